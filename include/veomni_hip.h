@@ -1,0 +1,148 @@
+/* veomni_hip.h — C ABI of libveomni_hip.so (gfx950 / MI355X hot-path kernels).
+ *
+ * This is the drop-in boundary under the Python operator API.
+ * Each entry replaces one reference kernel (anchor cited per function).
+ * Conventions:
+ *   - caller owns every buffer (device pointers from torch tensors);
+ *   - all entries are stream-ordered on the passed hipStream_t (as void*);
+ *   - return 0 on success, nonzero error code otherwise; vh_last_error()
+ *     returns a thread-local message for the last failure;
+ *   - no threads are spawned inside; no allocation except where stated;
+ *   - bf16 buffers are passed as uint16_t*.
+ */
+#ifndef VEOMNI_HIP_H
+#define VEOMNI_HIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Last error message (thread-local). */
+const char* vh_last_error(void);
+
+/* Build tag: returns a static string identifying arch + version. */
+const char* vh_build_info(void);
+
+/* ---- MoE token bookkeeping --------------------------------------------- */
+
+/* Histogram of expert ids. Bit-exact.
+ * Replaces: veomni/ops/kernels/moe/_kernels/kernel/moe.py:29-82.
+ * expert_index: int64 [n]; out: int32 [num_experts] (pre-zeroed NOT required;
+ * kernel zeroes it). */
+int vh_expert_histogram(const int64_t* expert_index, int64_t n,
+                        int num_experts, int32_t* out, void* stream);
+
+/* Row scatter: out[index[i]] = x[i / topk]'s row — i.e. for each token row t
+ * and slot k, out[index[t*topk+k]] = x[t]. bf16 rows.
+ * Replaces: kernel/moe.py:253-333 (_moe_scatter_kernel).
+ * x: [M, N] bf16; index: int32 [M*topk]; out: [M*topk, N] bf16. */
+int vh_moe_scatter_bf16(const uint16_t* x, const int32_t* index, uint16_t* out,
+                        int64_t M, int64_t N, int topk, void* stream);
+
+/* Row gather with fp32 accumulation over topk:
+ * out[t] = sum_k x[index[t*topk+k]].
+ * Replaces: kernel/moe.py:87-159 (_moe_gather_kernel).
+ * x: [M*topk, N] bf16; out: [M, N] bf16. */
+int vh_moe_gather_bf16(const uint16_t* x, const int32_t* index, uint16_t* out,
+                       int64_t M, int64_t N, int topk, void* stream);
+
+/* ---- Grouped GEMM (MFMA bf16, fp32 accumulate) ------------------------- */
+
+/* Per-group GEMM, shared N,K; group g owns rows [cumsum[g-1], cumsum[g]) of
+ * A and C.  trans_b!=0: B is [G,N,K] and C_g = A_g @ B_g^T;
+ * trans_b==0: B is [G,K,N] and C_g = A_g @ B_g.
+ * activation: 0 none; 1 fused SiLU on the output (bf16 rounding after).
+ * accumulate!=0: C += result (C read as bf16).
+ * cumsum: int64 device [G], inclusive.
+ * Replaces: kernel/group_gemm.py:66-234 (group_gemm_same_nk_kernel).
+ * Constraints (v1): N % 16 == 0, K % 64 == 0. */
+int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                          const int64_t* cumsum, int G, int64_t N, int64_t K,
+                          int64_t total_rows, int trans_b, int accumulate,
+                          int activation, void* stream);
+
+/* Per-group wgrad: C[g] = A_g^T @ B_g with per-group row count (k) from
+ * cumsum; A: [rows, M], B: [rows, N], C: [G, M, N] bf16 (fp32 accum).
+ * Groups with zero rows are zero-filled.
+ * Replaces: kernel/group_gemm.py:252-397 (group_gemm_same_mn_kernel,
+ * transpose_a=True, transpose_b=False — the only variant on the path). */
+int vh_group_gemm_mn_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                          const int64_t* cumsum, int G, int64_t M, int64_t N,
+                          void* stream);
+
+/* ---- Fused MoE elementwise --------------------------------------------- */
+
+/* Fused epilogue: act = silu(gate) * up * w_row, where gate/up are the two
+ * halves of fc1 [rows, 2I] and w_row is the per-row routing weight.
+ * Saves nothing; backward recomputes silu (ref moe_layer.py:328).
+ * fc1: [rows, 2I] bf16; w: fp32-or-bf16 per-row [rows]; out: [rows, I] bf16. */
+int vh_moe_silu_mul_weighted_bf16(const uint16_t* fc1, const uint16_t* w_row,
+                                  uint16_t* out, int64_t rows, int64_t I,
+                                  int has_w, void* stream);
+
+/* Backward of the fused epilogue:
+ * given dy [rows, I], fc1 [rows,2I] (pre-activation, saved), w_row,
+ * produce dfc1 [rows, 2I] and (optional) dw_row [rows] fp32
+ * (dw_row[t] = sum_i silu(g)*u * dy — the routing-weight grad numerator). */
+int vh_moe_silu_mul_weighted_bwd_bf16(const uint16_t* dy, const uint16_t* fc1,
+                                      const uint16_t* w_row, uint16_t* dfc1,
+                                      float* dw_row, int64_t rows, int64_t I,
+                                      int has_w, void* stream);
+
+/* ---- RMSNorm ------------------------------------------------------------ */
+
+/* y = w * cast_bf16(x_f32 * rsqrt(mean(x^2)+eps)); saves rstd fp32 [T].
+ * Replaces the Liger RMSNorm slot (ref ops/liger/__init__.py:28-60); math
+ * order matches eager Qwen3MoeRMSNorm (patched modeling :380-400). */
+int vh_rmsnorm_fwd_bf16(const uint16_t* x, const uint16_t* w, uint16_t* y,
+                        float* rstd, int64_t T, int64_t H, float eps,
+                        void* stream);
+
+/* Backward: dx bf16 [T,H]; dw accumulated fp32 [H] (caller zeroes dw). */
+int vh_rmsnorm_bwd_bf16(const uint16_t* dy, const uint16_t* x,
+                        const uint16_t* w, const float* rstd, uint16_t* dx,
+                        float* dw, int64_t T, int64_t H, void* stream);
+
+/* ---- RoPE --------------------------------------------------------------- */
+
+/* In-place-capable rotate-half RoPE on [B, h, S, D] q and k with cos/sin
+ * [B, S, D] (halves duplicated). backward = same call with negated sin
+ * (host passes sign=-1). Replaces Liger RoPE slot (liger/__init__.py:115-126);
+ * math matches patched modeling :94-111. */
+int vh_rope_bf16(const uint16_t* q, const uint16_t* k, const uint16_t* cos_t,
+                 const uint16_t* sin_t, uint16_t* q_out, uint16_t* k_out,
+                 int64_t B, int64_t hq, int64_t hk, int64_t S, int64_t D,
+                 int negate_sin, void* stream);
+
+/* ---- SwiGLU ------------------------------------------------------------- */
+
+/* out = silu(gate) * up  (bf16; fp32 internal).
+ * Replaces LigerSiLUMulFunction slot (liger/__init__.py:130-141). */
+int vh_silu_mul_bf16(const uint16_t* gate, const uint16_t* up, uint16_t* out,
+                     int64_t n, void* stream);
+
+/* dgate = dy * up * silu'(gate); dup = dy * silu(gate). */
+int vh_silu_mul_bwd_bf16(const uint16_t* dy, const uint16_t* gate,
+                         const uint16_t* up, uint16_t* dgate, uint16_t* dup,
+                         int64_t n, void* stream);
+
+/* ---- Fused chunked cross-entropy ---------------------------------------- */
+
+/* Per-row softmax CE over a bf16 logits chunk:
+ *   loss_rows[r]  = is_valid * (logsumexp(logits[r]) - logits[r][label]);
+ *   dlogits[r][v] = is_valid * (softmax - onehot) * grad_scale.
+ * fp32 online logsumexp; ignored rows (label == ignore_index) produce 0.
+ * Replaces the inner loop of chunk_loss (ref chunk_loss.py:109-127 +
+ * transformers fixed_cross_entropy). */
+int vh_ce_fwd_bf16(const uint16_t* logits, const int64_t* labels,
+                   float* loss_rows, uint16_t* dlogits, int64_t rows,
+                   int64_t V, float grad_scale, int64_t ignore_index,
+                   void* stream);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* VEOMNI_HIP_H */

@@ -1,0 +1,335 @@
+// Grouped GEMM on gfx950 MFMA (bf16 in, fp32 accumulate).
+//
+// Semantics anchors (restated, not copied):
+//   group_gemm_same_nk — ref kernel/group_gemm.py:66-234: G per-expert GEMMs,
+//     rows of A/C partitioned by an inclusive cumsum; shared N,K; out-of-range
+//     row loads wrapped modulo the group's row count and stores masked.
+//   group_gemm_same_mn — ref kernel/group_gemm.py:252-397: per-group wgrad
+//     C[g] = A_g^T @ B_g with per-group K = row count; zero-count groups are
+//     zero-filled.
+//
+// MI355X design (guide cdna_hip_programming.md §5):
+//   * v_mfma_f32_16x16x32_bf16; 128x128x64 tile; 256 threads = 4 waves in a
+//     2x2 wave grid, 64x64 per wave (4x4 fragments, f32x4 accumulators).
+//   * K-contiguous operands staged via global_load_lds width 16 into a
+//     lane-linear LDS image; the T2 XOR swizzle ((row&7)<<4) is applied on the
+//     per-lane SOURCE address and on the ds_read address (rule 21).
+//   * K-strided operands (dgrad's B, wgrad's A and B) staged through
+//     registers: per-lane column loads are coalesced ACROSS lanes (consecutive
+//     n per lane), written as 16-B ds_writes into the transposed image.
+//   * 2-phase double buffer: stage tile t+1, compute tile t, one
+//     __syncthreads per K-step (guide T3 "minimum 2-phase" recipe).
+
+#include "vh_common.h"
+
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int THREADS = 256;
+
+using bf16frag = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ void glds16(const bf16_t* g, bf16_t* l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
+}
+
+// byte-offset XOR swizzle within a 128-B row (8 x 16-B slots)
+__device__ __forceinline__ int swz(int row, int colb) {
+  return colb ^ ((row & 7) << 4);
+}
+
+// Stage a [128][BK] K-contiguous tile (A, or B when rows of B are
+// K-contiguous) via glds. `row_of(r)` maps tile row -> global row index.
+// LDS image: linear [row][colb], data swizzled so that
+// lds[row][colb] = src[row][colb ^ ((row&7)<<4)].
+template <typename RowFn>
+__device__ __forceinline__ void stage_kcontig(const bf16_t* src, int64_t ld_elems,
+                                              bf16_t* lds_tile, int64_t k0,
+                                              RowFn row_of, int tid) {
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    // glds writes wave-uniform LDS base + lane*16 (guide §5 caveat): pass the
+    // WAVE base; the per-lane SOURCE address carries the lane offset and the
+    // inverse swizzle (rule 21).
+    int base = i * 4096 + wave * 1024;
+    int o = base + lane * 16;  // this lane's effective dest byte offset
+    int row = o >> 7;          // 128 B per row
+    int colb = o & 127;
+    int src_colb = swz(row, colb);
+    const bf16_t* g = src + row_of(row) * ld_elems + k0 + (src_colb >> 1);
+    glds16(g, lds_tile + (base >> 1));
+  }
+}
+
+// Stage a [128][BK] tile TRANSPOSED from a [BK][128-col window] source
+// (element (out,k) read from src[k*ld + out]). Coalesced across lanes
+// (consecutive `out` per lane); 16-B ds_writes; same swizzled image as
+// stage_kcontig. kmax clamps the k range (zero fill) for ragged group rows.
+__device__ __forceinline__ void stage_transposed(const bf16_t* src, int64_t ld_elems,
+                                                 bf16_t* lds_tile, int64_t k0,
+                                                 int64_t kmax, int out0,
+                                                 int out_max, int tid) {
+  // thread t: out = t % 128, k chunk = (t / 128) * 32 .. +32
+  int out = tid & 127;
+  int kbase = (tid >> 7) * 32;
+  int64_t gout = out0 + out;
+  bool out_ok = gout < out_max;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {  // 4 chunks of 8 k
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int64_t k = k0 + kbase + c * 8 + j;
+      v.v[j] = (out_ok && k < kmax) ? src[k * ld_elems + gout] : (bf16_t)0;
+    }
+    int colb = (kbase + c * 8) * 2;
+    *reinterpret_cast<bf16x8*>(&lds_tile[(out * 128 + swz(out, colb)) >> 1]) = v;
+  }
+}
+
+// read one 16x32 MFMA A/B fragment from the swizzled [row][64] bf16 image
+__device__ __forceinline__ bf16frag frag_read(const bf16_t* lds_tile, int row0,
+                                              int ks, int lane) {
+  int row = row0 + (lane & 15);
+  int colb = (ks * 32 + ((lane >> 4) << 3)) * 2;
+  int off_b = row * 128 + swz(row, colb);
+  return *reinterpret_cast<const bf16frag*>(
+      reinterpret_cast<const char*>(lds_tile) + off_b);
+}
+
+__device__ __forceinline__ float maybe_act(float x, int ACT) {
+  return ACT == 1 ? siluf(x) : x;
+}
+
+// ---------------------------------------------------------------- same_nk
+template <bool TRANS_B, bool ACCUM, int ACT>
+__global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: A0, B0, A1, B1 — each 128*64*2 = 16 KiB
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 16384; };
+  auto tb = [&](int buf) { return sm + 8192 + buf * 16384; };
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t row_end = cumsum[gid];
+  const int64_t m_size = row_end - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+  if ((int64_t)bm * BM >= m_size) return;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid
+
+  const bf16_t* Ag = A + row_start * K;
+  const bf16_t* Bg = B + (int64_t)gid * N * K;
+  bf16_t* Cg = C + row_start * N;
+
+  auto a_row = [&](int r) -> int64_t {
+    int64_t gm = (int64_t)bm * BM + r;
+    return gm % m_size;  // wrap like the reference; stores are masked
+  };
+  auto b_row = [&](int r) -> int64_t {
+    int64_t gn = (int64_t)bn * BN + r;
+    return gn % N;
+  };
+
+  auto stage = [&](int buf, int64_t k0) {
+    stage_kcontig(Ag, K, ta(buf), k0, a_row, tid);
+    if (TRANS_B) {
+      stage_kcontig(Bg, K, tb(buf), k0, b_row, tid);
+    } else {
+      // B is [K, N]: transposed staging of the [k0..k0+BK) x [bn*BN ..) window
+      stage_transposed(Bg, N, tb(buf), k0, K, bn * BN, (int)N, tid);
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = (int)(K / BK);
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) stage(cur ^ 1, (int64_t)(t + 1) * BK);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16frag af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read(ta(cur), wr * 64 + i * 16, ks, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bf[j] = frag_read(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: C row = (lane>>4)*4 + reg, col = lane&15 within each 16x16 frag
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM + wr * 64 + i * 16 + row_base_in + rr;
+        int64_t n = (int64_t)bn * BN + wc * 64 + j * 16 + col_in;
+        if (m < m_size && n < N) {
+          float v = acc[i][j][rr];
+          if (ACCUM) v += bf2f(Cg[m * N + n]);
+          Cg[m * N + n] = f2bf(maybe_act(v, ACT));
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- same_mn
+// C[g, M, N] = A_g^T @ B_g ; A [rows, M], B [rows, N]; k = group row count.
+__global__ __launch_bounds__(THREADS, 2) void k_group_gemm_mn(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t M, int64_t N, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 16384; };
+  auto tb = [&](int buf) { return sm + 8192 + buf * 16384; };
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t row_end = cumsum[gid];
+  const int64_t kcount = row_end - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  bf16_t* Cg = C + (int64_t)gid * M * N;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  if (kcount > 0) {
+    const bf16_t* Ag = A + row_start * M;
+    const bf16_t* Bg = B + row_start * N;
+    const int nk = (int)((kcount + BK - 1) / BK);
+    auto stage = [&](int buf, int64_t k0) {
+      stage_transposed(Ag, M, ta(buf), k0, kcount, bm * BM, (int)M, tid);
+      stage_transposed(Bg, N, tb(buf), k0, kcount, bn * BN, (int)N, tid);
+    };
+    stage(0, 0);
+    __syncthreads();
+    int cur = 0;
+    for (int t = 0; t < nk; ++t) {
+      if (t + 1 < nk) stage(cur ^ 1, (int64_t)(t + 1) * BK);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16frag af[4], bf[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read(ta(cur), wr * 64 + i * 16, ks, lane);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) bf[j] = frag_read(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM + wr * 64 + i * 16 + row_base_in + rr;
+        int64_t n = (int64_t)bn * BN + wc * 64 + j * 16 + col_in;
+        if (m < M && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+}  // namespace
+
+extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
+                                     uint16_t* C, const int64_t* cumsum, int G,
+                                     int64_t N, int64_t K, int64_t total_rows,
+                                     int trans_b, int accumulate,
+                                     int activation, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % BK == 0, "K %% 64 != 0 (K=%lld)", (long long)K);
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0 (N=%lld)", (long long)N);
+  VH_CHECK(G >= 1, "G < 1");
+  int tiles_m = (int)((total_rows + BM - 1) / BM);
+  if (tiles_m < 1) tiles_m = 1;
+  int tiles_n = (int)((N + BN - 1) / BN);
+  dim3 grid(tiles_m * tiles_n, G);
+  size_t lds = 65536;
+
+#define VH_DISPATCH(TB, AC, ACT_)                                              \
+  hipLaunchKernelGGL((k_group_gemm_nk<TB, AC, ACT_>), grid, dim3(THREADS),     \
+                     lds, s, reinterpret_cast<const bf16_t*>(A),               \
+                     reinterpret_cast<const bf16_t*>(B),                       \
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_m,   \
+                     tiles_n)
+
+  if (trans_b) {
+    if (accumulate) VH_DISPATCH(true, true, 0);
+    else if (activation == 1) VH_DISPATCH(true, false, 1);
+    else VH_DISPATCH(true, false, 0);
+  } else {
+    if (accumulate) VH_DISPATCH(false, true, 0);
+    else if (activation == 1) VH_DISPATCH(false, false, 1);
+    else VH_DISPATCH(false, false, 0);
+  }
+#undef VH_DISPATCH
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+extern "C" int vh_group_gemm_mn_bf16(const uint16_t* A, const uint16_t* B,
+                                     uint16_t* C, const int64_t* cumsum, int G,
+                                     int64_t M, int64_t N, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(M % 16 == 0 && N % 16 == 0, "M/N %% 16 != 0");
+  int tiles_m = (int)((M + BM - 1) / BM);
+  int tiles_n = (int)((N + BN - 1) / BN);
+  dim3 grid(tiles_m * tiles_n, G);
+  hipLaunchKernelGGL(k_group_gemm_mn, grid, dim3(THREADS), 65536, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, M, N, tiles_m,
+                     tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

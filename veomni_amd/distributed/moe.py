@@ -1,0 +1,180 @@
+"""Expert-parallel MoE dispatch/combine over RCCL all-to-all.
+
+API parity target: /root/reference/veomni/distributed/moe/
+  moe_layer.py:48-224 (`preprocess`, `token_pre_all2all`,
+  `tokens_post_all2all`, `dispatch_to_ep_class`)
+  comm.py:20-54 (`_AllToAll` autograd over dist.all_to_all_single)
+  moe_utils.py:19-99 (`permute`, `unpermute`, `generate_weights_idx`,
+  `sort_chunks_by_idxs`).
+
+The expert-MLP autograd classes (grouped GEMMs on local experts) live in
+veomni_amd.ops.kernels.moe — this module is pure dispatch plumbing, so the
+gloo ws=2 CPU tests cover it with a torch ep_class stub while the GPU path
+passes the HIP grouped-GEMM classes.
+
+MI355X note: the a2a payload (~T·topk·H·2B per layer per rank) rides the
+xGMI full mesh where all-to-all is link-parallel — the cheap collective.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Callable, Optional
+
+import torch
+import torch.distributed as dist
+
+from .parallel_state import get_parallel_state
+
+
+class _AllToAll(torch.autograd.Function):
+    """Variable-split a2a; backward = a2a with the splits swapped
+    (ref comm.py:20-54)."""
+
+    @staticmethod
+    def forward(ctx, group, x, output_split_sizes, input_split_sizes):
+        ctx.group = group
+        ctx.output_split_sizes = output_split_sizes
+        ctx.input_split_sizes = input_split_sizes
+        if dist.get_world_size(group=group) == 1:
+            return x
+        x = x.contiguous()
+        if output_split_sizes is None:
+            out = torch.empty_like(x)
+        else:
+            out = torch.empty((sum(output_split_sizes), x.size(1)), dtype=x.dtype, device=x.device)
+        dist.all_to_all_single(
+            out, x, output_split_sizes=output_split_sizes,
+            input_split_sizes=input_split_sizes, group=group,
+        )
+        return out
+
+    @staticmethod
+    def backward(ctx, *grad_output):
+        return (None, _AllToAll.apply(ctx.group, *grad_output, ctx.input_split_sizes, ctx.output_split_sizes), None, None)
+
+
+def all_to_all(group, x, output_split_sizes=None, input_split_sizes=None):
+    return _AllToAll.apply(group, x, output_split_sizes, input_split_sizes)
+
+
+def permute(tokens: torch.Tensor, routing_map: torch.Tensor):
+    """Expert-major stable permutation (ref moe_utils.py:19-41).
+
+    routing_map: [num_experts, num_tokens] 0/1. Returns (permuted, mapping)
+    where mapping[i] = source token row of permuted row i.
+    """
+    num_tokens = tokens.shape[0]
+    num_experts = routing_map.shape[0]
+    token_indices = (
+        torch.arange(num_tokens, device=routing_map.device).unsqueeze(0).expand(num_experts, -1)
+    )
+    sorted_indices = token_indices.masked_select(routing_map.bool())
+    return tokens.index_select(0, sorted_indices), sorted_indices
+
+
+def unpermute(tokens, routing_weights, hidden_states_shape, permutation_mapping, routing_map):
+    """Weighted fp32 scatter_add unpermute (ref moe_utils.py:44-72)."""
+    tokens_weight = routing_weights.T.contiguous().masked_select(routing_map.bool())
+    tokens = tokens * tokens_weight.unsqueeze(-1)
+    hidden_dim = hidden_states_shape[-1]
+    out = torch.zeros(hidden_states_shape, device=tokens.device, dtype=torch.float32)
+    out.scatter_add_(0, permutation_mapping.unsqueeze(1).expand(-1, hidden_dim), tokens.float())
+    return out.to(tokens.dtype)
+
+
+def generate_weights_idx(routing_weights, selected_experts, num_experts):
+    """[T, topk] weights -> dense [T, E] (ref moe_utils.py:75-92)."""
+    T, _ = routing_weights.shape
+    out = torch.zeros((T, num_experts), dtype=routing_weights.dtype, device=routing_weights.device)
+    out.scatter_add_(1, selected_experts, routing_weights)
+    return out
+
+
+def sort_chunks_by_idxs(x, split_sizes, sorted_idxs):
+    """Reorder row-chunks (ref moe_utils.py:95-99)."""
+    chunks = torch.split(x, split_sizes.tolist() if torch.is_tensor(split_sizes) else split_sizes, dim=0)
+    return torch.cat([chunks[i] for i in sorted_idxs], dim=0)
+
+
+def preprocess(expert_mask: torch.Tensor, num_experts: int, ep_group) -> tuple:
+    """Exchange per-expert counts; derive a2a splits (ref moe_layer.py:48-87)."""
+    ep_size = ep_group.size()
+    num_local_experts = num_experts // ep_size
+    rank = dist.get_rank(ep_group)
+    num_local_tokens_per_expert = expert_mask.sum(dim=(1, 2))
+
+    input_splits = (
+        num_local_tokens_per_expert.reshape(ep_size, num_local_experts).sum(dim=1).tolist()
+    )
+    num_global = torch.zeros(
+        ep_size, num_local_tokens_per_expert.size(0),
+        dtype=num_local_tokens_per_expert.dtype, device=num_local_tokens_per_expert.device,
+    )
+    dist.all_gather_into_tensor(num_global, num_local_tokens_per_expert, group=ep_group)
+
+    start, end = rank * num_local_experts, (rank + 1) * num_local_experts
+    num_global_per_local = num_global[:, start:end].contiguous()
+    output_splits = num_global_per_local.sum(dim=1).tolist()
+    num_global_sum_per_local = num_global_per_local.sum(dim=0).to("cpu", non_blocking=True)
+    num_global_per_local = num_global_per_local.view(-1, num_local_experts).to("cpu", non_blocking=True)
+    return input_splits, output_splits, num_global_per_local, num_global_sum_per_local
+
+
+def token_pre_all2all(hidden_states, expert_mask, num_experts, input_splits,
+                      output_splits, num_global_tokens_per_local_expert, ep_group):
+    """Permute + dispatch a2a + expert-major chunk resort (ref moe_layer.py:90-117)."""
+    hidden_dim = hidden_states.size(-1)
+    hidden_states = hidden_states.reshape(-1, hidden_dim)
+    org_shape = hidden_states.shape
+    routing_map = expert_mask.sum(dim=1)
+
+    local_permuted, local_mapping = permute(hidden_states, routing_map)
+    global_permuted = all_to_all(ep_group, local_permuted, output_splits, input_splits)
+
+    num_local_experts = num_experts // ep_group.size()
+    permute_order = torch.arange(num_experts).reshape(-1, num_local_experts).T.ravel().tolist()
+    global_permuted = sort_chunks_by_idxs(
+        global_permuted, num_global_tokens_per_local_expert.ravel(), permute_order
+    )
+    return global_permuted, routing_map, local_mapping, org_shape
+
+
+def tokens_post_all2all(expert_outputs, routing_weights, selected_experts, num_experts,
+                        input_splits, output_splits, num_global_tokens_per_local_expert,
+                        routing_map, local_input_permutation_mapping,
+                        org_hidden_states_shape, ep_group):
+    """Inverse resort + combine a2a + weighted unpermute (ref moe_layer.py:189-224)."""
+    num_local_experts = num_experts // ep_group.size()
+    unpermute_order = torch.arange(num_experts).reshape(num_local_experts, -1).T.ravel().tolist()
+    expert_outputs = sort_chunks_by_idxs(
+        expert_outputs, num_global_tokens_per_local_expert.T.ravel(), unpermute_order
+    )
+    out = all_to_all(ep_group, expert_outputs, input_splits, output_splits)
+    weights_idx = generate_weights_idx(routing_weights, selected_experts, num_experts)
+    return unpermute(out, weights_idx, org_hidden_states_shape,
+                     local_input_permutation_mapping, routing_map)
+
+
+def dispatch_to_ep_class(ep_class: Callable, num_experts: int, routing_weights,
+                         selected_experts, hidden_states, *ep_class_args: Any):
+    """Shared EP plumbing (ref moe_layer.py:120-186): preprocess ->
+    token_pre_all2all -> ep_class.apply(permute_tokens, cumsum, *args) ->
+    tokens_post_all2all."""
+    ep_state = get_parallel_state()
+    expert_mask = torch.nn.functional.one_hot(
+        selected_experts, num_classes=num_experts
+    ).permute(2, 1, 0)
+    input_splits, output_splits, num_global_per_local, num_global_sum_per_local = preprocess(
+        expert_mask, num_experts, ep_state.ep_group
+    )
+    permute_tokens, routing_map, local_mapping, org_shape = token_pre_all2all(
+        hidden_states, expert_mask, num_experts, input_splits, output_splits,
+        num_global_per_local, ep_state.ep_group,
+    )
+    cumsum = torch.cumsum(num_global_sum_per_local, dim=0).to(permute_tokens.device)
+    final_permute_tokens = ep_class.apply(permute_tokens, cumsum, *ep_class_args)
+    return tokens_post_all2all(
+        final_permute_tokens, routing_weights, selected_experts, num_experts,
+        input_splits, output_splits, num_global_per_local, routing_map,
+        local_mapping, org_shape, ep_state.ep_group,
+    )

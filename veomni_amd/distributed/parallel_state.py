@@ -1,0 +1,198 @@
+"""Parallel-state topology for the FSDP2 training step.
+
+API parity target: /root/reference/veomni/distributed/parallel_state.py
+(`init_parallel_state` :444-656, `get_parallel_state` :694-701, property set
+:60-419). Round-1 scope covers the dims the §8 hot path uses:
+(dp_shard, ulysses) as one device mesh, flattened `dp_shard_sp` as the FSDP
+shard mesh, and a separate (ep_fsdp, ep) mesh with ep_size | dp_shard_sp
+(ref :598-627). pp/tp/cp/dp_replicate are fixed at 1 this round.
+
+MI355X mapping: one process per GPU; the mesh's process groups are RCCL over
+xGMI on "cuda" (= ROCm) devices, gloo on CPU for tests.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+from torch.distributed.device_mesh import DeviceMesh, init_device_mesh
+
+_PARALLEL_STATE: Optional["ParallelState"] = None
+
+
+@dataclass
+class ParallelState:
+    world_size: int = 1
+    dp_size: int = 1           # dp_shard size
+    ulysses_size: int = 1
+    ep_size: int = 1
+    dp_mode: str = "fsdp2"
+    device_type: str = "cpu"
+    device_mesh: Optional[DeviceMesh] = None      # (dp_shard, ulysses)
+    ep_device_mesh: Optional[DeviceMesh] = None   # (ep_fsdp, ep)
+    _fsdp_mesh: Optional[DeviceMesh] = None       # flattened dp_shard_sp
+    extra_parallel_names: tuple = ("ep",)
+
+    # ---------------------------------------------------------------- flags
+    @property
+    def ulysses_enabled(self) -> bool:
+        return self.ulysses_size > 1
+
+    @property
+    def sp_enabled(self) -> bool:
+        # Ulysses is the only SP flavour on the §8 path (cp is dead in the
+        # reference too: parallel_state.py:81-82).
+        return self.ulysses_enabled
+
+    @property
+    def ep_enabled(self) -> bool:
+        return self.ep_size > 1
+
+    def extra_parallel_enabled(self, name: str) -> bool:
+        return name == "ep" and self.ep_enabled
+
+    # ---------------------------------------------------------------- meshes
+    @property
+    def fsdp_mesh(self) -> Optional[DeviceMesh]:
+        return self._fsdp_mesh
+
+    @property
+    def fsdp_size(self) -> int:
+        return self.dp_size * self.ulysses_size
+
+    @property
+    def ep_fsdp_mesh(self) -> Optional[DeviceMesh]:
+        if self.ep_device_mesh is None:
+            return None
+        return self.ep_device_mesh["ep_fsdp"]
+
+    @property
+    def ep_fsdp_size(self) -> int:
+        return self.fsdp_size // self.ep_size
+
+    # ---------------------------------------------------------------- groups
+    @property
+    def dp_group(self):
+        if self.device_mesh is None:
+            return None
+        return self.device_mesh["dp_shard"].get_group()
+
+    @property
+    def ulysses_group(self):
+        if self.device_mesh is None or not self.ulysses_enabled:
+            return None
+        return self.device_mesh["ulysses"].get_group()
+
+    @property
+    def sp_group(self):
+        return self.ulysses_group
+
+    @property
+    def ep_group(self):
+        if self.ep_device_mesh is None:
+            return None
+        return self.ep_device_mesh["ep"].get_group()
+
+    # ------------------------------------------------------------ ranks/sizes
+    @property
+    def sp_size(self) -> int:
+        return self.ulysses_size
+
+    @property
+    def sp_rank(self) -> int:
+        return self.ulysses_rank
+
+    @property
+    def ulysses_rank(self) -> int:
+        if not self.ulysses_enabled:
+            return 0
+        return dist.get_rank(self.ulysses_group)
+
+    @property
+    def ep_rank(self) -> int:
+        if not self.ep_enabled:
+            return 0
+        return dist.get_rank(self.ep_group)
+
+    @property
+    def dp_rank(self) -> int:
+        if self.device_mesh is None:
+            return 0
+        return dist.get_rank(self.dp_group)
+
+    # grad divide factor for EP modules — always world size
+    # (ref parallel_state.py:348-356).
+    def extra_parallel_gradient_divide_factor(self, name: str) -> int:
+        return self.world_size
+
+    @property
+    def ep_gradient_divide_factor(self) -> int:
+        return self.extra_parallel_gradient_divide_factor("ep")
+
+
+def init_parallel_state(
+    dp_size: Optional[int] = None,
+    ulysses_size: int = 1,
+    ep_size: int = 1,
+    dp_mode: str = "fsdp2",
+    device_type: Optional[str] = None,
+) -> ParallelState:
+    """Build the device meshes and register the global state.
+
+    Mirrors the mesh construction of reference parallel_state.py:444-656:
+    one mesh (dp_shard, ulysses), a flattened `dp_shard_sp` sub-mesh used as
+    the FSDP shard mesh, and a second mesh (ep_fsdp, ep) with the constraint
+    ep_size | dp_shard_sp_size (ref :598-627).
+    """
+    global _PARALLEL_STATE
+    assert dist.is_initialized(), "init_process_group first"
+    world_size = dist.get_world_size()
+    if device_type is None:
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    if dp_size is None:
+        assert world_size % ulysses_size == 0
+        dp_size = world_size // ulysses_size
+    assert dp_size * ulysses_size == world_size, (dp_size, ulysses_size, world_size)
+    dp_shard_sp = dp_size * ulysses_size
+    assert dp_shard_sp % ep_size == 0, f"ep_size {ep_size} must divide dp_shard*sp {dp_shard_sp}"
+
+    mesh = init_device_mesh(
+        device_type, (dp_size, ulysses_size), mesh_dim_names=("dp_shard", "ulysses")
+    )
+    fsdp_mesh = mesh["dp_shard", "ulysses"]._flatten(mesh_dim_name="dp_shard_sp")
+
+    ep_mesh = None
+    if ep_size > 1:
+        ep_mesh = init_device_mesh(
+            device_type, (dp_shard_sp // ep_size, ep_size), mesh_dim_names=("ep_fsdp", "ep")
+        )
+
+    _PARALLEL_STATE = ParallelState(
+        world_size=world_size,
+        dp_size=dp_size,
+        ulysses_size=ulysses_size,
+        ep_size=ep_size,
+        dp_mode=dp_mode,
+        device_type=device_type,
+        device_mesh=mesh,
+        ep_device_mesh=ep_mesh,
+        _fsdp_mesh=fsdp_mesh,
+    )
+    return _PARALLEL_STATE
+
+
+def get_parallel_state() -> ParallelState:
+    """Current state; an uninitialized process gets a single-process default
+    (ref parallel_state.py:694-701)."""
+    global _PARALLEL_STATE
+    if _PARALLEL_STATE is None:
+        return ParallelState()
+    return _PARALLEL_STATE
+
+
+def set_parallel_state(state: Optional[ParallelState]) -> None:
+    global _PARALLEL_STATE
+    _PARALLEL_STATE = state

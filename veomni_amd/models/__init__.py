@@ -1,0 +1,58 @@
+"""Model presets for the BASELINE.json configs."""
+
+from .modeling import ForCausalLM, ModelConfig, bind_ops  # noqa: F401
+
+PRESETS = {
+    # BASELINE config 1 plumbing model (Qwen2-0.5B-shaped)
+    "qwen2-0.5b": ModelConfig(
+        name="qwen2-0.5b", vocab_size=151936, hidden_size=896, intermediate_size=4864,
+        num_hidden_layers=24, num_attention_heads=14, num_key_value_heads=2,
+        head_dim=64, rope_theta=1000000.0, attention_bias=True, qk_norm=False,
+        tie_word_embeddings=True,
+    ),
+    # BASELINE config 2: Llama-3-8B
+    "llama3-8b": ModelConfig(
+        name="llama3-8b", vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+        num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
+        head_dim=128, rope_theta=500000.0, rms_norm_eps=1e-5, qk_norm=False,
+    ),
+    # BASELINE config 3 (north star): Qwen3-30B-A3B
+    "qwen3-moe-30b": ModelConfig(
+        name="qwen3-moe-30b", vocab_size=151936, hidden_size=2048, intermediate_size=6144,
+        num_hidden_layers=48, num_attention_heads=32, num_key_value_heads=4,
+        head_dim=128, rope_theta=1000000.0, qk_norm=True,
+        num_experts=128, num_experts_per_tok=8, moe_intermediate_size=768,
+        norm_topk_prob=True, router_aux_loss_coef=0.001, initializer_range=0.02,
+    ),
+    # toy fixture matching the reference's tests/toy_config/qwen3_moe_toy
+    "qwen3-moe-toy": ModelConfig(
+        name="qwen3-moe-toy", vocab_size=151936, hidden_size=2048, intermediate_size=6144,
+        num_hidden_layers=4, num_attention_heads=32, num_key_value_heads=4,
+        head_dim=128, rope_theta=1000000.0, qk_norm=True, initializer_range=0.05,
+        num_experts=16, num_experts_per_tok=2, moe_intermediate_size=768,
+        norm_topk_prob=True, router_aux_loss_coef=0.001,
+    ),
+    # small variants for unit tests
+    "tiny-dense": ModelConfig(
+        name="tiny-dense", vocab_size=512, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=32, qk_norm=False,
+    ),
+    "tiny-moe": ModelConfig(
+        name="tiny-moe", vocab_size=512, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=32, qk_norm=True, num_experts=8, num_experts_per_tok=2,
+        moe_intermediate_size=64, router_aux_loss_coef=0.001,
+    ),
+}
+
+
+def build_model(preset: str, dtype=None, device=None) -> ForCausalLM:
+    import torch
+
+    cfg = PRESETS[preset]
+    with torch.device(device) if device is not None else torch.device("cpu"):
+        model = ForCausalLM(cfg)
+    if dtype is not None:
+        model = model.to(dtype)
+    return model
