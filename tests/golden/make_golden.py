@@ -268,6 +268,43 @@ golden["ep/sorted_chunks"] = sort_chunks_by_idxs(toks, split_sizes, order)
 golden["ep/sorted_chunks_sizes"] = split_sizes
 golden["ep/sorted_chunks_order"] = torch.tensor(order)
 
+# ── tiny end-to-end reference model fwd+bwd ─ ref: patched_modeling_qwen3_moe
+# _gpu.py (full model), loss via install_loss_mapping("chunk_loss") — the
+# reference's default CE backend (cross_entropy/__init__.py:445-516).
+from transformers.models.qwen3_moe.configuration_qwen3_moe import Qwen3MoeConfig  # noqa: E402
+from veomni.ops.kernels.cross_entropy import install_loss_mapping  # noqa: E402
+
+install_loss_mapping("chunk_loss")
+tiny_cfg = dict(
+    vocab_size=512, hidden_size=128, intermediate_size=256, num_hidden_layers=2,
+    num_attention_heads=4, num_key_value_heads=2, head_dim=32, num_experts=8,
+    num_experts_per_tok=2, moe_intermediate_size=64, norm_topk_prob=True,
+    rope_theta=1000000.0, rms_norm_eps=1e-6, tie_word_embeddings=False,
+    attention_bias=False, output_router_logits=False,
+)
+torch.manual_seed(21)
+ref_model = m.Qwen3MoeForCausalLM(Qwen3MoeConfig(**tiny_cfg, attn_implementation="sdpa"))
+with torch.no_grad():
+    for name, p in sorted(ref_model.named_parameters(), key=lambda kv: kv[0]):
+        gseed = torch.Generator().manual_seed(abs(hash(name)) % (2**31))
+        if "norm" in name and name.endswith("weight") and p.dim() == 1:
+            p.fill_(1.0)
+        else:
+            p.copy_(torch.randn(p.shape, generator=gseed) * 0.03)
+ids = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(99))
+out = ref_model(input_ids=ids, labels=ids.clone())
+out.loss.backward()
+golden["e2e/config"] = tiny_cfg
+golden["e2e/input_ids"] = ids
+golden["e2e/loss"] = out.loss.detach()
+golden["e2e/state_dict"] = {k: v.detach().clone() for k, v in ref_model.state_dict().items()}
+golden["e2e/grads"] = {
+    n: p.grad.detach().clone() for n, p in ref_model.named_parameters() if p.grad is not None
+}
+with torch.no_grad():
+    logits = ref_model(input_ids=ids).logits
+golden["e2e/logits"] = logits.detach()
+
 out_path = os.path.join(os.path.dirname(__file__), "golden.pt")
 torch.save(golden, out_path)
 print(f"wrote {out_path} with {len(golden)} entries")

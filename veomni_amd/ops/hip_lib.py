@@ -1,0 +1,231 @@
+"""ctypes binding of libveomni_hip.so (the C ABI in include/veomni_hip.h).
+
+The product path calls HIP kernels ONLY through here. There is no fallback:
+if the library is missing on a machine with a GPU, `get_lib()` raises — the
+"hip" kernel registrations must never silently degrade to eager/torch.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional
+
+import torch
+
+_LIB: Optional[ctypes.CDLL] = None
+
+_PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+SO_PATH = os.path.join(_PKG_DIR, "libveomni_hip.so")
+
+c_i64 = ctypes.c_int64
+c_int = ctypes.c_int
+c_f32 = ctypes.c_float
+c_p = ctypes.c_void_p
+
+
+def _sig(lib: ctypes.CDLL, name: str, *argtypes) -> None:
+    fn = getattr(lib, name)
+    fn.restype = ctypes.c_int
+    fn.argtypes = list(argtypes)
+
+
+def get_lib() -> ctypes.CDLL:
+    global _LIB
+    if _LIB is not None:
+        return _LIB
+    if not os.path.exists(SO_PATH):
+        raise RuntimeError(
+            f"libveomni_hip.so not found at {SO_PATH}. Build it with "
+            "`python veomni_amd/csrc/build.py` (or __graft_entry__.build()). "
+            "The HIP op path has NO fallback by design."
+        )
+    lib = ctypes.CDLL(SO_PATH)
+    lib.vh_last_error.restype = ctypes.c_char_p
+    lib.vh_build_info.restype = ctypes.c_char_p
+    _sig(lib, "vh_expert_histogram", c_p, c_i64, c_int, c_p, c_p)
+    _sig(lib, "vh_moe_scatter_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
+    _sig(lib, "vh_moe_gather_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
+    _sig(lib, "vh_group_gemm_nk_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64,
+         c_i64, c_int, c_int, c_int, c_p)
+    _sig(lib, "vh_group_gemm_mn_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
+    _sig(lib, "vh_moe_silu_mul_weighted_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
+    _sig(lib, "vh_moe_silu_mul_weighted_bwd_bf16", c_p, c_p, c_p, c_p, c_p,
+         c_i64, c_i64, c_int, c_p)
+    _sig(lib, "vh_rmsnorm_fwd_bf16", c_p, c_p, c_p, c_p, c_i64, c_i64, c_f32, c_p)
+    _sig(lib, "vh_rmsnorm_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_i64, c_i64, c_p)
+    _sig(lib, "vh_rope_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_i64, c_i64, c_i64,
+         c_i64, c_i64, c_int, c_p)
+    _sig(lib, "vh_silu_mul_bf16", c_p, c_p, c_p, c_i64, c_p)
+    _sig(lib, "vh_silu_mul_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
+    _sig(lib, "vh_ce_fwd_bf16", c_p, c_p, c_p, c_p, c_i64, c_i64, c_f32, c_i64, c_p)
+    _LIB = lib
+    return lib
+
+
+def cur_stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def check(rc: int, what: str) -> None:
+    if rc != 0:
+        raise RuntimeError(f"{what} failed (rc={rc}): {get_lib().vh_last_error().decode()}")
+
+
+def dptr(t: torch.Tensor) -> int:
+    return t.data_ptr()
+
+
+# ------------------------------------------------------------- typed wrappers
+def expert_histogram(expert_index: torch.Tensor, num_experts: int) -> torch.Tensor:
+    assert expert_index.dtype == torch.int64
+    flat = expert_index.flatten().contiguous()
+    out = torch.empty(num_experts, dtype=torch.int32, device=flat.device)
+    check(get_lib().vh_expert_histogram(dptr(flat), flat.numel(), num_experts,
+                                        dptr(out), cur_stream()), "vh_expert_histogram")
+    return out
+
+
+def moe_scatter(x: torch.Tensor, index: torch.Tensor) -> torch.Tensor:
+    M, N = x.shape
+    topk = index.shape[1] if index.dim() == 2 else 1
+    idx = index.reshape(-1).to(torch.int32).contiguous()
+    out = torch.empty(M * topk, N, dtype=x.dtype, device=x.device)
+    check(get_lib().vh_moe_scatter_bf16(dptr(x.contiguous()), dptr(idx), dptr(out),
+                                        M, N, topk, cur_stream()), "vh_moe_scatter")
+    return out
+
+
+def moe_gather(x: torch.Tensor, index: torch.Tensor) -> torch.Tensor:
+    M, topk = index.shape
+    N = x.shape[1]
+    idx = index.reshape(-1).to(torch.int32).contiguous()
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    check(get_lib().vh_moe_gather_bf16(dptr(x.contiguous()), dptr(idx), dptr(out),
+                                       M, N, topk, cur_stream()), "vh_moe_gather")
+    return out
+
+
+def group_gemm_nk(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
+                  trans_b: bool, c: torch.Tensor | None = None,
+                  activation: int = 0) -> torch.Tensor:
+    """C_g = A_g @ (B_g^T if trans_b else B_g); rows partitioned by cumsum."""
+    assert a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+    G = b.shape[0]
+    N = b.shape[1] if trans_b else b.shape[2]
+    K = b.shape[2] if trans_b else b.shape[1]
+    assert a.shape[1] == K, (a.shape, b.shape, trans_b)
+    accumulate = c is not None
+    if c is None:
+        c = torch.empty(a.shape[0], N, dtype=a.dtype, device=a.device)
+    cs = cumsum.to(torch.int64).contiguous()
+    check(get_lib().vh_group_gemm_nk_bf16(
+        dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, N, K,
+        a.shape[0], int(trans_b), int(accumulate), activation, cur_stream()),
+        "vh_group_gemm_nk")
+    return c
+
+
+def group_gemm_mn(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
+                  G: int) -> torch.Tensor:
+    """C[g] = A_g^T @ B_g; A [rows, M], B [rows, N] -> C [G, M, N]."""
+    assert a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+    M, N = a.shape[1], b.shape[1]
+    c = torch.empty(G, M, N, dtype=a.dtype, device=a.device)
+    cs = cumsum.to(torch.int64).contiguous()
+    check(get_lib().vh_group_gemm_mn_bf16(
+        dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, M, N,
+        cur_stream()), "vh_group_gemm_mn")
+    return c
+
+
+def silu_mul_weighted(fc1: torch.Tensor, w_row: torch.Tensor | None) -> torch.Tensor:
+    rows, twoI = fc1.shape
+    I = twoI // 2
+    out = torch.empty(rows, I, dtype=fc1.dtype, device=fc1.device)
+    has_w = w_row is not None
+    check(get_lib().vh_moe_silu_mul_weighted_bf16(
+        dptr(fc1.contiguous()), dptr(w_row.contiguous()) if has_w else None,
+        dptr(out), rows, I, int(has_w), cur_stream()), "vh_moe_silu_mul_weighted")
+    return out
+
+
+def silu_mul_weighted_bwd(dy: torch.Tensor, fc1: torch.Tensor,
+                          w_row: torch.Tensor | None):
+    rows, twoI = fc1.shape
+    I = twoI // 2
+    dfc1 = torch.empty_like(fc1)
+    has_w = w_row is not None
+    dw_row = torch.zeros(rows, dtype=torch.float32, device=fc1.device) if has_w else None
+    check(get_lib().vh_moe_silu_mul_weighted_bwd_bf16(
+        dptr(dy.contiguous()), dptr(fc1.contiguous()),
+        dptr(w_row.contiguous()) if has_w else None, dptr(dfc1),
+        dptr(dw_row) if has_w else None, rows, I, int(has_w), cur_stream()),
+        "vh_moe_silu_mul_weighted_bwd")
+    return dfc1, dw_row
+
+
+def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float):
+    T = x.numel() // x.shape[-1]
+    H = x.shape[-1]
+    y = torch.empty_like(x)
+    rstd = torch.empty(T, dtype=torch.float32, device=x.device)
+    check(get_lib().vh_rmsnorm_fwd_bf16(dptr(x.contiguous()), dptr(w.contiguous()),
+                                        dptr(y), dptr(rstd), T, H, eps,
+                                        cur_stream()), "vh_rmsnorm_fwd")
+    return y, rstd
+
+
+def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                rstd: torch.Tensor):
+    T = x.numel() // x.shape[-1]
+    H = x.shape[-1]
+    dx = torch.empty_like(x)
+    dw = torch.zeros(H, dtype=torch.float32, device=x.device)
+    check(get_lib().vh_rmsnorm_bwd_bf16(dptr(dy.contiguous()), dptr(x.contiguous()),
+                                        dptr(w.contiguous()), dptr(rstd), dptr(dx),
+                                        dptr(dw), T, H, cur_stream()), "vh_rmsnorm_bwd")
+    return dx, dw
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+         negate_sin: bool = False):
+    B, hq, S, D = q.shape
+    hk = k.shape[1]
+    qo = torch.empty_like(q)
+    ko = torch.empty_like(k)
+    check(get_lib().vh_rope_bf16(dptr(q.contiguous()), dptr(k.contiguous()),
+                                 dptr(cos.contiguous()), dptr(sin.contiguous()),
+                                 dptr(qo), dptr(ko), B, hq, hk, S, D,
+                                 int(negate_sin), cur_stream()), "vh_rope")
+    return qo, ko
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    out = torch.empty_like(gate)
+    check(get_lib().vh_silu_mul_bf16(dptr(gate.contiguous()), dptr(up.contiguous()),
+                                     dptr(out), gate.numel(), cur_stream()),
+          "vh_silu_mul")
+    return out
+
+
+def silu_mul_bwd(dy: torch.Tensor, gate: torch.Tensor, up: torch.Tensor):
+    dg = torch.empty_like(gate)
+    du = torch.empty_like(up)
+    check(get_lib().vh_silu_mul_bwd_bf16(dptr(dy.contiguous()), dptr(gate.contiguous()),
+                                         dptr(up.contiguous()), dptr(dg), dptr(du),
+                                         gate.numel(), cur_stream()), "vh_silu_mul_bwd")
+    return dg, du
+
+
+def ce_fwd(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float,
+           ignore_index: int = -100):
+    rows, V = logits.shape
+    loss_rows = torch.zeros(rows, dtype=torch.float32, device=logits.device)
+    dlogits = torch.empty_like(logits)
+    check(get_lib().vh_ce_fwd_bf16(dptr(logits.contiguous()),
+                                   dptr(labels.to(torch.int64).contiguous()),
+                                   dptr(loss_rows), dptr(dlogits), rows, V,
+                                   grad_scale, ignore_index, cur_stream()),
+          "vh_ce_fwd")
+    return loss_rows, dlogits

@@ -1,0 +1,112 @@
+"""Attention with Ulysses SP (impl name "hip").
+
+Parity target: flash_attention_forward + prepare/restore Ulysses exchange
+(ops/kernels/attention/flash.py:153-301, attention/ulysses.py:27-91):
+seq-sharded/full-heads -> (a2a) -> full-seq/head-sharded -> kernel ->
+(a2a back); GQA KV head repeat when sp > kv heads.
+
+Round-1 core kernel: torch scaled_dot_product_attention on ROCm (the flash
+path inside PyTorch) — a hand-written CDNA4 flash kernel replaces it in a
+later round. The Ulysses layout exchanges are the §8a items
+and are implemented here.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+from ...distributed.parallel_state import get_parallel_state
+from ...distributed.sequence_parallel import (
+    gather_heads_scatter_seq,
+    gather_seq_scatter_heads,
+)
+from ..kernel_registry import KERNEL_REGISTRY, HardwareRequirement, KernelSpec
+
+
+def prepare_ulysses_qkv(query, key, value, *, group, ulysses_size):
+    """[B, S/sp, h, D] -> [B, S, h/sp, D] for q/k/v (ref attention/ulysses.py:27-65)."""
+    q_heads = query.shape[2]
+    kv_heads = key.shape[2]
+    assert q_heads % ulysses_size == 0
+    if ulysses_size > kv_heads:
+        assert ulysses_size % kv_heads == 0
+        rep = ulysses_size // kv_heads
+        key = torch.repeat_interleave(key, dim=2, repeats=rep)
+        value = torch.repeat_interleave(value, dim=2, repeats=rep)
+    else:
+        assert kv_heads % ulysses_size == 0
+    if query.ndim == 4 and query.size(0) == 1:
+        q, k, v = query.squeeze(0), key.squeeze(0), value.squeeze(0)
+        q = gather_seq_scatter_heads(q, seq_dim=0, head_dim=1, group=group)
+        k = gather_seq_scatter_heads(k, seq_dim=0, head_dim=1, group=group)
+        v = gather_seq_scatter_heads(v, seq_dim=0, head_dim=1, group=group)
+        return q.unsqueeze(0), k.unsqueeze(0), v.unsqueeze(0), q_heads
+    q = gather_seq_scatter_heads(query, seq_dim=1, head_dim=2, group=group)
+    k = gather_seq_scatter_heads(key, seq_dim=1, head_dim=2, group=group)
+    v = gather_seq_scatter_heads(value, seq_dim=1, head_dim=2, group=group)
+    return q, k, v, q_heads
+
+
+def restore_ulysses_output(output, *, group):
+    """[B, S, h/sp, D] -> [B, S/sp, h, D] (ref attention/ulysses.py:83-91)."""
+    if output.ndim == 4 and output.size(0) == 1:
+        out = output.squeeze(0)
+        out = gather_heads_scatter_seq(out, seq_dim=0, head_dim=1, group=group)
+        return out.unsqueeze(0)
+    return gather_heads_scatter_seq(output, seq_dim=1, head_dim=2, group=group)
+
+
+def _repeat_kv(x, n_rep):
+    b, h, s, d = x.shape
+    if n_rep == 1:
+        return x
+    return x[:, :, None, :, :].expand(b, h, n_rep, s, d).reshape(b, h * n_rep, s, d)
+
+
+def hip_attention_forward(module, query, key, value, attention_mask,
+                          dropout=0.0, scaling=None, sliding_window=None,
+                          softcap=None, skip_ulysses=False, **kwargs):
+    """Slot signature parity: flash.py:153-165. Inputs [B, h, S, D]
+    (pre-transpose, HF convention); returns ([B, S_local, h, D] reshaped by
+    the caller, None)."""
+    ps = get_parallel_state()
+    ulysses = ps.ulysses_enabled and not skip_ulysses
+    # FA kernels take [B, S, h, D]
+    query = query.transpose(1, 2)
+    key = key.transpose(1, 2)
+    value = value.transpose(1, 2)
+    if ulysses:
+        group = ps.ulysses_group
+        query, key, value, _ = prepare_ulysses_qkv(
+            query, key, value, group=group, ulysses_size=ps.ulysses_size
+        )
+    # core attention (torch flash/sdpa path on ROCm)
+    q = query.transpose(1, 2)
+    k = key.transpose(1, 2)
+    v = value.transpose(1, 2)
+    n_rep = q.shape[1] // k.shape[1]
+    k = _repeat_kv(k, n_rep)
+    v = _repeat_kv(v, n_rep)
+    out = F.scaled_dot_product_attention(q, k, v, attn_mask=None,
+                                         dropout_p=dropout, scale=scaling,
+                                         is_causal=True)
+    out = out.transpose(1, 2)  # [B, S, h, D]
+    if ulysses:
+        out = restore_ulysses_output(out, group=ps.ulysses_group)
+    return out, None
+
+
+KERNEL_REGISTRY.register(
+    KernelSpec(
+        name="hip", op_name="attention", variant="sdpa_with_sp",
+        factory=lambda: hip_attention_forward,
+        hardware=HardwareRequirement(device_type="gpu"),
+        description="Ulysses SP exchange + torch flash/sdpa core (own FA kernel: next round)",
+    )
+)
+
+# the reference exposes the same callable under this name in HF's
+# ALL_ATTENTION_FUNCTIONS; keep the alias for drop-in callers.
+veomni_flash_attention_2_with_sp = hip_attention_forward

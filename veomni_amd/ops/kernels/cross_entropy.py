@@ -1,0 +1,91 @@
+"""HIP chunked fused-linear cross-entropy (impl name "hip").
+
+Parity targets:
+  - ForCausalLMLoss 3-tuple contract + SP handling:
+    ops/kernels/cross_entropy/__init__.py:89-221;
+  - chunked fused-linear semantics (never materializes [T,V]):
+    chunk_loss.py:43-144 — grads computed IN forward per chunk, saved,
+    scaled by upstream grad in backward;
+  - inner CE = fixed_cross_entropy: fp32 log-softmax, sum-NLL / num_items.
+
+Device split: chunk logits/dW/dx GEMMs ride hipBLASLt via torch.matmul
+(plain library GEMMs); the softmax/NLL/grad fusion is vh_ce_fwd (C ABI).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ...distributed.parallel_state import get_parallel_state
+from ...distributed.sequence_parallel import reduce_sequence_parallel_loss
+from .. import hip_lib
+from ..kernel_registry import KERNEL_REGISTRY, HardwareRequirement, KernelSpec
+
+IGNORE_INDEX = -100
+
+
+class HipChunkCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, hidden_states, weight, labels, chunk_size):
+        # hidden [B, T, H] (already causally aligned by the caller), labels [B, T]
+        B, T, H = hidden_states.shape
+        flat_h = hidden_states.reshape(-1, H)
+        flat_l = labels.reshape(-1)
+        num_items = (flat_l != IGNORE_INDEX).sum()
+        inv = 1.0 / max(int(num_items), 1)
+
+        total = torch.zeros((), dtype=torch.float32, device=hidden_states.device)
+        grad_h = torch.empty_like(flat_h)
+        grad_w = torch.zeros_like(weight, dtype=torch.float32)
+        for s in range(0, flat_h.shape[0], chunk_size):
+            e = min(s + chunk_size, flat_h.shape[0])
+            h = flat_h[s:e]
+            logits = torch.matmul(h, weight.t())          # bf16 (hipBLASLt)
+            loss_rows, dlogits = hip_lib.ce_fwd(logits, flat_l[s:e], inv, IGNORE_INDEX)
+            total += loss_rows.sum() * inv
+            grad_h[s:e] = torch.matmul(dlogits, weight)
+            grad_w += torch.matmul(dlogits.t().float(), h.float())
+        ctx.save_for_backward(grad_h, grad_w.to(weight.dtype))
+        ctx.hshape = hidden_states.shape
+        return total
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        grad_h, grad_w = ctx.saved_tensors
+        if grad_output is not None and not torch.equal(
+            grad_output, torch.ones_like(grad_output)
+        ):
+            grad_h = grad_h * grad_output
+            grad_w = grad_w * grad_output
+        return grad_h.reshape(ctx.hshape), grad_w, None, None
+
+
+def hip_causal_lm_loss(hidden_states=None, weights=None, labels=None,
+                       vocab_size=None, num_items_in_batch=None,
+                       ignore_index=IGNORE_INDEX, shift_labels=None,
+                       chunk_size: int = 2048, **kwargs):
+    """ForCausalLMLoss-shaped entry returning (loss, logits|None, aux|None).
+
+    Causal shift applied here unless SP pre-shifted the labels in the
+    collator (ref __init__.py:184-195)."""
+    assert hidden_states is not None and weights is not None
+    sp_enabled = get_parallel_state().sp_enabled
+    orig_labels = labels
+    if not sp_enabled:
+        labels = labels[..., 1:].contiguous()
+        hidden_states = hidden_states[..., :-1, :].contiguous()
+    loss = HipChunkCE.apply(hidden_states, weights, labels, chunk_size)
+    if sp_enabled:
+        num_valid = (orig_labels != ignore_index).sum()
+        loss = reduce_sequence_parallel_loss(loss, num_valid)
+    return loss, None, None
+
+
+KERNEL_REGISTRY.register(
+    KernelSpec(
+        name="hip", op_name="cross_entropy_loss", variant="causal",
+        factory=lambda: hip_causal_lm_loss,
+        hardware=HardwareRequirement(device_type="gpu"),
+        description="gfx950 chunked fused-linear CE (vh_ce_fwd + hipBLASLt chunks)",
+    )
+)
