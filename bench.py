@@ -1,0 +1,242 @@
+#!/usr/bin/env python3
+"""Training-step benchmark (driver contract — see repo prompt).
+
+Default (no flags): N=1, Llama-3-8B FSDP2 bf16 seq4096 (BASELINE configs[1],
+the largest single-GPU configuration — Qwen3-30B-A3B does not fit one GPU
+with optimizer state). --gpus N>=2: Qwen3-30B-A3B FSDP2 + EP (configs[2],
+the north-star config). A step = one forward+backward+clip+optimizer over
+one synthetic packed seq-4096 batch per rank (weak scaling).
+
+Rank 0 prints ONE JSON line with metric/value/roofline/cpu_baseline.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
+
+import torch
+import torch.distributed as dist
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def active_params(cfg):
+    """Dense-equivalent active params/token (reference count_flops.py:123-155)."""
+    H, L = cfg.hidden_size, cfg.num_hidden_layers
+    V = cfg.vocab_size
+    hq, hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    attn = H * (hq * D) + 2 * H * (hkv * D) + (hq * D) * H
+    if cfg.is_moe:
+        mlp = 3 * H * cfg.moe_intermediate_size * cfg.num_experts_per_tok
+        router = H * cfg.num_experts
+    else:
+        mlp = 3 * H * cfg.intermediate_size
+        router = 0
+    emb = V * H  # lm_head
+    return L * (attn + mlp + router) + emb
+
+
+def step_flops(cfg, tokens, seq_len):
+    """Analytic fwd+bwd FLOPs per step (factor 6·P·T + attention term,
+    reference count_flops.py:123-155,502-528; causal 1/2)."""
+    p = active_params(cfg)
+    attn = (
+        cfg.num_hidden_layers
+        * 2 * 2 * seq_len * cfg.num_attention_heads * cfg.head_dim * 0.5
+    ) * 3  # qk^T + pv, fwd+2x bwd
+    return 6.0 * p * tokens + attn * tokens
+
+
+def run_cpu_baseline(preset, seq_len, budget_s=15.0):
+    """Oracle (host-mirror, eager ops) timed on host cores on a BOUNDED
+    sample of the same workload. kind="port" (DESIGN.md §c/§d)."""
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch
+
+    bind_ops("eager")
+    sample_seq = min(seq_len, 512)
+    model = build_model(preset, dtype=torch.bfloat16, device="cpu")
+    model.use_checkpoint = False
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-5)
+    batch = synthetic_batch(model.config.vocab_size, sample_seq, seed=42, device="cpu")
+    t0 = time.time()
+    steps = 0
+    while time.time() - t0 < budget_s and steps < 8:
+        loss, _ = model(**batch)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        steps += 1
+    dt = time.time() - t0
+    toks = steps * sample_seq / dt
+    return {
+        "value": round(toks, 3),
+        "unit": "tokens/s",
+        "cores": torch.get_num_threads(),
+        "kind": "port",
+        "sample": f"{steps} steps of {preset} seq{sample_seq} bf16 eager on host ({dt:.1f}s)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", type=str, default=None)
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world, 1)
+
+    assert torch.cuda.is_available(), "bench needs an MI355X"
+    torch.cuda.set_device(local_rank)
+
+    if world > 1:
+        dist.init_process_group("nccl")
+    else:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29555")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+
+    import veomni_amd.ops  # registrations
+    from veomni_amd.ops import HIP_OPS_CONFIG, hip_lib
+    from veomni_amd.distributed.fsdp2 import build_parallelize_model
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+    from veomni_amd.models import PRESETS, build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch
+
+    preset = args.model or ("llama3-8b" if n_gpus == 1 else "qwen3-moe-30b")
+    cfg = PRESETS[preset]
+    ep_size = n_gpus if (cfg.is_moe and n_gpus > 1) else 1
+    init_parallel_state(ep_size=ep_size, device_type="cuda")
+    bind_ops(HIP_OPS_CONFIG)
+
+    log(f"building {preset} on cuda:{local_rank} (ep={ep_size}, ws={world})")
+    t_build = time.time()
+    model = build_model(preset, dtype=torch.bfloat16, device="cuda")
+    model.use_checkpoint = True
+    model = build_parallelize_model(model)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-5, betas=(0.9, 0.95), foreach=True)
+    log(f"built in {time.time() - t_build:.1f}s; mem {torch.cuda.memory_allocated()/2**30:.1f} GiB")
+
+    seq = args.seq_len
+    batch = synthetic_batch(cfg.vocab_size, seq, seed=42 + rank, device="cuda")
+
+    def one_step():
+        loss, _ = model(**batch)
+        loss.backward()
+        model.clip_grad_norm_(1.0)
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        return loss
+
+    for i in range(args.warmup):
+        loss = one_step()
+        log(f"warmup {i}: loss {float(loss):.4f}")
+
+    # profile one extra (untimed) step for the per-kernel roofline leg
+    hip_lib.profile_enable(True)
+    one_step()
+    prof = hip_lib.profile_summary()
+    hip_lib.profile_enable(False)
+
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        one_step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    dt = time.time() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([dt], device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dt = float(t)
+
+    tokens_per_step = seq * n_gpus
+    toks_per_s = tokens_per_step * args.steps / dt
+    ms_per_step = dt / args.steps * 1000.0
+
+    flops = step_flops(cfg, seq, seq) * n_gpus  # per step whole job
+    mfu = flops * args.steps / dt / (n_gpus * 2.5e15)
+
+    # roofline: dominant hand-written kernel
+    roofline = None
+    if "group_gemm_nk" in prof:
+        rec = prof["group_gemm_nk"]
+        fl = sum(rec["work"]) / rec["count"]
+        ach = fl / (rec["ms_avg"] / 1000.0) / 1e12
+        roofline = {"bound": "mfma", "achieved": round(ach, 1), "peak": 2500.0,
+                    "unit": "TFLOP/s", "frac": round(ach / 2500.0, 4), "traffic": None,
+                    "kernel": "vh_group_gemm_nk_bf16"}
+    elif "ce_fwd" in prof:
+        rec = prof["ce_fwd"]
+        by = sum(rec["work"]) / rec["count"]
+        ach = by / (rec["ms_avg"] / 1000.0) / 1e9
+        roofline = {"bound": "hbm", "achieved": round(ach, 1), "peak": 8000.0,
+                    "unit": "GB/s", "frac": round(ach / 8000.0, 4), "traffic": None,
+                    "kernel": "vh_ce_fwd_bf16"}
+
+    cpu_baseline = None
+    if rank == 0 and n_gpus == 1 and not args.no_cpu_baseline:
+        log("timing CPU baseline (host cores, bounded sample)")
+        try:
+            cpu_baseline = run_cpu_baseline(preset, seq)
+        except Exception as e:  # report, never break the bench line
+            cpu_baseline = {"value": None, "unit": "tokens/s", "cores": None,
+                            "kind": "port", "sample": f"failed: {e}"}
+        bind_ops(HIP_OPS_CONFIG)
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec/node",
+            "value": round(toks_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "mfu": round(mfu, 4),
+            "config": {
+                "workload": f"{preset} FSDP2 bf16 seq{seq}"
+                + (f" EP{ep_size}" if ep_size > 1 else ""),
+                "model": preset,
+                "global_batch": n_gpus,
+                "seq_len": seq,
+                "parallelism": f"dp{world}" + (f"_ep{ep_size}" if ep_size > 1 else ""),
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,222 @@
+"""Multi-process (gloo, world_size=2) CPU tests of the distributed plumbing:
+Ulysses a2a re-sharding, SP loss reduce, EP dispatch/combine, parallel state,
+EP parameter slicing, and FSDP2-vs-single-process grad-norm equivalence.
+
+Pattern parity: the reference tests the same logic with mp.spawn + gloo
+fallback (tests/tools/launch_utils.py:18-47) and compares grad_norm across
+parallelisms (tests/distributed/test_fsdp_equivalence.py)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from oracle import losses as oracle_losses
+from oracle import moe as oracle_moe
+
+WORLD = 2
+
+
+def _run(rank, world_size, fn, port, args):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    torch.manual_seed(0)
+    try:
+        fn(rank, world_size, *args)
+    finally:
+        from veomni_amd.distributed.parallel_state import set_parallel_state
+
+        set_parallel_state(None)
+        dist.destroy_process_group()
+
+
+def spawn(fn, *args, world_size=WORLD):
+    import random
+
+    port = random.randint(20000, 40000)
+    mp.spawn(_run, args=(world_size, fn, port, args), nprocs=world_size, join=True)
+
+
+# ---------------------------------------------------------------- ulysses a2a
+def _ulysses_roundtrip(rank, ws):
+    from veomni_amd.distributed.sequence_parallel import (
+        gather_heads_scatter_seq,
+        gather_seq_scatter_heads,
+        set_ulysses_sequence_parallel_group,
+    )
+
+    set_ulysses_sequence_parallel_group(dist.group.WORLD)
+    S, h, D = 8, 4, 6
+    full = torch.arange(ws * S * h * D, dtype=torch.float32).reshape(ws * S, h, D)
+    local = full[rank * S : (rank + 1) * S]
+    # seq-sharded/full-heads -> full-seq/head-sharded
+    g = gather_seq_scatter_heads(local, seq_dim=0, head_dim=1)
+    assert g.shape == (ws * S, h // ws, D)
+    torch.testing.assert_close(g, full[:, rank * (h // ws) : (rank + 1) * (h // ws)])
+    # and back
+    back = gather_heads_scatter_seq(g, head_dim=1, seq_dim=0)
+    torch.testing.assert_close(back, local)
+    set_ulysses_sequence_parallel_group(None)
+
+
+def test_ulysses_roundtrip():
+    spawn(_ulysses_roundtrip)
+
+
+def _sp_loss_reduce(rank, ws):
+    from veomni_amd.distributed.sequence_parallel import (
+        reduce_sequence_parallel_loss,
+        set_ulysses_sequence_parallel_group,
+    )
+
+    set_ulysses_sequence_parallel_group(dist.group.WORLD)
+    loss_vals = [2.0, 3.0]
+    n_vals = [5, 3]
+    loss = torch.tensor(loss_vals[rank], requires_grad=True)
+    n = torch.tensor(float(n_vals[rank]))
+    out = reduce_sequence_parallel_loss(loss * 1.0, n.clone())
+    expect = oracle_losses.sp_loss_reduce(loss_vals, n_vals)
+    torch.testing.assert_close(out, expect)
+    out.backward()
+    # d out / d loss_r = ws * n_r / sum(n)
+    torch.testing.assert_close(loss.grad, torch.tensor(ws * n_vals[rank] / sum(n_vals)))
+    set_ulysses_sequence_parallel_group(None)
+
+
+def test_sp_loss_reduce():
+    spawn(_sp_loss_reduce)
+
+
+# ----------------------------------------------------------------- EP dispatch
+class _TorchEpClass(torch.autograd.Function):
+    """Test stub ep_class: per-expert dense MLP on the permuted buffer, same
+    contract as EPMergedFc1HipGroupGemm but in plain torch (CPU)."""
+
+    @staticmethod
+    def forward(ctx, permute_tokens, cumsum, gate_up, down):
+        ctx.save_for_backward(permute_tokens, cumsum, gate_up, down)
+        out = torch.empty(permute_tokens.shape[0], down.shape[1], dtype=permute_tokens.dtype)
+        start = 0
+        for g in range(gate_up.shape[0]):
+            end = int(cumsum[g])
+            x = permute_tokens[start:end]
+            gu = x @ gate_up[g].t()
+            gate, up = gu.chunk(2, -1)
+            out[start:end] = (torch.nn.functional.silu(gate) * up) @ down[g].t()
+            start = end
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        permute_tokens, cumsum, gate_up, down = ctx.saved_tensors
+        with torch.enable_grad():
+            x = permute_tokens.detach().requires_grad_(True)
+            gu_w = gate_up.detach().requires_grad_(True)
+            d_w = down.detach().requires_grad_(True)
+            out = _TorchEpClass.forward(type("c", (), {"save_for_backward": lambda *a: None})(),
+                                        x, cumsum, gu_w, d_w)
+            out.backward(dy)
+        return x.grad, None, gu_w.grad, d_w.grad
+
+
+def _ep_dispatch(rank, ws):
+    from veomni_amd.distributed.moe import dispatch_to_ep_class
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+
+    init_parallel_state(ep_size=ws)
+    E, topk, T, H, I = 8, 2, 16, 32, 24
+    torch.manual_seed(5)
+    gate_up = torch.randn(E, 2 * I, H) * 0.1
+    down = torch.randn(E, H, I) * 0.1
+    # identical routing on every rank's own tokens
+    torch.manual_seed(100 + rank)
+    hidden = torch.randn(T, H)
+    sel = torch.randint(0, E, (T, topk))
+    sel[:, 1] = (sel[:, 0] + 1) % E  # distinct experts per token
+    rw = torch.softmax(torch.randn(T, topk), -1)
+
+    local_e = E // ws
+    my_gu = gate_up[rank * local_e : (rank + 1) * local_e]
+    my_down = down[rank * local_e : (rank + 1) * local_e]
+    hidden_g = hidden.clone().requires_grad_(True)
+    out = dispatch_to_ep_class(_TorchEpClass, E, rw, sel, hidden_g, my_gu, my_down)
+
+    # single-process expectation: eager per-expert loop, weights AFTER fc2 is
+    # NOT the contract here — dispatch applies weights in unpermute, matching
+    # the fused order (weights after expert MLP, before sum) — both orders
+    # coincide because the weight multiplies the fc2 output row.
+    expect = torch.zeros_like(hidden)
+    for t in range(T):
+        for kk in range(topk):
+            e = int(sel[t, kk])
+            x = hidden[t : t + 1]
+            gu = x @ gate_up[e].t()
+            gate, up = gu.chunk(2, -1)
+            y = (torch.nn.functional.silu(gate) * up) @ down[e].t()
+            expect[t] += (y * rw[t, kk]).squeeze(0)
+    torch.testing.assert_close(out, expect, rtol=1e-4, atol=1e-5)
+    out.sum().backward()
+    assert hidden_g.grad is not None and torch.isfinite(hidden_g.grad).all()
+
+
+def test_ep_dispatch_combine():
+    spawn(_ep_dispatch)
+
+
+# ------------------------------------------------- FSDP2 grad-norm equivalence
+def _fsdp_equivalence(rank, ws):
+    from veomni_amd.distributed.fsdp2 import build_parallelize_model
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch
+
+    init_parallel_state()
+    bind_ops("eager")
+    model = build_model("tiny-dense")
+    ref = build_model("tiny-dense")  # identical seeded init
+
+    model = build_parallelize_model(model, param_dtype=torch.float32,
+                                    reduce_dtype=torch.float32)
+    batch = synthetic_batch(512, 64, seed=7)  # same batch on both ranks
+    loss, _ = model(**batch)
+    loss.backward()
+    gn = model.clip_grad_norm_(1e9)
+
+    # single-process reference on the same (replicated) batch
+    rloss, _ = ref(**batch)
+    rloss.backward()
+    rgn = torch.nn.utils.get_total_norm([p.grad for p in ref.parameters() if p.grad is not None])
+    torch.testing.assert_close(loss.detach().float(), rloss.detach().float(), rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(gn.float(), rgn.float(), rtol=1e-3, atol=1e-4)
+
+
+def test_fsdp2_equivalence():
+    spawn(_fsdp_equivalence)
+
+
+# -------------------------------------------------------------- EP param slice
+def _ep_slice(rank, ws):
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+
+    init_parallel_state(ep_size=ws)
+    bind_ops("eager")
+    model = build_model("tiny-moe")
+    full = model.model.layers[0].mlp.experts.gate_up_proj.detach().clone()
+    plan = model.get_parallel_plan()
+    plan.apply(model)
+    p = model.model.layers[0].mlp.experts.gate_up_proj
+    E = full.shape[0]
+    local = E // ws
+    assert p.shape[0] == local
+    torch.testing.assert_close(p.detach(), full[rank * local : (rank + 1) * local])
+    assert getattr(p, "_ep_param", False)
+
+
+def test_ep_param_slice():
+    spawn(_ep_slice)

@@ -1,0 +1,262 @@
+"""MI355X kernel parity tests (pytest -m gpu): every HIP kernel vs the CPU
+oracle / fp32 torch references, at small sizes and with the edge cases the
+reference tests (empty experts, ragged group sizes, duplicate routes)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from oracle import moe as omoe  # noqa: E402
+from oracle import norms as onorms  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def lib():
+    from veomni_amd.ops import hip_lib
+
+    return hip_lib
+
+
+def bf(x):
+    return x.to(torch.bfloat16)
+
+
+# ------------------------------------------------------------- token bookkeeping
+def test_histogram_bitexact(lib, golden):
+    for name in ("small", "tiny", "one_expert", "full"):
+        idx = golden[f"scatter/{name}/expert_index"].cuda()
+        E = int(golden[f"scatter/{name}/num_experts"])
+        out = lib.expert_histogram(idx, E)
+        assert torch.equal(out.cpu(), golden[f"scatter/{name}/histogram"]), name
+
+
+def test_scatter_gather_vs_oracle(lib):
+    torch.manual_seed(0)
+    M, N, topk, E = 64, 128, 4, 8
+    x = bf(torch.randn(M, N)).cuda()
+    sel = torch.randint(0, E, (M, topk)).cuda()
+    _, scatter_index = omoe.compute_expert_scatter_index(sel.cpu())
+    scatter_index = scatter_index.cuda()
+    out = lib.moe_scatter(x, scatter_index)
+    ref = omoe.moe_scatter(x.cpu(), scatter_index.cpu())
+    assert torch.equal(out.cpu(), ref)  # pure row copies: bit-exact
+    g = lib.moe_gather(out, scatter_index)
+    gref = omoe.moe_gather(ref, scatter_index.cpu())
+    torch.testing.assert_close(g.cpu().float(), gref.float(), rtol=1e-2, atol=1e-2)
+
+
+# ----------------------------------------------------------------- grouped GEMM
+def _mk_groups(counts):
+    cumsum = torch.tensor(counts).cumsum(0)
+    return cumsum
+
+
+@pytest.mark.parametrize("counts,N,K", [
+    ([37, 0, 91, 128], 128, 64),       # ragged + empty group
+    ([256, 256], 256, 128),            # aligned
+    ([5, 1000, 3], 1536, 2048),        # skew at real widths
+    ([64], 768, 2048),                 # single group, real fc1 shape
+])
+def test_group_gemm_nk_transb(lib, counts, N, K):
+    torch.manual_seed(1)
+    G = len(counts)
+    rows = sum(counts)
+    a = bf(torch.randn(rows, K) * 0.5).cuda()
+    b = bf(torch.randn(G, N, K) * 0.5).cuda()
+    cumsum = _mk_groups(counts).cuda()
+    c = lib.group_gemm_nk(a, b, cumsum, trans_b=True)
+    # fp32 reference per group
+    start = 0
+    for g in range(G):
+        end = int(cumsum[g])
+        if end == start:
+            continue
+        ref = (a[start:end].float() @ b[g].float().t())
+        torch.testing.assert_close(c[start:end].float(), ref, rtol=2e-2, atol=2e-2)
+        start = end
+
+
+@pytest.mark.parametrize("counts,N,K", [
+    ([37, 0, 91, 128], 64, 128),
+    ([100, 400], 768, 1536),           # dgrad-like: B [K,N] with K=2I
+])
+def test_group_gemm_nk_notransb(lib, counts, N, K):
+    torch.manual_seed(2)
+    G = len(counts)
+    rows = sum(counts)
+    a = bf(torch.randn(rows, K) * 0.5).cuda()
+    b = bf(torch.randn(G, K, N) * 0.5).cuda()
+    cumsum = _mk_groups(counts).cuda()
+    c = lib.group_gemm_nk(a, b, cumsum, trans_b=False)
+    start = 0
+    for g in range(G):
+        end = int(cumsum[g])
+        if end == start:
+            continue
+        ref = (a[start:end].float() @ b[g].float())
+        torch.testing.assert_close(c[start:end].float(), ref, rtol=2e-2, atol=2e-2)
+        start = end
+
+
+@pytest.mark.parametrize("counts,M,N", [
+    ([37, 0, 91, 60], 128, 64),
+    ([300, 100], 1536, 2048),          # wgrad fc1 shape [G,2I,H]
+])
+def test_group_gemm_mn(lib, counts, M, N):
+    torch.manual_seed(3)
+    G = len(counts)
+    rows = sum(counts)
+    a = bf(torch.randn(rows, M) * 0.5).cuda()
+    b = bf(torch.randn(rows, N) * 0.5).cuda()
+    cumsum = _mk_groups(counts).cuda()
+    c = lib.group_gemm_mn(a, b, cumsum, G)
+    start = 0
+    for g in range(G):
+        end = int(cumsum[g])
+        if end == start:
+            assert torch.all(c[g] == 0), f"group {g} not zero-filled"
+            continue
+        ref = a[start:end].float().t() @ b[start:end].float()
+        torch.testing.assert_close(c[g].float(), ref, rtol=2e-2, atol=2e-2)
+        start = end
+
+
+def test_group_gemm_asymmetric_layout(lib):
+    """Transpose-detecting check (guide G9/G16): asymmetric B, M=N."""
+    counts = [128]
+    a = torch.zeros(128, 64)
+    b = torch.zeros(1, 128, 64)
+    for i in range(128):
+        for j in range(64):
+            b[0, i, j] = (i * 3 + j) % 7 - 3
+    a[:, 0] = 1.0  # row r of C = b[:,0] broadcast? no: C[r,n] = sum_k a[r,k] b[n,k] = b[n,0]
+    c = lib.group_gemm_nk(bf(a).cuda(), bf(b).cuda(), _mk_groups(counts).cuda(), trans_b=True)
+    ref = a.float() @ b[0].float().t()
+    torch.testing.assert_close(c.cpu().float(), ref, rtol=1e-3, atol=1e-3)
+
+
+# ------------------------------------------------------------------ norms/rope
+def test_rmsnorm_fwd_bwd(lib):
+    torch.manual_seed(4)
+    for T, H in [(64, 128), (33, 2048), (256, 4096)]:
+        x = bf(torch.randn(T, H)).cuda()
+        w = bf(torch.randn(H) * 0.1 + 1).cuda()
+        y, rstd = lib.rmsnorm_fwd(x, w, 1e-6)
+        ref = onorms.rms_norm(x.cpu(), w.cpu(), 1e-6)  # bf16 math, same order
+        torch.testing.assert_close(y.cpu().float(), ref.float(), rtol=1e-2, atol=1e-2)
+        dy = bf(torch.randn(T, H)).cuda()
+        dx, dw = lib.rmsnorm_bwd(dy, x, w, rstd)
+        rdx, rdw = onorms.rms_norm_bwd(dy.cpu().float(), x.cpu().float(), w.cpu().float(), 1e-6)
+        torch.testing.assert_close(dx.cpu().float(), rdx, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(dw.cpu(), rdw, rtol=5e-2, atol=5e-2)
+
+
+def test_rope_fwd_bwd(lib):
+    torch.manual_seed(5)
+    B, hq, hk, S, D = 2, 4, 2, 32, 64
+    q = bf(torch.randn(B, hq, S, D)).cuda()
+    k = bf(torch.randn(B, hk, S, D)).cuda()
+    cos, sin = onorms.rope_cos_sin(D, S, 10000.0)
+    cosb = bf(cos)[None].expand(B, S, D).contiguous().cuda()
+    sinb = bf(sin)[None].expand(B, S, D).contiguous().cuda()
+    qe, ke = lib.rope(q, k, cosb, sinb)
+    rq2, rk2 = onorms.apply_rotary_pos_emb(q.cpu().float(), k.cpu().float(),
+                                           cosb.cpu().float(), sinb.cpu().float(), unsqueeze_dim=1)
+    torch.testing.assert_close(qe.cpu().float(), rq2, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(ke.cpu().float(), rk2, rtol=2e-2, atol=2e-2)
+    # backward = negated sin: rope(rope(x)) with -sin returns x
+    qb, kb = lib.rope(qe, ke, cosb, sinb, negate_sin=True)
+    torch.testing.assert_close(qb.cpu().float(), q.cpu().float(), rtol=2e-2, atol=2e-2)
+
+
+def test_silu_mul(lib):
+    torch.manual_seed(6)
+    g = bf(torch.randn(64, 256)).cuda()
+    u = bf(torch.randn(64, 256)).cuda()
+    out = lib.silu_mul(g, u)
+    ref = onorms.silu_mul(g.cpu().float(), u.cpu().float())
+    torch.testing.assert_close(out.cpu().float(), ref, rtol=2e-2, atol=2e-2)
+    dy = bf(torch.randn(64, 256)).cuda()
+    dg, du = lib.silu_mul_bwd(dy, g, u)
+    gg = g.cpu().float().requires_grad_(True)
+    uu = u.cpu().float().requires_grad_(True)
+    onorms.silu_mul(gg, uu).backward(dy.cpu().float())
+    torch.testing.assert_close(dg.cpu().float(), gg.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(du.cpu().float(), uu.grad, rtol=3e-2, atol=3e-2)
+
+
+# ------------------------------------------------------------------------- CE
+def test_ce_fwd(lib):
+    torch.manual_seed(7)
+    rows, V = 33, 512
+    logits = bf(torch.randn(rows, V) * 2).cuda()
+    labels = torch.randint(0, V, (rows,)).cuda()
+    labels[5] = -100
+    num = int((labels != -100).sum())
+    loss_rows, dlogits = lib.ce_fwd(logits, labels, 1.0 / num)
+    lf = logits.cpu().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, labels.cpu(), ignore_index=-100,
+                                            reduction="sum") / num
+    ref.backward()
+    torch.testing.assert_close(loss_rows.sum().cpu() / num, ref.detach(), rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(dlogits.cpu().float(), lf.grad, rtol=5e-2, atol=1e-3)
+
+
+# ------------------------------------------------------------------ fused MoE
+def test_fused_moe_vs_eager_bf16(lib, golden):
+    """HIP fused path (weights before fc2) vs the reference's eager bf16
+    output (golden) within the documented operator-ordering tolerance
+    (ref tests/ops/test_fused_moe_split_vs_merged.py:30-38)."""
+    from veomni_amd.ops.kernels.moe import HipFusedMoeFunction
+
+    hidden = bf(golden["moe/hidden"]).cuda()
+    gup = bf(golden["moe/gate_up_proj"]).cuda()
+    down = bf(golden["moe/down_proj"]).cuda()
+    top_i = golden["moe/top_i"].cuda()
+    top_w = bf(golden["moe/top_w"]).cuda()
+    # pad dims: toy golden has H=64, I=48 -> K=64 ok, N=2I=96 %16=0 ok
+    out = HipFusedMoeFunction.apply(gup.shape[0], top_w, top_i, hidden, gup, down)
+    ref = golden["moe/out_bf16"].float()
+    torch.testing.assert_close(out.cpu().float(), ref, rtol=5e-2, atol=5e-2)
+
+
+def test_fused_moe_fwd_bwd_vs_eager_gpu(lib):
+    """HIP fused MoE (fwd+bwd) vs the eager per-expert loop run on GPU with
+    identical bf16 weights — mirrors the reference's fused-vs-eager test."""
+    from veomni_amd.models.modeling import Experts, ModelConfig
+    from veomni_amd.ops.kernels.moe import HipFusedMoeFunction
+
+    torch.manual_seed(8)
+    cfg = ModelConfig(hidden_size=128, num_experts=8, num_experts_per_tok=2,
+                      moe_intermediate_size=64, vocab_size=64)
+    T = 96
+    experts = Experts(cfg).cuda().to(torch.bfloat16)
+    with torch.no_grad():
+        experts.gate_up_proj.normal_(0, 0.05)
+        experts.down_proj.normal_(0, 0.05)
+    hidden = bf(torch.randn(T, 128) * 0.5).cuda()
+    rw = torch.softmax(torch.randn(T, 8), -1)
+    top_w, top_i = torch.topk(rw, 2, -1)
+    top_w = bf(top_w / top_w.sum(-1, keepdim=True)).cuda()
+    top_i = top_i.cuda()
+
+    h1 = hidden.clone().requires_grad_(True)
+    gup1 = experts.gate_up_proj.detach().clone().requires_grad_(True)
+    down1 = experts.down_proj.detach().clone().requires_grad_(True)
+    tw1 = top_w.clone().requires_grad_(True)
+    out_hip = HipFusedMoeFunction.apply(8, tw1, top_i, h1, gup1, down1)
+    dy = bf(torch.randn_like(out_hip) * 0.1)
+    out_hip.backward(dy)
+
+    h2 = hidden.clone().requires_grad_(True)
+    tw2 = top_w.clone().requires_grad_(True)
+    out_eager = experts(h2, top_i, tw2)
+    out_eager.backward(dy)
+
+    tol = dict(rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(out_hip.float(), out_eager.float(), **tol)
+    torch.testing.assert_close(h1.grad.float(), h2.grad.float(), **tol)
+    torch.testing.assert_close(gup1.grad.float(), experts.gate_up_proj.grad.float(), **tol)
+    torch.testing.assert_close(down1.grad.float(), experts.down_proj.grad.float(), **tol)
+    torch.testing.assert_close(tw1.grad.float(), tw2.grad.float(), rtol=1e-1, atol=1e-1)

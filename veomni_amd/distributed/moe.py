@@ -107,10 +107,11 @@ def preprocess(expert_mask: torch.Tensor, num_experts: int, ep_group) -> tuple:
         num_local_tokens_per_expert.reshape(ep_size, num_local_experts).sum(dim=1).tolist()
     )
     num_global = torch.zeros(
-        ep_size, num_local_tokens_per_expert.size(0),
+        ep_size * num_local_tokens_per_expert.size(0),
         dtype=num_local_tokens_per_expert.dtype, device=num_local_tokens_per_expert.device,
     )
-    dist.all_gather_into_tensor(num_global, num_local_tokens_per_expert, group=ep_group)
+    dist.all_gather_into_tensor(num_global, num_local_tokens_per_expert.contiguous(), group=ep_group)
+    num_global = num_global.view(ep_size, -1)
 
     start, end = rank * num_local_experts, (rank + 1) * num_local_experts
     num_global_per_local = num_global[:, start:end].contiguous()

@@ -356,18 +356,27 @@ class ForCausalLM(nn.Module):
 
     @torch.no_grad()
     def reset_parameters(self, seed: int = 1234):
-        """Seeded init (HF _init_weights pattern: normal(0, initializer_range))."""
-        g = torch.Generator().manual_seed(seed)
+        """Seeded init (HF _init_weights pattern: normal(0, initializer_range)).
+
+        Deterministic across ranks: one generator, sorted param order. On GPU
+        the generator lives on-device (fast for multi-B-param models) — every
+        rank draws the identical sequence, so replicated params agree before
+        FSDP sharding."""
         std = self.config.initializer_range
+        try:
+            dev = next(self.parameters()).device
+        except StopIteration:
+            return
+        g = torch.Generator(device=dev).manual_seed(seed)
         for name, p in sorted(self.named_parameters(), key=lambda kv: kv[0]):
-            cpu = torch.empty(p.shape, dtype=torch.float32)
             if name.endswith("layernorm.weight") or name.endswith("norm.weight") or ".q_norm" in name or ".k_norm" in name:
-                cpu.fill_(1.0)
+                p.fill_(1.0)
             elif name.endswith(".bias"):
-                cpu.zero_()
+                p.zero_()
             else:
-                cpu.normal_(0.0, std, generator=g)
-            p.copy_(cpu.to(p.dtype))
+                buf = torch.empty(p.shape, dtype=torch.float32, device=dev)
+                buf.normal_(0.0, std, generator=g)
+                p.copy_(buf.to(p.dtype))
 
     # ------------------------------------------------------------- forward
     def forward(self, input_ids, labels=None, position_ids=None, **kwargs):

@@ -67,6 +67,58 @@ def cur_stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
+# --------------------------------------------------------- event profiling
+# bench.py's roofline leg: when enabled, each wrapped kernel call records a
+# (start, end) HIP event pair on ITS launch stream plus the call's
+# algorithmic work, so average per-launch durations can be read back.
+_PROFILE = False
+_PROF_EVENTS: dict = {}
+
+
+def profile_enable(on: bool = True) -> None:
+    global _PROFILE
+    _PROFILE = on
+    if on:
+        _PROF_EVENTS.clear()
+
+
+def profile_events() -> dict:
+    return _PROF_EVENTS
+
+
+def profile_summary() -> dict:
+    """{name: {"count": n, "ms_avg": per-launch ms, "work": [payloads]}}."""
+    torch.cuda.synchronize()
+    out = {}
+    for name, recs in _PROF_EVENTS.items():
+        times = [s.elapsed_time(e) for s, e, _ in recs]
+        out[name] = {
+            "count": len(recs),
+            "ms_avg": sum(times) / max(len(times), 1),
+            "ms_total": sum(times),
+            "work": [w for _, _, w in recs],
+        }
+    return out
+
+
+class _prof:
+    def __init__(self, name: str, work=None):
+        self.name = name
+        self.work = work
+
+    def __enter__(self):
+        if _PROFILE:
+            self.s = torch.cuda.Event(enable_timing=True)
+            self.e = torch.cuda.Event(enable_timing=True)
+            self.s.record()
+        return self
+
+    def __exit__(self, *a):
+        if _PROFILE:
+            self.e.record()
+            _PROF_EVENTS.setdefault(self.name, []).append((self.s, self.e, self.work))
+
+
 def check(rc: int, what: str) -> None:
     if rc != 0:
         raise RuntimeError(f"{what} failed (rc={rc}): {get_lib().vh_last_error().decode()}")
@@ -115,14 +167,23 @@ def group_gemm_nk(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
     N = b.shape[1] if trans_b else b.shape[2]
     K = b.shape[2] if trans_b else b.shape[1]
     assert a.shape[1] == K, (a.shape, b.shape, trans_b)
+    if K % 64 != 0:
+        # zero-pad the reduction dim to the kernel's BK multiple (only test
+        # shapes hit this; every production K — H, I, 2I — is a 64-multiple)
+        pad = 64 - K % 64
+        a = torch.nn.functional.pad(a, (0, pad))
+        b = torch.nn.functional.pad(b, (0, pad) if trans_b else (0, 0, 0, pad))
+        K = K + pad
     accumulate = c is not None
     if c is None:
         c = torch.empty(a.shape[0], N, dtype=a.dtype, device=a.device)
     cs = cumsum.to(torch.int64).contiguous()
-    check(get_lib().vh_group_gemm_nk_bf16(
-        dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, N, K,
-        a.shape[0], int(trans_b), int(accumulate), activation, cur_stream()),
-        "vh_group_gemm_nk")
+    # algorithmic flops = 2 * total_rows * N * K (every A row belongs to one group)
+    with _prof("group_gemm_nk", 2.0 * a.shape[0] * N * K):
+        check(get_lib().vh_group_gemm_nk_bf16(
+            dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, N, K,
+            a.shape[0], int(trans_b), int(accumulate), activation, cur_stream()),
+            "vh_group_gemm_nk")
     return c
 
 
@@ -133,9 +194,10 @@ def group_gemm_mn(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
     M, N = a.shape[1], b.shape[1]
     c = torch.empty(G, M, N, dtype=a.dtype, device=a.device)
     cs = cumsum.to(torch.int64).contiguous()
-    check(get_lib().vh_group_gemm_mn_bf16(
-        dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, M, N,
-        cur_stream()), "vh_group_gemm_mn")
+    with _prof("group_gemm_mn", 2.0 * a.shape[0] * M * N):
+        check(get_lib().vh_group_gemm_mn_bf16(
+            dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, M, N,
+            cur_stream()), "vh_group_gemm_mn")
     return c
 
 
@@ -170,9 +232,10 @@ def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float):
     H = x.shape[-1]
     y = torch.empty_like(x)
     rstd = torch.empty(T, dtype=torch.float32, device=x.device)
-    check(get_lib().vh_rmsnorm_fwd_bf16(dptr(x.contiguous()), dptr(w.contiguous()),
-                                        dptr(y), dptr(rstd), T, H, eps,
-                                        cur_stream()), "vh_rmsnorm_fwd")
+    with _prof("rmsnorm_fwd", 2.0 * T * H * 2 + T * H * 2):
+        check(get_lib().vh_rmsnorm_fwd_bf16(dptr(x.contiguous()), dptr(w.contiguous()),
+                                            dptr(y), dptr(rstd), T, H, eps,
+                                            cur_stream()), "vh_rmsnorm_fwd")
     return y, rstd
 
 
@@ -203,9 +266,10 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
 
 def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     out = torch.empty_like(gate)
-    check(get_lib().vh_silu_mul_bf16(dptr(gate.contiguous()), dptr(up.contiguous()),
-                                     dptr(out), gate.numel(), cur_stream()),
-          "vh_silu_mul")
+    with _prof("silu_mul", 3.0 * gate.numel() * 2):
+        check(get_lib().vh_silu_mul_bf16(dptr(gate.contiguous()), dptr(up.contiguous()),
+                                         dptr(out), gate.numel(), cur_stream()),
+              "vh_silu_mul")
     return out
 
 
@@ -223,9 +287,10 @@ def ce_fwd(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float,
     rows, V = logits.shape
     loss_rows = torch.zeros(rows, dtype=torch.float32, device=logits.device)
     dlogits = torch.empty_like(logits)
-    check(get_lib().vh_ce_fwd_bf16(dptr(logits.contiguous()),
-                                   dptr(labels.to(torch.int64).contiguous()),
-                                   dptr(loss_rows), dptr(dlogits), rows, V,
-                                   grad_scale, ignore_index, cur_stream()),
-          "vh_ce_fwd")
+    with _prof("ce_fwd", 3.0 * rows * V * 2):
+        check(get_lib().vh_ce_fwd_bf16(dptr(logits.contiguous()),
+                                       dptr(labels.to(torch.int64).contiguous()),
+                                       dptr(loss_rows), dptr(dlogits), rows, V,
+                                       grad_scale, ignore_index, cur_stream()),
+              "vh_ce_fwd")
     return loss_rows, dlogits
