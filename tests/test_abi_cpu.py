@@ -33,7 +33,8 @@ def test_header_covers_wrappers():
     from veomni_amd.ops import hip_lib  # noqa: F401
 
     syms = set(header_symbols())
-    bound = set(re.findall(r'"(vh_\w+)"', open(os.path.join(
+    bound = set(re.findall(r'_sig\(lib, "(vh_\w+)"', open(os.path.join(
         REPO, "veomni_amd", "ops", "hip_lib.py")).read()))
-    missing = {b for b in bound if b not in syms and b.startswith("vh_")}
+    assert bound, "no ctypes signatures found"
+    missing = bound - syms
     assert not missing, missing
