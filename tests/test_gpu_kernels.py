@@ -149,7 +149,13 @@ def test_rmsnorm_fwd_bwd(lib):
         dx, dw = lib.rmsnorm_bwd(dy, x, w, rstd)
         rdx, rdw = onorms.rms_norm_bwd(dy.cpu().float(), x.cpu().float(), w.cpu().float(), 1e-6)
         torch.testing.assert_close(dx.cpu().float(), rdx, rtol=5e-2, atol=5e-2)
-        torch.testing.assert_close(dw.cpu(), rdw, rtol=5e-2, atol=5e-2)
+        # dw reference with the kernel's exact rounding: dw = sum dy * bf16(x*rs)
+        # (forward downcasts x_hat to bf16 BEFORE the weight multiply, so the
+        # exact autograd dw sees the rounded x_hat; the fp32 oracle does not)
+        xh = (x.cpu().float() * rstd.cpu()[:, None]).to(torch.bfloat16).float()
+        rdw_exact = (dy.cpu().float() * xh).sum(0)
+        torch.testing.assert_close(dw.cpu(), rdw_exact, rtol=1e-2, atol=1e-2)
+        torch.testing.assert_close(dw.cpu(), rdw, rtol=5e-2, atol=2e-1)
 
 
 def test_rope_fwd_bwd(lib):
