@@ -14,9 +14,12 @@
 //   * K-contiguous operands staged via global_load_lds width 16 into a
 //     lane-linear LDS image; the T2 XOR swizzle ((row&7)<<4) is applied on the
 //     per-lane SOURCE address and on the ds_read address (rule 21).
+//   * Per-lane stage addresses (incl. the reference's modulo row wrap) are
+//     hoisted OUT of the K-loop — the loop only adds k0 (int64 div/mod per
+//     stage was VALU-bound).
 //   * K-strided operands (dgrad's B, wgrad's A and B) staged through
 //     registers: per-lane column loads are coalesced ACROSS lanes (consecutive
-//     n per lane), written as 16-B ds_writes into the transposed image.
+//     out-dim per lane), written as 16-B ds_writes into the transposed image.
 //   * 2-phase double buffer: stage tile t+1, compute tile t, one
 //     __syncthreads per K-step (guide T3 "minimum 2-phase" recipe).
 
@@ -40,56 +43,73 @@ __device__ __forceinline__ int swz(int row, int colb) {
   return colb ^ ((row & 7) << 4);
 }
 
-// Stage a [128][BK] K-contiguous tile (A, or B when rows of B are
-// K-contiguous) via glds. `row_of(r)` maps tile row -> global row index.
-// LDS image: linear [row][colb], data swizzled so that
-// lds[row][colb] = src[row][colb ^ ((row&7)<<4)].
-template <typename RowFn>
-__device__ __forceinline__ void stage_kcontig(const bf16_t* src, int64_t ld_elems,
-                                              bf16_t* lds_tile, int64_t k0,
-                                              RowFn row_of, int tid) {
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    // glds writes wave-uniform LDS base + lane*16 (guide §5 caveat): pass the
-    // WAVE base; the per-lane SOURCE address carries the lane offset and the
-    // inverse swizzle (rule 21).
-    int base = i * 4096 + wave * 1024;
-    int o = base + lane * 16;  // this lane's effective dest byte offset
-    int row = o >> 7;          // 128 B per row
-    int colb = o & 127;
-    int src_colb = swz(row, colb);
-    const bf16_t* g = src + row_of(row) * ld_elems + k0 + (src_colb >> 1);
-    glds16(g, lds_tile + (base >> 1));
-  }
-}
+// Per-lane staging context for a [128][BK] K-contiguous tile via glds:
+// 4 pre-swizzled per-lane source pointers (advance by k each stage) + the
+// wave-uniform LDS byte bases.
+struct KStage {
+  const bf16_t* src[4];
+  int lds_base[4];
 
-// Stage a [128][BK] tile TRANSPOSED from a [BK][128-col window] source
-// (element (out,k) read from src[k*ld + out]). Coalesced across lanes
-// (consecutive `out` per lane); 16-B ds_writes; same swizzled image as
-// stage_kcontig. kmax clamps the k range (zero fill) for ragged group rows.
-__device__ __forceinline__ void stage_transposed(const bf16_t* src, int64_t ld_elems,
-                                                 bf16_t* lds_tile, int64_t k0,
-                                                 int64_t kmax, int out0,
-                                                 int out_max, int tid) {
-  // thread t: out = t % 128, k chunk = (t / 128) * 32 .. +32
-  int out = tid & 127;
-  int kbase = (tid >> 7) * 32;
-  int64_t gout = out0 + out;
-  bool out_ok = gout < out_max;
+  template <typename RowFn>
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       RowFn row_of, int tid) {
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {  // 4 chunks of 8 k
-    bf16x8 v;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int64_t k = k0 + kbase + c * 8 + j;
-      v.v[j] = (out_ok && k < kmax) ? src[k * ld_elems + gout] : (bf16_t)0;
+    for (int i = 0; i < 4; ++i) {
+      int base = i * 4096 + wave * 1024;
+      int o = base + lane * 16;
+      int row = o >> 7;
+      int colb = o & 127;
+      src[i] = s + row_of(row) * ld_elems + (swz(row, colb) >> 1);
+      lds_base[i] = base;
     }
-    int colb = (kbase + c * 8) * 2;
-    *reinterpret_cast<bf16x8*>(&lds_tile[(out * 128 + swz(out, colb)) >> 1]) = v;
   }
-}
+
+  __device__ __forceinline__ void stage(bf16_t* lds_tile, int64_t k0) const {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) glds16(src[i] + k0, lds_tile + (lds_base[i] >> 1));
+  }
+};
+
+// Transposed register staging: element (out, k) read from src[k*ld + out];
+// coalesced across lanes (consecutive `out` per lane), 16-B ds_writes into
+// the same swizzled image. CLAMP guards ragged k (zero fill).
+template <bool CLAMP>
+struct TStage {
+  const bf16_t* base;   // src + gout (this thread's column)
+  int64_t ld;
+  int64_t kmax;
+  bool out_ok;
+  int out, kbase;
+
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       int64_t kmax_, int out0, int out_max,
+                                       int tid) {
+    out = tid & 127;
+    kbase = (tid >> 7) * 32;
+    int64_t gout = out0 + out;
+    out_ok = gout < out_max;
+    base = s + (out_ok ? gout : 0);
+    ld = ld_elems;
+    kmax = kmax_;
+  }
+
+  __device__ __forceinline__ void stage(bf16_t* lds_tile, int64_t k0) const {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int64_t k = k0 + kbase + c * 8 + j;
+        bool ok = out_ok && (!CLAMP || k < kmax);
+        v.v[j] = ok ? base[k * ld] : (bf16_t)0;
+      }
+      int colb = (kbase + c * 8) * 2;
+      *reinterpret_cast<bf16x8*>(&lds_tile[(out * 128 + swz(out, colb)) >> 1]) = v;
+    }
+  }
+};
 
 // read one 16x32 MFMA A/B fragment from the swizzled [row][64] bf16 image
 __device__ __forceinline__ bf16frag frag_read(const bf16_t* lds_tile, int row0,
@@ -105,6 +125,20 @@ __device__ __forceinline__ float maybe_act(float x, int ACT) {
   return ACT == 1 ? siluf(x) : x;
 }
 
+// shared MFMA inner step over both K-sub-tiles of one staged K-step
+#define VH_MFMA_STEP(TA_, TB_)                                                 \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                           \
+    bf16frag af[4], bfr[4];                                                    \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                              \
+        af[i] = frag_read(TA_, wr * 64 + i * 16, ks, lane);                    \
+    _Pragma("unroll") for (int j = 0; j < 4; ++j)                              \
+        bfr[j] = frag_read(TB_, wc * 64 + j * 16, ks, lane);                   \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                              \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                          \
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(               \
+                af[i], bfr[j], acc[i][j], 0, 0, 0);                            \
+  }
+
 // ---------------------------------------------------------------- same_nk
 template <bool TRANS_B, bool ACCUM, int ACT>
 __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
@@ -112,7 +146,6 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
     bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
     int64_t N, int64_t K, int tiles_m, int tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: A0, B0, A1, B1 — each 128*64*2 = 16 KiB
   bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
   auto ta = [&](int buf) { return sm + buf * 16384; };
   auto tb = [&](int buf) { return sm + 8192 + buf * 16384; };
@@ -134,23 +167,27 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
   const bf16_t* Bg = B + (int64_t)gid * N * K;
   bf16_t* Cg = C + row_start * N;
 
-  auto a_row = [&](int r) -> int64_t {
+  KStage sa;
+  sa.init(Ag, K, [&](int r) -> int64_t {
     int64_t gm = (int64_t)bm * BM + r;
     return gm % m_size;  // wrap like the reference; stores are masked
-  };
-  auto b_row = [&](int r) -> int64_t {
-    int64_t gn = (int64_t)bn * BN + r;
-    return gn % N;
-  };
+  }, tid);
+
+  KStage sb_k;          // TRANS_B path
+  TStage<false> sb_t;   // !TRANS_B path (K % 64 == 0 asserted by the wrapper)
+  if (TRANS_B) {
+    sb_k.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN + r;
+      return gn % N;
+    }, tid);
+  } else {
+    sb_t.init(Bg, N, K, bn * BN, (int)N, tid);
+  }
 
   auto stage = [&](int buf, int64_t k0) {
-    stage_kcontig(Ag, K, ta(buf), k0, a_row, tid);
-    if (TRANS_B) {
-      stage_kcontig(Bg, K, tb(buf), k0, b_row, tid);
-    } else {
-      // B is [K, N]: transposed staging of the [k0..k0+BK) x [bn*BN ..) window
-      stage_transposed(Bg, N, tb(buf), k0, K, bn * BN, (int)N, tid);
-    }
+    sa.stage(ta(buf), k0);
+    if (TRANS_B) sb_k.stage(tb(buf), k0);
+    else sb_t.stage(tb(buf), k0);
   };
 
   f32x4 acc[4][4];
@@ -165,18 +202,10 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
   int cur = 0;
   for (int t = 0; t < nk; ++t) {
     if (t + 1 < nk) stage(cur ^ 1, (int64_t)(t + 1) * BK);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bf16frag af[4], bf[4];
-#pragma unroll
-      for (int i = 0; i < 4; ++i) af[i] = frag_read(ta(cur), wr * 64 + i * 16, ks, lane);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) bf[j] = frag_read(tb(cur), wc * 64 + j * 16, ks, lane);
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+    {
+      const bf16_t* TA = ta(cur);
+      const bf16_t* TB = tb(cur);
+      VH_MFMA_STEP(TA, TB)
     }
     __syncthreads();
     cur ^= 1;
@@ -238,27 +267,22 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_mn(
     const bf16_t* Ag = A + row_start * M;
     const bf16_t* Bg = B + row_start * N;
     const int nk = (int)((kcount + BK - 1) / BK);
-    auto stage = [&](int buf, int64_t k0) {
-      stage_transposed(Ag, M, ta(buf), k0, kcount, bm * BM, (int)M, tid);
-      stage_transposed(Bg, N, tb(buf), k0, kcount, bn * BN, (int)N, tid);
-    };
-    stage(0, 0);
+    TStage<true> sta, stb;
+    sta.init(Ag, M, kcount, bm * BM, (int)M, tid);
+    stb.init(Bg, N, kcount, bn * BN, (int)N, tid);
+    sta.stage(ta(0), 0);
+    stb.stage(tb(0), 0);
     __syncthreads();
     int cur = 0;
     for (int t = 0; t < nk; ++t) {
-      if (t + 1 < nk) stage(cur ^ 1, (int64_t)(t + 1) * BK);
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        bf16frag af[4], bf[4];
-#pragma unroll
-        for (int i = 0; i < 4; ++i) af[i] = frag_read(ta(cur), wr * 64 + i * 16, ks, lane);
-#pragma unroll
-        for (int j = 0; j < 4; ++j) bf[j] = frag_read(tb(cur), wc * 64 + j * 16, ks, lane);
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int j = 0; j < 4; ++j)
-            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+      if (t + 1 < nk) {
+        sta.stage(ta(cur ^ 1), (int64_t)(t + 1) * BK);
+        stb.stage(tb(cur ^ 1), (int64_t)(t + 1) * BK);
+      }
+      {
+        const bf16_t* TA = ta(cur);
+        const bf16_t* TB = tb(cur);
+        VH_MFMA_STEP(TA, TB)
       }
       __syncthreads();
       cur ^= 1;

@@ -12,18 +12,73 @@
 #include "vh_common.h"
 
 // --------------------------------------------------------------- RMSNorm fwd
-__global__ void k_rmsnorm_fwd(const bf16x8* __restrict__ x,
-                              const bf16x8* __restrict__ w,
-                              bf16x8* __restrict__ y, float* __restrict__ rstd,
-                              int64_t T, int64_t Hv, float eps, float invH) {
+// Block-per-row (256 threads): one or two bf16x8 chunks per thread stay in
+// registers across the reduce, so the second pass re-reads nothing from HBM.
+// Wave shuffle + 4-slot LDS cross-wave reduce.
+__global__ __launch_bounds__(256) void k_rmsnorm_fwd(
+    const bf16x8* __restrict__ x, const bf16x8* __restrict__ w,
+    bf16x8* __restrict__ y, float* __restrict__ rstd, int64_t T, int64_t Hv,
+    float eps, float invH) {
+  __shared__ float red[4];
+  constexpr int MAXC = 4;  // up to 4 chunks/thread = H <= 8192
+  int wave = threadIdx.x / kWave;
+  int lane = threadIdx.x & (kWave - 1);
+  for (int64_t r = blockIdx.x; r < T; r += gridDim.x) {
+    const bf16x8* xr = x + r * Hv;
+    bf16x8 v[MAXC];
+    int nc = 0;
+    float ss = 0.f;
+    for (int64_t c = threadIdx.x; c < Hv; c += blockDim.x) {
+      bf16x8 vv = xr[c];
+      if (nc < MAXC) v[nc] = vv;
+      ++nc;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(vv.v[j]);
+        ss += f * f;
+      }
+    }
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1) ss += __shfl_down(ss, off, kWave);
+    if (lane == 0) red[wave] = ss;
+    __syncthreads();
+    float tot = red[0] + red[1] + red[2] + red[3];
+    float rs = rsqrtf(tot * invH + eps);
+    if (threadIdx.x == 0 && rstd != nullptr) rstd[r] = rs;
+    bf16x8* yr = y + r * Hv;
+    int ci = 0;
+    for (int64_t c = threadIdx.x; c < Hv; c += blockDim.x) {
+      bf16x8 vv = (ci < MAXC) ? v[ci] : xr[c];
+      ++ci;
+      bf16x8 wv = w[c];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        // downcast the normalized value to bf16 FIRST, then bf16 multiply
+        float nh = bf2f(f2bf(bf2f(vv.v[j]) * rs));
+        o.v[j] = f2bf(bf2f(wv.v[j]) * nh);
+      }
+      yr[c] = o;
+    }
+    __syncthreads();
+  }
+}
+
+// wave-per-row variant for small rows (per-head q/k norms: H = head_dim)
+__global__ void k_rmsnorm_fwd_small(const bf16x8* __restrict__ x,
+                                    const bf16x8* __restrict__ w,
+                                    bf16x8* __restrict__ y,
+                                    float* __restrict__ rstd, int64_t T,
+                                    int64_t Hv, float eps, float invH) {
   int wave = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
   int lane = threadIdx.x & (kWave - 1);
   int num_waves = (gridDim.x * blockDim.x) / kWave;
   for (int64_t r = wave; r < T; r += num_waves) {
     const bf16x8* xr = x + r * Hv;
+    bf16x8 v = {};
     float ss = 0.f;
-    for (int64_t c = lane; c < Hv; c += kWave) {
-      bf16x8 v = xr[c];
+    if (lane < Hv) {
+      v = xr[lane];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float f = bf2f(v.v[j]);
@@ -35,18 +90,15 @@ __global__ void k_rmsnorm_fwd(const bf16x8* __restrict__ x,
     ss = __shfl(ss, 0, kWave);
     float rs = rsqrtf(ss * invH + eps);
     if (lane == 0 && rstd != nullptr) rstd[r] = rs;
-    bf16x8* yr = y + r * Hv;
-    for (int64_t c = lane; c < Hv; c += kWave) {
-      bf16x8 v = xr[c];
-      bf16x8 wv = w[c];
+    if (lane < Hv) {
+      bf16x8 wv = w[lane];
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        // downcast the normalized value to bf16 FIRST, then bf16 multiply
         float nh = bf2f(f2bf(bf2f(v.v[j]) * rs));
         o.v[j] = f2bf(bf2f(wv.v[j]) * nh);
       }
-      yr[c] = o;
+      y[r * Hv + lane] = o;
     }
   }
 }
@@ -56,15 +108,25 @@ extern "C" int vh_rmsnorm_fwd_bf16(const uint16_t* x, const uint16_t* w,
                                    int64_t H, float eps, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(H % 8 == 0, "H %% 8 != 0 (H=%lld)", (long long)H);
+  VH_CHECK(H <= 65536, "H too large");
   int64_t Hv = H / 8;
-  int blocks = (int)((T + 3) / 4);
-  if (blocks > 2048) blocks = 2048;
-  if (blocks < 1) blocks = 1;
-  hipLaunchKernelGGL(k_rmsnorm_fwd, dim3(blocks), dim3(256), 0, s,
-                     reinterpret_cast<const bf16x8*>(x),
-                     reinterpret_cast<const bf16x8*>(w),
-                     reinterpret_cast<bf16x8*>(y), rstd, T, Hv, eps,
-                     1.0f / (float)H);
+  if (Hv <= kWave) {
+    int blocks = (int)((T + 3) / 4);
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_rmsnorm_fwd_small, dim3(blocks), dim3(256), 0, s,
+                       reinterpret_cast<const bf16x8*>(x),
+                       reinterpret_cast<const bf16x8*>(w),
+                       reinterpret_cast<bf16x8*>(y), rstd, T, Hv, eps,
+                       1.0f / (float)H);
+  } else {
+    int blocks = (int)(T < 2048 ? (T ? T : 1) : 2048);
+    hipLaunchKernelGGL(k_rmsnorm_fwd, dim3(blocks), dim3(256), 0, s,
+                       reinterpret_cast<const bf16x8*>(x),
+                       reinterpret_cast<const bf16x8*>(w),
+                       reinterpret_cast<bf16x8*>(y), rstd, T, Hv, eps,
+                       1.0f / (float)H);
+  }
   VH_HIP(hipGetLastError());
   return 0;
 }

@@ -112,21 +112,22 @@ def build_parallelize_model(
 # ---------------------------------------------------------------- grad norm
 def _local_pth_sum(params, norm_type: float) -> torch.Tensor:
     """Sum of |g|^p over the LOCAL shards, fp32 accumulation without
-    materializing fp32 grads (ref fsdp2/clip_grad_norm.py:246-270)."""
-    dev = None
-    total = None
+    materializing fp32 grads (ref fsdp2/clip_grad_norm.py:246-270).
+    Uses foreach-norm (one fused kernel sweep) instead of a per-param
+    abs/pow/sum kernel triple."""
+    grads = []
     for p in params:
         g = p.grad
         if g is None:
             continue
         if isinstance(g, DTensor):
             g = g.to_local()
-        s = g.float().abs().pow(norm_type).sum()
-        total = s if total is None else total + s
-        dev = g.device
-    if total is None:
-        total = torch.zeros((), device=dev or "cpu")
-    return total
+        if g.numel() > 0:
+            grads.append(g)
+    if not grads:
+        return torch.zeros(())
+    norms = torch._foreach_norm(grads, norm_type)
+    return torch.stack([n.float() for n in norms]).pow(norm_type).sum()
 
 
 @torch.no_grad()
@@ -156,9 +157,13 @@ def clip_grad_norm(model: nn.Module, max_norm: float, norm_type: float = 2.0,
 
     clip_coef = max_norm / (total_norm + 1e-6)
     clip_coef = torch.clamp(clip_coef, max=1.0)
+    grads = []
     for p in params:
         g = p.grad
         if isinstance(g, DTensor):
             g = g.to_local()
-        g.mul_(clip_coef.to(g.device, g.dtype))
+        if g.numel() > 0:
+            grads.append(g)
+    if grads:
+        torch._foreach_mul_(grads, clip_coef.to(grads[0].device))
     return total_norm

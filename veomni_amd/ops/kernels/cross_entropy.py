@@ -44,7 +44,8 @@ class HipChunkCE(torch.autograd.Function):
             loss_rows, dlogits = hip_lib.ce_fwd(logits, flat_l[s:e], inv, IGNORE_INDEX)
             total += loss_rows.sum() * inv
             grad_h[s:e] = torch.matmul(dlogits, weight)
-            grad_w += torch.matmul(dlogits.t().float(), h.float())
+            # bf16 GEMM per chunk, fp32 accumulation across chunks
+            grad_w += torch.matmul(dlogits.t(), h)
         ctx.save_for_backward(grad_h, grad_w.to(weight.dtype))
         ctx.hshape = hidden_states.shape
         return total
