@@ -38,9 +38,14 @@ __device__ __forceinline__ void glds16(const bf16_t* g, bf16_t* l) {
       (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
 }
 
-// byte-offset XOR swizzle within a 128-B row (8 x 16-B slots)
+// byte-offset XOR swizzle within a 128-B row (8 x 16-B slots).
+// ds_read_b128 lane groups read 16 CONSECUTIVE tile rows at one col range;
+// the bank row is 256 B, so tile rows alternate half-banks (+0 / +32 dwords).
+// slot = (row>>1)&7 gives the 8 even rows (and the 8 odd rows) 8 DISTINCT
+// slots each -> all 16 (slot, half-bank) pairs distinct -> conflict-free
+// reads (a plain (row&7) slot map leaves rows r and r+8 2-way conflicted).
 __device__ __forceinline__ int swz(int row, int colb) {
-  return colb ^ ((row & 7) << 4);
+  return colb ^ (((row >> 1) & 7) << 4);
 }
 
 // Per-lane staging context for a [128][BK] K-contiguous tile via glds:
