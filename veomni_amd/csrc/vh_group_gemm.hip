@@ -41,11 +41,14 @@ __device__ __forceinline__ void glds16(const bf16_t* g, bf16_t* l) {
 // byte-offset XOR swizzle within a 128-B row (8 x 16-B slots).
 // ds_read_b128 lane groups read 16 CONSECUTIVE tile rows at one col range;
 // the bank row is 256 B, so tile rows alternate half-banks (+0 / +32 dwords).
-// slot = (row>>1)&7 gives the 8 even rows (and the 8 odd rows) 8 DISTINCT
-// slots each -> all 16 (slot, half-bank) pairs distinct -> conflict-free
-// reads (a plain (row&7) slot map leaves rows r and r+8 2-way conflicted).
+// slot = ((row>>1) ^ (row>>3)) & 7:
+//   * reads: over 16 consecutive rows each parity class hits 8 DISTINCT
+//     slots -> all 16 (slot, half-bank) pairs distinct -> conflict-free;
+//   * transposed-staging writes (ds_write_b64, rows 8m+j at one k-col):
+//     slots over m = 0..7 are the distinct set m ^ 4(m&1) -> only the m+8
+//     repeat collides -> 2-way (vs 8-way with a (row>>1)-only slot map).
 __device__ __forceinline__ int swz(int row, int colb) {
-  return colb ^ (((row >> 1) & 7) << 4);
+  return colb ^ ((((row >> 1) ^ (row >> 3)) & 7) << 4);
 }
 
 // Per-lane staging context for a [128][BK] K-contiguous tile via glds:
@@ -77,41 +80,58 @@ struct KStage {
   }
 };
 
-// Transposed register staging: element (out, k) read from src[k*ld + out];
-// coalesced across lanes (consecutive `out` per lane), 16-B ds_writes into
-// the same swizzled image. CLAMP guards ragged k (zero fill).
+// Transposed register staging: element (out, k) read from src[k*ld + out].
+// Thread t owns a [4k][8out] sub-block: o0 = (t&15)*8, ks0 = (t>>4)*4.
+// Loads: 4 x 16-B bf16x8 (8 consecutive `out` at one k) — lanes 0..15 read
+// 256 contiguous bytes (fully coalesced). Writes: 8 x ds_write_b64 (4
+// consecutive k at one out — k IS contiguous in the [out][k] image).
+// CLAMP guards ragged k (zero fill); ragged out takes a per-element path.
 template <bool CLAMP>
 struct TStage {
-  const bf16_t* base;   // src + gout (this thread's column)
+  const bf16_t* base;   // src + out0 + o0 (this thread's out chunk)
   int64_t ld;
   int64_t kmax;
-  bool out_ok;
-  int out, kbase;
+  bool out_ok;          // whole 8-out chunk in range
+  int64_t gout_left;    // number of valid outs in this chunk (ragged tiles)
+  int o0, ks0;
 
   __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
                                        int64_t kmax_, int out0, int out_max,
                                        int tid) {
-    out = tid & 127;
-    kbase = (tid >> 7) * 32;
-    int64_t gout = out0 + out;
-    out_ok = gout < out_max;
-    base = s + (out_ok ? gout : 0);
+    o0 = (tid & 15) * 8;
+    ks0 = (tid >> 4) * 4;
+    int64_t gout = (int64_t)out0 + o0;
+    gout_left = out_max - gout;  // may be <= 0
+    out_ok = gout_left >= 8;
+    base = s + gout;
     ld = ld_elems;
     kmax = kmax_;
   }
 
   __device__ __forceinline__ void stage(bf16_t* lds_tile, int64_t k0) const {
+    bf16x8 v[4];
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 v;
+    for (int i = 0; i < 4; ++i) {
+      int64_t k = k0 + ks0 + i;
+      bool kok = !CLAMP || k < kmax;
+      if (kok && out_ok) {
+        v[i] = *reinterpret_cast<const bf16x8*>(base + k * ld);
+      } else if (kok && gout_left > 0) {
+        const bf16_t* p = base + k * ld;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int64_t k = k0 + kbase + c * 8 + j;
-        bool ok = out_ok && (!CLAMP || k < kmax);
-        v.v[j] = ok ? base[k * ld] : (bf16_t)0;
+        for (int j = 0; j < 8; ++j) v[i].v[j] = (j < gout_left) ? p[j] : bf16_t(0);
+      } else {
+        v[i] = bf16x8{};
       }
-      int colb = (kbase + c * 8) * 2;
-      *reinterpret_cast<bf16x8*>(&lds_tile[(out * 128 + swz(out, colb)) >> 1]) = v;
+    }
+    const int colb = ks0 * 2;  // 8-B aligned; swz flips only bits 4-6
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = o0 + j;
+      ushort2 lo = {v[0].v[j], v[1].v[j]};
+      ushort2 hi = {v[2].v[j], v[3].v[j]};
+      uint2 pack = {*reinterpret_cast<uint32_t*>(&lo), *reinterpret_cast<uint32_t*>(&hi)};
+      *reinterpret_cast<uint2*>(&lds_tile[(row * 128 + swz(row, colb)) >> 1]) = pack;
     }
   }
 };
