@@ -108,8 +108,9 @@ struct TStage {
     kmax = kmax_;
   }
 
-  __device__ __forceinline__ void stage(bf16_t* lds_tile, int64_t k0) const {
-    bf16x8 v[4];
+  // T14 split (guide G15): issue the global loads EARLY (before the MFMA
+  // phase that covers their latency), write to LDS LATE (after it).
+  __device__ __forceinline__ void load(bf16x8 (&v)[4], int64_t k0) const {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int64_t k = k0 + ks0 + i;
@@ -124,15 +125,25 @@ struct TStage {
         v[i] = bf16x8{};
       }
     }
+  }
+
+  __device__ __forceinline__ void flush(bf16_t* lds_tile, const bf16x8 (&v)[4]) const {
     const int colb = ks0 * 2;  // 8-B aligned; swz flips only bits 4-6
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int row = o0 + j;
       ushort2 lo = {v[0].v[j], v[1].v[j]};
       ushort2 hi = {v[2].v[j], v[3].v[j]};
-      uint2 pack = {*reinterpret_cast<uint32_t*>(&lo), *reinterpret_cast<uint32_t*>(&hi)};
+      uint2 pack = {*reinterpret_cast<const uint32_t*>(&lo),
+                    *reinterpret_cast<const uint32_t*>(&hi)};
       *reinterpret_cast<uint2*>(&lds_tile[(row * 128 + swz(row, colb)) >> 1]) = pack;
     }
+  }
+
+  __device__ __forceinline__ void stage(bf16_t* lds_tile, int64_t k0) const {
+    bf16x8 v[4];
+    load(v, k0);
+    flush(lds_tile, v);
   }
 };
 
@@ -209,12 +220,6 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
     sb_t.init(Bg, N, K, bn * BN, (int)N, tid);
   }
 
-  auto stage = [&](int buf, int64_t k0) {
-    sa.stage(ta(buf), k0);
-    if (TRANS_B) sb_k.stage(tb(buf), k0);
-    else sb_t.stage(tb(buf), k0);
-  };
-
   f32x4 acc[4][4];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
@@ -222,16 +227,26 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_nk(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int nk = (int)(K / BK);
-  stage(0, 0);
+  sa.stage(ta(0), 0);
+  if (TRANS_B) sb_k.stage(tb(0), 0);
+  else sb_t.stage(tb(0), 0);
   __syncthreads();
   int cur = 0;
   for (int t = 0; t < nk; ++t) {
-    if (t + 1 < nk) stage(cur ^ 1, (int64_t)(t + 1) * BK);
+    bf16x8 vb[4];
+    const bool more = t + 1 < nk;
+    const int64_t k1 = (int64_t)(t + 1) * BK;
+    if (more) {
+      sa.stage(ta(cur ^ 1), k1);           // glds: async fire-and-forget
+      if (TRANS_B) sb_k.stage(tb(cur ^ 1), k1);
+      else sb_t.load(vb, k1);              // T14: issue loads before MFMA
+    }
     {
       const bf16_t* TA = ta(cur);
       const bf16_t* TB = tb(cur);
       VH_MFMA_STEP(TA, TB)
     }
+    if (more && !TRANS_B) sb_t.flush(tb(cur ^ 1), vb);  // write late
     __syncthreads();
     cur ^= 1;
   }
@@ -300,14 +315,21 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_mn(
     __syncthreads();
     int cur = 0;
     for (int t = 0; t < nk; ++t) {
-      if (t + 1 < nk) {
-        sta.stage(ta(cur ^ 1), (int64_t)(t + 1) * BK);
-        stb.stage(tb(cur ^ 1), (int64_t)(t + 1) * BK);
+      bf16x8 va[4], vb[4];
+      const bool more = t + 1 < nk;
+      const int64_t k1 = (int64_t)(t + 1) * BK;
+      if (more) {  // T14 split: loads before the MFMA phase, writes after
+        sta.load(va, k1);
+        stb.load(vb, k1);
       }
       {
         const bf16_t* TA = ta(cur);
         const bf16_t* TB = tb(cur);
         VH_MFMA_STEP(TA, TB)
+      }
+      if (more) {
+        sta.flush(ta(cur ^ 1), va);
+        stb.flush(tb(cur ^ 1), vb);
       }
       __syncthreads();
       cur ^= 1;
