@@ -63,6 +63,13 @@ int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                           int64_t total_rows, int trans_b, int accumulate,
                           int activation, void* stream);
 
+/* Deep-pipelined 256x256 variant of vh_group_gemm_nk_bf16 (same semantics,
+ * no accumulate/activation); auto-dispatched for large shapes, exported for
+ * direct benchmarking. K % 32 == 0, K >= 64. */
+int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                           const int64_t* cumsum, int G, int64_t N, int64_t K,
+                           int64_t total_rows, int trans_b, void* stream);
+
 /* Per-group wgrad: C[g] = A_g^T @ B_g with per-group row count (k) from
  * cumsum; A: [rows, M], B: [rows, N], C: [G, M, N] bf16 (fp32 accum).
  * Groups with zero rows are zero-filled.

@@ -266,3 +266,25 @@ def test_fused_moe_fwd_bwd_vs_eager_gpu(lib):
     torch.testing.assert_close(gup1.grad.float(), experts.gate_up_proj.grad.float(), **tol)
     torch.testing.assert_close(down1.grad.float(), experts.down_proj.grad.float(), **tol)
     torch.testing.assert_close(tw1.grad.float(), tw2.grad.float(), rtol=1e-1, atol=1e-1)
+
+
+def test_group_gemm_nk8_ragged(lib):
+    """Exercises the 256x256 8-phase kernel (auto-dispatch at large rows),
+    K%32 (not 64), empty group, ragged tails, both B layouts."""
+    torch.manual_seed(12)
+    counts = [500, 0, 279, 333]
+    G, rows = len(counts), sum(counts)
+    cumsum = torch.tensor(counts).cumsum(0).cuda()
+    for trans_b, N, K in [(True, 288, 96), (False, 288, 96), (True, 256, 2048)]:
+        a = bf(torch.randn(rows, K) * 0.5).cuda()
+        b = bf(torch.randn(G, N, K) * 0.5).cuda() if trans_b else bf(torch.randn(G, K, N) * 0.5).cuda()
+        c = lib.group_gemm_nk(a, b, cumsum, trans_b=trans_b)
+        start = 0
+        for g in range(G):
+            end = int(cumsum[g])
+            if end == start:
+                continue
+            ref = a[start:end].float() @ (b[g].float().t() if trans_b else b[g].float())
+            torch.testing.assert_close(c[start:end].float(), ref, rtol=3e-2, atol=3e-2,
+                                       msg=lambda m: f"g={g} trans_b={trans_b} N={N} K={K}: {m}")
+            start = end

@@ -18,6 +18,7 @@ SOURCES = [
     "vh_abi.cpp",
     "vh_moe_ops.hip",
     "vh_group_gemm.hip",
+    "vh_group_gemm8.hip",
     "vh_norms.hip",
     "vh_ce.hip",
 ]

@@ -352,12 +352,25 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_mn(
 
 }  // namespace
 
+extern "C" int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B,
+                                      uint16_t* C, const int64_t* cumsum,
+                                      int G, int64_t N, int64_t K,
+                                      int64_t total_rows, int trans_b,
+                                      void* stream);
+
 extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
                                      uint16_t* C, const int64_t* cumsum, int G,
                                      int64_t N, int64_t K, int64_t total_rows,
                                      int trans_b, int accumulate,
                                      int activation, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  // deep-pipelined 256x256 variant for the large shapes (no fused epilogue /
+  // accumulate there; those callers stay on the 128x128 kernel)
+  if (!accumulate && activation == 0 && K % 32 == 0 && K >= 64 && N >= 256 &&
+      total_rows >= 16 * G * 16) {
+    return vh_group_gemm_nk8_bf16(A, B, C, cumsum, G, N, K, total_rows,
+                                  trans_b, stream);
+  }
   VH_CHECK(K % BK == 0, "K %% 64 != 0 (K=%lld)", (long long)K);
   VH_CHECK(N % 16 == 0, "N %% 16 != 0 (N=%lld)", (long long)N);
   VH_CHECK(G >= 1, "G < 1");

@@ -1,0 +1,291 @@
+// 256x256 8-phase grouped GEMM for gfx950 (bf16, fp32 accumulate).
+//
+// The deep-pipelined variant of vh_group_gemm.hip's same-NK kernel, built on
+// the guide's 256-square 8-phase recipe (cdna_hip_programming.md §5 "The 256²
+// 8-phase template" + T3/T4/T5): counted vmcnt (loads stay in flight across
+// barriers), raw s_barrier (never __syncthreads — it would drain the LDS-DMA
+// queue), s_setprio around each MFMA cluster.
+//
+// Geometry:
+//   tile 256x256, K consumed in 32-deep k-subs; 512 threads = 8 waves in a
+//   2(M) x 4(N) grid, 128x64 per wave = 8x4 16x16 fragments (128 acc VGPRs).
+//   LDS = 4-deep ring of k-sub slots x {A, B}, slot = [256 rows][32 k] bf16
+//   (16 KiB) -> 128 KiB total, 1 block/CU, 2 waves/SIMD.
+//   Per k-sub: 2 phases x {8 ds_read_b128 + 16 MFMA + raw barrier}.
+//   Staging: 4 glds (A 2 + B 2) issued at phase 0 of each k-sub for slot
+//   s+3; vmcnt(12) then admits slot s while 3 slots stay in flight.
+//
+// Dispatched by vh_group_gemm_nk_bf16 (vh_group_gemm.hip) for large shapes.
+
+#include "vh_common.h"
+
+namespace {
+
+constexpr int BM8 = 256, BN8 = 256, KSUB = 32;
+constexpr int THREADS8 = 512;
+
+using bf16frag = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ void glds16(const bf16_t* g, bf16_t* l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
+}
+
+// swizzle for the [256][32] k-sub slot (row = 64 B): 4 16-B slots per row;
+// bank row = 256 B = 4 tile rows, so rows r, r+4, r+8, r+12 share a bank
+// phase -> slot = (row>>2)&3 makes each mod-4 class conflict-free over the
+// 16 consecutive rows a ds_read_b128 lane group touches.
+__device__ __forceinline__ int swz32(int row, int colb) {
+  return colb ^ (((row >> 2) & 3) << 4);
+}
+
+__device__ __forceinline__ void raw_barrier() {
+  asm volatile("" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+#define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+
+// K-contiguous staging of one [256][32] slot via glds (2 instructions of
+// 8 KiB; per-lane source carries the inverse swizzle).
+struct KStage8 {
+  const bf16_t* src[2];
+  int lds_base[2];
+
+  template <typename RowFn>
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       RowFn row_of, int tid) {
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int base = i * 8192 + wave * 1024;
+      int o = base + lane * 16;
+      int row = o >> 6;           // 64 B per row
+      int colb = o & 63;
+      src[i] = s + row_of(row) * ld_elems + (swz32(row, colb) >> 1);
+      lds_base[i] = base;
+    }
+  }
+
+  __device__ __forceinline__ void stage(bf16_t* slot, int64_t k0) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) glds16(src[i] + k0, slot + (lds_base[i] >> 1));
+  }
+};
+
+// Transposed staging of one [256 out][32 k] slot from a [k][out] source.
+// Thread t: o0 = (t&31)*8, kp = (t>>5)*2 -> 2 x 16-B loads (8 consecutive
+// outs, k and k+1; 32 lanes x 16 B = 512 B coalesced), 8 x ds_write_b32
+// (the (k, k+1) pair is contiguous in the [out][k] image).
+struct TStage8 {
+  const bf16_t* base;
+  int64_t ld;
+  bool out_ok;
+  int64_t gout_left;
+  int o0, kp;
+
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       int out0, int out_max, int tid) {
+    o0 = (tid & 31) * 8;
+    kp = (tid >> 5) * 2;
+    int64_t gout = (int64_t)out0 + o0;
+    gout_left = out_max - gout;
+    out_ok = gout_left >= 8;
+    base = s + gout;
+    ld = ld_elems;
+  }
+
+  __device__ __forceinline__ void stage(bf16_t* slot, int64_t k0) const {
+    bf16x8 v[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int64_t k = k0 + kp + i;
+      if (out_ok) {
+        v[i] = *reinterpret_cast<const bf16x8*>(base + k * ld);
+      } else if (gout_left > 0) {
+        const bf16_t* p = base + k * ld;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[i].v[j] = (j < gout_left) ? p[j] : bf16_t(0);
+      } else {
+        v[i] = bf16x8{};
+      }
+    }
+    const int colb = kp * 2;  // 4-B aligned pair position
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = o0 + j;
+      ushort2 pk = {v[0].v[j], v[1].v[j]};
+      *reinterpret_cast<uint32_t*>(&slot[(row * 64 + swz32(row, colb & ~15) + (colb & 15)) >> 1]) =
+          *reinterpret_cast<uint32_t*>(&pk);
+    }
+  }
+};
+
+__device__ __forceinline__ bf16frag frag_read8(const bf16_t* slot, int row0,
+                                               int lane) {
+  int row = row0 + (lane & 15);
+  int colb = ((lane >> 4) << 3) * 2;  // k position within the 32-k slot
+  int off_b = row * 64 + swz32(row, colb);
+  return *reinterpret_cast<const bf16frag*>(
+      reinterpret_cast<const char*>(slot) + off_b);
+}
+
+template <bool TRANS_B>
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk8(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto slotA = [&](int s) { return sm + (s & 3) * 8192; };          // 4 x 16 KiB
+  auto slotB = [&](int s) { return sm + 32768 + (s & 3) * 8192; };  // 4 x 16 KiB
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t m_size = cumsum[gid] - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+  if ((int64_t)bm * BM8 >= m_size) return;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;  // 2x4 wave grid; 128x64 per wave
+
+  const bf16_t* Ag = A + row_start * K;
+  const bf16_t* Bg = B + (int64_t)gid * N * K;
+  bf16_t* Cg = C + row_start * N;
+
+  KStage8 sa;
+  sa.init(Ag, K, [&](int r) -> int64_t {
+    int64_t gm = (int64_t)bm * BM8 + r;
+    return gm % m_size;
+  }, tid);
+  KStage8 sb_k;
+  TStage8 sb_t;
+  if (TRANS_B) {
+    sb_k.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+  } else {
+    sb_t.init(Bg, N, bn * BN8, (int)N, tid);
+  }
+
+  auto stage = [&](int s, int64_t k0) {
+    sa.stage(slotA(s), k0);
+    if (TRANS_B) sb_k.stage(slotB(s), k0);
+    else sb_t.stage(slotB(s), k0);
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nsub = (int)(K / KSUB);
+  // prologue: slots 0..2 in flight
+  stage(0, 0);
+  stage(1, KSUB);
+  if (nsub > 2) stage(2, 2 * KSUB);
+  for (int s = 0; s < nsub; ++s) {
+    // phase 0: issue next staging, admit slot s, compute fragment half 0
+    if (s + 3 < nsub) stage(s + 3, (int64_t)(s + 3) * KSUB);
+    // counted wait: slot s must have landed; younger slots stay in flight.
+    // glds per in-flight slot: 4 (A+B) on the TRANS_B path, 2 (A only — B is
+    // synchronous ds_writes) otherwise. Tail k-subs have fewer in flight.
+    {
+      const int rem = nsub - s;  // slots still live: s .. min(s+3, nsub-1)
+      if (TRANS_B) {
+        if (rem >= 4) VMCNT(12);
+        else if (rem == 3) VMCNT(8);
+        else if (rem == 2) VMCNT(4);
+        else VMCNT(0);
+      } else {
+        if (rem >= 4) VMCNT(6);
+        else if (rem == 3) VMCNT(4);
+        else if (rem == 2) VMCNT(2);
+        else VMCNT(0);
+      }
+    }
+    raw_barrier();
+    {
+      const bf16_t* TA = slotA(s);
+      const bf16_t* TB = slotB(s);
+      bf16frag af[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(TA, wr * 128 + i * 16, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(TB, wc * 64 + j * 16, lane);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      raw_barrier();
+      // phase 1: fragment half 1 (rows 64..127 of this wave's panel)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(TA, wr * 128 + 64 + i * 16, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(TB, wc * 64 + j * 16, lane);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    raw_barrier();
+  }
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                    row_base_in + rr;
+        int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+        if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+}  // namespace
+
+extern "C" int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B,
+                                      uint16_t* C, const int64_t* cumsum,
+                                      int G, int64_t N, int64_t K,
+                                      int64_t total_rows, int trans_b,
+                                      void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % KSUB == 0 && K / KSUB >= 2, "K must be a multiple of 32, >= 64");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  int tiles_m = (int)((total_rows + BM8 - 1) / BM8);
+  if (tiles_m < 1) tiles_m = 1;
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  dim3 grid(tiles_m * tiles_n, G);
+  if (trans_b)
+    hipLaunchKernelGGL((k_group_gemm_nk8<true>), grid, dim3(THREADS8), 131072,
+                       s, reinterpret_cast<const bf16_t*>(A),
+                       reinterpret_cast<const bf16_t*>(B),
+                       reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_m,
+                       tiles_n);
+  else
+    hipLaunchKernelGGL((k_group_gemm_nk8<false>), grid, dim3(THREADS8), 131072,
+                       s, reinterpret_cast<const bf16_t*>(A),
+                       reinterpret_cast<const bf16_t*>(B),
+                       reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_m,
+                       tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
