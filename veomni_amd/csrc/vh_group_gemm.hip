@@ -364,10 +364,11 @@ extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
                                      int trans_b, int accumulate,
                                      int activation, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  // deep-pipelined 256x256 variant for the large shapes (no fused epilogue /
-  // accumulate there; those callers stay on the 128x128 kernel)
-  if (!accumulate && activation == 0 && K % 32 == 0 && K >= 64 && N >= 256 &&
-      total_rows >= 16 * G * 16) {
+  // deep-pipelined 256x256 variant — currently wins only on the wide
+  // !trans_b (dgrad) shapes (measured: profiles/r01_groupgemm_microbench);
+  // everything else stays on the 128x128 2-phase kernel.
+  if (!accumulate && activation == 0 && !trans_b && K % 32 == 0 && K >= 1024 &&
+      N >= 1024 && total_rows >= 16 * G * 16) {
     return vh_group_gemm_nk8_bf16(A, B, C, cumsum, G, N, K, total_rows,
                                   trans_b, stream);
   }

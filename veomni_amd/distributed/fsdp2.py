@@ -55,6 +55,12 @@ def build_parallelize_model(
     if plan is not None and ps.ep_enabled:
         plan.apply(model)
 
+    if ps.fsdp_size == 1 and not ps.ep_enabled:
+        # degenerate world: fully_shard would only add allgather/copy churn
+        # (no actual sharding); keep the plain module + the same clip entry.
+        model.clip_grad_norm_ = functools.partial(clip_grad_norm, model)
+        return model
+
     mp_policy = MixedPrecisionPolicy(param_dtype=param_dtype, reduce_dtype=reduce_dtype)
     fsdp_kwargs = dict(
         mesh=ps.fsdp_mesh,
