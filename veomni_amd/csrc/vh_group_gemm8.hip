@@ -289,3 +289,301 @@ extern "C" int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B,
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ============================================================================
+// Register-staged 256x256 ring kernels (no glds — every load is
+// compiler-visible, so hipcc emits counted waits and the T14 split pipelines
+// cleanly; mixing glds with ordinary loads makes the compiler drain vmcnt(0)
+// at every ds_write — the §5 "Three .s-level traps" (b) case measured as 61%
+// SQ_WAIT_ANY on the earlier dgrad variant).
+//
+//   k_group_gemm_dgrad8: C_g = A_g @ B_g, A rows K-contiguous (wrapped),
+//       B [K,N] transposed-staged. Replaces the !trans_b nk path.
+//   k_group_gemm_mn8:    C[g] = A_g^T @ B_g, both operands transposed-staged,
+//       per-group ragged K (row count).
+//
+// Shared ring: 4 k-sub slots x {A, B} (16 KiB each, 128 KiB LDS, 1 block/CU,
+// 8 waves). Per k-sub: load slot s+2's registers, two 16-MFMA phases on slot
+// s, flush slot s+2's ds_writes, one barrier.
+// ============================================================================
+
+namespace {
+
+// K-contiguous register staging of a [256 row][32 k] slot with row wrap.
+struct KRegStage8 {
+  const bf16_t* src[2];
+  int lds_off[2];  // bf16-element offsets
+
+  template <typename RowFn>
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       RowFn row_of, int tid) {
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int unit = tid + u * 512;          // 1024 units of [1 row][8 k]
+      int row = unit & 255;
+      int ko = (unit >> 8) * 8;          // 0,8,16,24
+      src[u] = s + row_of(row) * ld_elems + ko;
+      lds_off[u] = (row * 64 + swz32(row, ko * 2)) >> 1;
+    }
+  }
+
+  __device__ __forceinline__ void load(bf16x8 (&v)[2], int64_t k0) const {
+    v[0] = *reinterpret_cast<const bf16x8*>(src[0] + k0);
+    v[1] = *reinterpret_cast<const bf16x8*>(src[1] + k0);
+  }
+
+  __device__ __forceinline__ void flush(bf16_t* slot, const bf16x8 (&v)[2]) const {
+    *reinterpret_cast<bf16x8*>(&slot[lds_off[0]]) = v[0];
+    *reinterpret_cast<bf16x8*>(&slot[lds_off[1]]) = v[1];
+  }
+};
+
+// Transposed register staging with split load/flush and optional k clamp.
+template <bool CLAMP>
+struct TRegStage8 {
+  const bf16_t* base;
+  int64_t ld, kmax;
+  bool out_ok;
+  int64_t gout_left;
+  int o0, kp;
+
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       int64_t kmax_, int out0, int out_max,
+                                       int tid) {
+    o0 = (tid & 31) * 8;
+    kp = (tid >> 5) * 2;
+    int64_t gout = (int64_t)out0 + o0;
+    gout_left = out_max - gout;
+    out_ok = gout_left >= 8;
+    base = s + gout;
+    ld = ld_elems;
+    kmax = kmax_;
+  }
+
+  __device__ __forceinline__ void load(bf16x8 (&v)[2], int64_t k0) const {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int64_t k = k0 + kp + i;
+      bool kok = !CLAMP || k < kmax;
+      if (kok && out_ok) {
+        v[i] = *reinterpret_cast<const bf16x8*>(base + k * ld);
+      } else if (kok && gout_left > 0) {
+        const bf16_t* p = base + k * ld;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[i].v[j] = (j < gout_left) ? p[j] : bf16_t(0);
+      } else {
+        v[i] = bf16x8{};
+      }
+    }
+  }
+
+  __device__ __forceinline__ void flush(bf16_t* slot, const bf16x8 (&v)[2]) const {
+    const int colb = kp * 2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = o0 + j;
+      ushort2 pk = {v[0].v[j], v[1].v[j]};
+      *reinterpret_cast<uint32_t*>(&slot[(row * 64 + swz32(row, colb & ~15) + (colb & 15)) >> 1]) =
+          *reinterpret_cast<uint32_t*>(&pk);
+    }
+  }
+};
+
+#define VH_MFMA_PHASE8(TA_, TB_, ROWOFF, ACCOFF)                               \
+  {                                                                            \
+    bf16frag af[4], bfr[4];                                                    \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                              \
+        af[i] = frag_read8(TA_, wr * 128 + (ROWOFF) + i * 16, lane);           \
+    _Pragma("unroll") for (int j = 0; j < 4; ++j)                              \
+        bfr[j] = frag_read8(TB_, wc * 64 + j * 16, lane);                      \
+    __builtin_amdgcn_s_setprio(1);                                             \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                              \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                          \
+            acc[i + (ACCOFF)][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+                af[i], bfr[j], acc[i + (ACCOFF)][j], 0, 0, 0);                 \
+    __builtin_amdgcn_s_setprio(0);                                             \
+  }
+
+// ring skeleton shared by both kernels
+#define VH_RING_LOOP(SA, SB, NSUB)                                             \
+  {                                                                            \
+    bf16x8 va0[2], vb0[2];                                                     \
+    SA.load(va0, 0);                                                           \
+    SB.load(vb0, 0);                                                           \
+    SA.flush(slotA(0), va0);                                                   \
+    SB.flush(slotB(0), vb0);                                                   \
+    if ((NSUB) > 1) {                                                          \
+      SA.load(va0, KSUB);                                                      \
+      SB.load(vb0, KSUB);                                                      \
+      SA.flush(slotA(1), va0);                                                 \
+      SB.flush(slotB(1), vb0);                                                 \
+    }                                                                          \
+    __syncthreads();                                                           \
+    for (int s = 0; s < (NSUB); ++s) {                                         \
+      bf16x8 va[2], vb[2];                                                     \
+      const bool more = s + 2 < (NSUB);                                        \
+      if (more) {                                                              \
+        SA.load(va, (int64_t)(s + 2) * KSUB);                                  \
+        SB.load(vb, (int64_t)(s + 2) * KSUB);                                  \
+      }                                                                        \
+      {                                                                        \
+        const bf16_t* TA = slotA(s);                                           \
+        const bf16_t* TB = slotB(s);                                           \
+        VH_MFMA_PHASE8(TA, TB, 0, 0)                                           \
+        VH_MFMA_PHASE8(TA, TB, 64, 4)                                          \
+      }                                                                        \
+      if (more) {                                                              \
+        SA.flush(slotA(s + 2), va);                                            \
+        SB.flush(slotB(s + 2), vb);                                            \
+      }                                                                        \
+      __syncthreads();                                                         \
+    }                                                                          \
+  }
+
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_dgrad8(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto slotA = [&](int s) { return sm + (s & 3) * 8192; };
+  auto slotB = [&](int s) { return sm + 32768 + (s & 3) * 8192; };
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t m_size = cumsum[gid] - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+  if ((int64_t)bm * BM8 >= m_size) return;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const bf16_t* Ag = A + row_start * K;
+  const bf16_t* Bg = B + (int64_t)gid * K * N;  // [K, N]
+  bf16_t* Cg = C + row_start * N;
+
+  KRegStage8 sa;
+  sa.init(Ag, K, [&](int r) -> int64_t {
+    int64_t gm = (int64_t)bm * BM8 + r;
+    return gm % m_size;
+  }, tid);
+  TRegStage8<false> sb;
+  sb.init(Bg, N, K, bn * BN8, (int)N, tid);
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nsub = (int)(K / KSUB);
+  VH_RING_LOOP(sa, sb, nsub)
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                    row_base_in + rr;
+        int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+        if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_mn8(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t M, int64_t N, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto slotA = [&](int s) { return sm + (s & 3) * 8192; };
+  auto slotB = [&](int s) { return sm + 32768 + (s & 3) * 8192; };
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t kcount = cumsum[gid] - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  bf16_t* Cg = C + (int64_t)gid * M * N;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  if (kcount > 0) {
+    const bf16_t* Ag = A + row_start * M;
+    const bf16_t* Bg = B + row_start * N;
+    TRegStage8<true> sa, sb;
+    sa.init(Ag, M, kcount, bm * BM8, (int)M, tid);
+    sb.init(Bg, N, kcount, bn * BN8, (int)N, tid);
+    const int nsub = (int)((kcount + KSUB - 1) / KSUB);
+    VH_RING_LOOP(sa, sb, nsub)
+  }
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                    row_base_in + rr;
+        int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+        if (m < M && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+}  // namespace
+
+extern "C" int vh_group_gemm_dgrad8_bf16(const uint16_t* A, const uint16_t* B,
+                                         uint16_t* C, const int64_t* cumsum,
+                                         int G, int64_t N, int64_t K,
+                                         int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % KSUB == 0, "K %% 32 != 0");
+  int tiles_m = (int)((total_rows + BM8 - 1) / BM8);
+  if (tiles_m < 1) tiles_m = 1;
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  dim3 grid(tiles_m * tiles_n, G);
+  hipLaunchKernelGGL(k_group_gemm_dgrad8, grid, dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_m,
+                     tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+extern "C" int vh_group_gemm_mn8_bf16(const uint16_t* A, const uint16_t* B,
+                                      uint16_t* C, const int64_t* cumsum,
+                                      int G, int64_t M, int64_t N,
+                                      void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  int tiles_m = (int)((M + BM8 - 1) / BM8);
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  dim3 grid(tiles_m * tiles_n, G);
+  hipLaunchKernelGGL(k_group_gemm_mn8, grid, dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, M, N, tiles_m,
+                     tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
