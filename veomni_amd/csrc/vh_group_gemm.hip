@@ -372,13 +372,13 @@ extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
                                      int trans_b, int accumulate,
                                      int activation, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  // register-staged 256x256 ring kernel for the dgrad (!trans_b) shapes —
-  // every load is compiler-visible so the counted waits pipeline cleanly;
-  // trans_b fwd stays on the glds 128x128 2-phase kernel (measured faster).
-  if (!accumulate && activation == 0 && !trans_b && K % 32 == 0 && K >= 64 &&
-      N >= 256 && total_rows >= 16 * G * 16) {
-    return vh_group_gemm_dgrad8_bf16(A, B, C, cumsum, G, N, K, total_rows,
-                                     stream);
+  // Per-shape dispatch from measured data (profiles/r01_groupgemm_microbench):
+  // wide !trans_b (fc1 dgrad) -> 256x256 glds nk8; everything else -> the
+  // 128x128 2-phase kernel (fastest on fwd and narrow dgrad).
+  if (!accumulate && activation == 0 && !trans_b && K % 32 == 0 && K >= 1024 &&
+      N >= 1024 && total_rows >= 16 * G * 16) {
+    return vh_group_gemm_nk8_bf16(A, B, C, cumsum, G, N, K, total_rows, 0,
+                                  stream);
   }
   VH_CHECK(K % BK == 0, "K %% 64 != 0 (K=%lld)", (long long)K);
   VH_CHECK(N % 16 == 0, "N %% 16 != 0 (N=%lld)", (long long)N);
@@ -415,7 +415,7 @@ extern "C" int vh_group_gemm_mn_bf16(const uint16_t* A, const uint16_t* B,
                                      int64_t M, int64_t N, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(M % 16 == 0 && N % 16 == 0, "M/N %% 16 != 0");
-  if (M >= 512 && N >= 512) {
+  if (M >= 1024 && N >= 1024) {  // measured: mn8 wins only at wide M,N
     return vh_group_gemm_mn8_bf16(A, B, C, cumsum, G, M, N, stream);
   }
   int tiles_m = (int)((M + BM - 1) / BM);
