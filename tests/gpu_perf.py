@@ -75,6 +75,27 @@ def main():
     t = timeit(lambda: L.group_gemm_mn(g2, act, cumsum, E))
     print(f"mn fc2 wgrad(M{H} N{I} k{rows} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
 
+    # nk256 variant (A/B vs the 128^2 fwd above)
+    fl = 2.0 * rows * 2 * I * H
+    lib2 = L.get_lib()
+    c256 = torch.empty(rows, 2 * I, dtype=torch.bfloat16, device=dev)
+    def run256():
+        rc = lib2.vh_group_gemm_nk256_bf16(a.data_ptr(), w1.data_ptr(), c256.data_ptr(),
+                                           cumsum.data_ptr(), E, 2 * I, H, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    t = timeit(run256)
+    ref = L.group_gemm_nk(a, w1, cumsum, trans_b=True)
+    ok = torch.allclose(c256.float(), ref.float(), rtol=2e-2, atol=2e-2)
+    print(f"nk256 fc1 fwd (M{rows} N{2*I} K{H} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s  correct={ok}")
+    fl = 2.0 * rows * H * I
+    c256b = torch.empty(rows, H, dtype=torch.bfloat16, device=dev)
+    def run256b():
+        rc = lib2.vh_group_gemm_nk256_bf16(act.data_ptr(), w2.data_ptr(), c256b.data_ptr(),
+                                           cumsum.data_ptr(), E, H, I, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    t = timeit(run256b)
+    print(f"nk256 fc2 fwd (M{rows} N{H} K{I} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
+
     # hipBLASLt comparison: one dense bf16 GEMM of the fc1-fwd size
     bd = w1.reshape(E * 2 * I, H)
     t = timeit(lambda: torch.matmul(a, bd.t()))

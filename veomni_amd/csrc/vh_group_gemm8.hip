@@ -587,3 +587,170 @@ extern "C" int vh_group_gemm_mn8_bf16(const uint16_t* A, const uint16_t* B,
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ============================================================================
+// 256x256 x BK=64 double-buffered glds kernel ("nk256"): the 128x128 2-phase
+// structure scaled to a 256-square tile — 2x the MFMA work per staged byte
+// and per barrier drain, with the proven conflict-free 128-B-row swizzle.
+// 8 waves (2x4), per wave 128x64; LDS = 2 x (A 32K + B 32K) = 128 KiB.
+// ============================================================================
+
+namespace {
+
+constexpr int BK64 = 64;
+
+// conflict-free slot map for 128-B rows (see vh_group_gemm.hip swz comment)
+__device__ __forceinline__ int swz64(int row, int colb) {
+  return colb ^ ((((row >> 1) ^ (row >> 3)) & 7) << 4);
+}
+
+struct KStage64 {  // [256][64] tile via 4 glds of 8 KiB (512 threads)
+  const bf16_t* src[4];
+  int lds_base[4];
+
+  template <typename RowFn>
+  __device__ __forceinline__ void init(const bf16_t* s, int64_t ld_elems,
+                                       RowFn row_of, int tid) {
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int base = i * 8192 + wave * 1024;
+      int o = base + lane * 16;
+      int row = o >> 7;  // 128 B per row
+      int colb = o & 127;
+      src[i] = s + row_of(row) * ld_elems + (swz64(row, colb) >> 1);
+      lds_base[i] = base;
+    }
+  }
+
+  __device__ __forceinline__ void stage(bf16_t* tile, int64_t k0) const {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) glds16(src[i] + k0, tile + (lds_base[i] >> 1));
+  }
+};
+
+__device__ __forceinline__ bf16frag frag_read64(const bf16_t* tile, int row0,
+                                                int ks, int lane) {
+  int row = row0 + (lane & 15);
+  int colb = (ks * 32 + ((lane >> 4) << 3)) * 2;
+  int off_b = row * 128 + swz64(row, colb);
+  return *reinterpret_cast<const bf16frag*>(
+      reinterpret_cast<const char*>(tile) + off_b);
+}
+
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 32768; };          // 2 x 32 KiB
+  auto tb = [&](int buf) { return sm + 16384 + buf * 32768; };  // 2 x 32 KiB
+
+  const int gid = blockIdx.y;
+  const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+  const int64_t m_size = cumsum[gid] - row_start;
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+  if ((int64_t)bm * BM8 >= m_size) return;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const bf16_t* Ag = A + row_start * K;
+  const bf16_t* Bg = B + (int64_t)gid * N * K;
+  bf16_t* Cg = C + row_start * N;
+
+  KStage64 sa, sb;
+  sa.init(Ag, K, [&](int r) -> int64_t {
+    int64_t gm = (int64_t)bm * BM8 + r;
+    return gm % m_size;
+  }, tid);
+  sb.init(Bg, K, [&](int r) -> int64_t {
+    int64_t gn = (int64_t)bn * BN8 + r;
+    return gn % N;
+  }, tid);
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = (int)(K / BK64);
+  sa.stage(ta(0), 0);
+  sb.stage(tb(0), 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) {
+      sa.stage(ta(cur ^ 1), (int64_t)(t + 1) * BK64);
+      sb.stage(tb(cur ^ 1), (int64_t)(t + 1) * BK64);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16frag af[4], bfr[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bfr[j] = frag_read64(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + i * 16, ks, lane);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + 64 + i * 16, ks, lane);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                    row_base_in + rr;
+        int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+        if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+}  // namespace
+
+extern "C" int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B,
+                                        uint16_t* C, const int64_t* cumsum,
+                                        int G, int64_t N, int64_t K,
+                                        int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % BK64 == 0, "K %% 64 != 0");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  int tiles_m = (int)((total_rows + BM8 - 1) / BM8);
+  if (tiles_m < 1) tiles_m = 1;
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  dim3 grid(tiles_m * tiles_n, G);
+  hipLaunchKernelGGL(k_group_gemm_nk256, grid, dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_m,
+                     tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

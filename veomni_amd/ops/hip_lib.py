@@ -50,6 +50,8 @@ def get_lib() -> ctypes.CDLL:
          c_i64, c_int, c_int, c_int, c_p)
     _sig(lib, "vh_group_gemm_nk8_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64,
          c_i64, c_int, c_p)
+    _sig(lib, "vh_group_gemm_nk256_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
+         c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_mn_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
     _sig(lib, "vh_moe_silu_mul_weighted_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
     _sig(lib, "vh_moe_silu_mul_weighted_bwd_bf16", c_p, c_p, c_p, c_p, c_p,
