@@ -352,6 +352,10 @@ __global__ __launch_bounds__(THREADS, 2) void k_group_gemm_mn(
 
 }  // namespace
 
+extern "C" int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B,
+                                        uint16_t* C, const int64_t* cumsum,
+                                        int G, int64_t N, int64_t K,
+                                        int64_t total_rows, void* stream);
 extern "C" int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B,
                                       uint16_t* C, const int64_t* cumsum,
                                       int G, int64_t N, int64_t K,
@@ -373,8 +377,14 @@ extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
                                      int activation, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   // Per-shape dispatch from measured data (profiles/r01_groupgemm_microbench):
-  // wide !trans_b (fc1 dgrad) -> 256x256 glds nk8; everything else -> the
-  // 128x128 2-phase kernel (fastest on fwd and narrow dgrad).
+  //   trans_b large   -> nk256 (256-sq double-buffered glds; fastest fwd)
+  //   !trans_b wide   -> nk8   (glds A + transposed B ring)
+  //   everything else -> the 128x128 2-phase kernel
+  if (!accumulate && activation == 0 && trans_b && K % 64 == 0 && N >= 256 &&
+      total_rows >= (int64_t)G * 256) {
+    return vh_group_gemm_nk256_bf16(A, B, C, cumsum, G, N, K, total_rows,
+                                    stream);
+  }
   if (!accumulate && activation == 0 && !trans_b && K % 32 == 0 && K >= 1024 &&
       N >= 1024 && total_rows >= 16 * G * 16) {
     return vh_group_gemm_nk8_bf16(A, B, C, cumsum, G, N, K, total_rows, 0,

@@ -73,8 +73,11 @@ class HipFusedMoeFunction(torch.autograd.Function):
         grad_output = grad_output.view(-1, grad_output.shape[-1]).contiguous()
 
         grad_fc2_out = hip_lib.moe_scatter(grad_output, scatter_index)
-        # dgrad through fc2
-        d_weighted = hip_lib.group_gemm_nk(grad_fc2_out, fc2_weight, cumsum_t, trans_b=False)
+        # dgrad through fc2 — transpose the (small, persistent) expert weights
+        # once so dgrad runs the fast trans_b kernel (HBM copy ~50 us vs
+        # ~2x slower K-strided dgrad staging; profiles/r01_groupgemm_microbench)
+        d_weighted = hip_lib.group_gemm_nk(
+            grad_fc2_out, fc2_weight.transpose(1, 2).contiguous(), cumsum_t, trans_b=True)
         # wgrad fc2: [G, H, I]
         d_fc2_w = hip_lib.group_gemm_mn(grad_fc2_out, weighted, cumsum_t, G)
 
@@ -83,8 +86,9 @@ class HipFusedMoeFunction(torch.autograd.Function):
         grad_gate = dw_rows[scatter_index.flatten().to(torch.int64)]
         grad_gate = grad_gate.reshape(gate_weights.shape).to(gate_weights.dtype)
 
-        # dgrad + wgrad through merged fc1
-        d_scatter = hip_lib.group_gemm_nk(d_fc1, fc1_1_2_weight, cumsum_t, trans_b=False)
+        # dgrad + wgrad through merged fc1 (same weight-transpose trick)
+        d_scatter = hip_lib.group_gemm_nk(
+            d_fc1, fc1_1_2_weight.transpose(1, 2).contiguous(), cumsum_t, trans_b=True)
         d_fc1_w = hip_lib.group_gemm_mn(d_fc1, scatter_output, cumsum_t, G)
 
         grad_hidden = hip_lib.moe_gather(d_scatter, scatter_index)
@@ -108,10 +112,12 @@ class EPMergedFc1HipGroupGemm(torch.autograd.Function):
         permute_tokens, cumsum, fc1_1_2_weight, fc2_weight, fc1, act = ctx.saved_tensors
         G = fc2_weight.shape[0]
         grad_output = grad_output.contiguous()
-        d_act = hip_lib.group_gemm_nk(grad_output, fc2_weight, cumsum, trans_b=False)
+        d_act = hip_lib.group_gemm_nk(
+            grad_output, fc2_weight.transpose(1, 2).contiguous(), cumsum, trans_b=True)
         d_fc2_w = hip_lib.group_gemm_mn(grad_output, act, cumsum, G)
         d_fc1, _ = hip_lib.silu_mul_weighted_bwd(d_act, fc1, None)
-        d_tokens = hip_lib.group_gemm_nk(d_fc1, fc1_1_2_weight, cumsum, trans_b=False)
+        d_tokens = hip_lib.group_gemm_nk(
+            d_fc1, fc1_1_2_weight.transpose(1, 2).contiguous(), cumsum, trans_b=True)
         d_fc1_w = hip_lib.group_gemm_mn(d_fc1, permute_tokens, cumsum, G)
         return d_tokens, None, d_fc1_w, d_fc2_w
 
