@@ -70,6 +70,22 @@ int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                            const int64_t* cumsum, int G, int64_t N, int64_t K,
                            int64_t total_rows, int trans_b, void* stream);
 
+/* 256-square double-buffered glds variant (trans_b semantics; K % 64 == 0):
+ * the fastest forward kernel; auto-dispatched for large trans_b shapes. */
+int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                             const int64_t* cumsum, int G, int64_t N,
+                             int64_t K, int64_t total_rows, void* stream);
+
+/* Register-staged 256-square ring variants (auto-dispatched):
+ * dgrad8 = !trans_b semantics; mn8 = wgrad (A^T B) semantics. */
+int vh_group_gemm_dgrad8_bf16(const uint16_t* A, const uint16_t* B,
+                              uint16_t* C, const int64_t* cumsum, int G,
+                              int64_t N, int64_t K, int64_t total_rows,
+                              void* stream);
+int vh_group_gemm_mn8_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                           const int64_t* cumsum, int G, int64_t M, int64_t N,
+                           void* stream);
+
 /* Per-group wgrad: C[g] = A_g^T @ B_g with per-group row count (k) from
  * cumsum; A: [rows, M], B: [rows, N], C: [G, M, N] bf16 (fp32 accum).
  * Groups with zero rows are zero-filled.
