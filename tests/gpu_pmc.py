@@ -29,21 +29,18 @@ def main():
     lib = L.get_lib()
     c1 = torch.empty(rows, 2 * I, dtype=torch.bfloat16, device=dev)
     c2 = torch.empty(rows, H, dtype=torch.bfloat16, device=dev)
+    cw = torch.empty(E, 2 * I, H, dtype=torch.bfloat16, device=dev)
     s = L.cur_stream()
     for _ in range(5):
-        # fc1 fwd on the 128x128 2-phase kernel
-        lib.vh_group_gemm_nk_bf16(a.data_ptr(), w1.data_ptr(), c1.data_ptr(),
-                                  cumsum.data_ptr(), E, 2 * I, H, rows, 1, 0, 0, s)
-        # fc1 fwd on the 256x256 8-phase kernel
-        lib.vh_group_gemm_nk8_bf16(a.data_ptr(), w1.data_ptr(), c1.data_ptr(),
-                                   cumsum.data_ptr(), E, 2 * I, H, rows, 1, s)
-        # fc1 dgrad on the 8-phase kernel (!trans_b)
+        # fwd on nk256 (current dispatch target)
+        lib.vh_group_gemm_nk256_bf16(a.data_ptr(), w1.data_ptr(), c1.data_ptr(),
+                                     cumsum.data_ptr(), E, 2 * I, H, rows, s)
+        # fc1 dgrad on nk8 (current dispatch target)
         lib.vh_group_gemm_nk8_bf16(g1.data_ptr(), w1.data_ptr(), c2.data_ptr(),
                                    cumsum.data_ptr(), E, H, 2 * I, rows, 0, s)
-        # wgrad
-        lib.vh_group_gemm_mn_bf16(g1.data_ptr(), a.data_ptr(),
-                                  torch.empty(E, 2 * I, H, dtype=torch.bfloat16, device=dev).data_ptr(),
-                                  cumsum.data_ptr(), E, 2 * I, H, s)
+        # wgrad on mn8 (current dispatch target)
+        lib.vh_group_gemm_mn8_bf16(g1.data_ptr(), a.data_ptr(), cw.data_ptr(),
+                                   cumsum.data_ptr(), E, 2 * I, H, s)
     torch.cuda.synchronize()
     print("pmc probe done")
 
