@@ -404,7 +404,10 @@ struct TRegStage8 {
     __builtin_amdgcn_s_setprio(0);                                             \
   }
 
-// ring skeleton shared by both kernels
+// ring skeleton shared by both kernels. Two pending register sets: slot
+// s+3's loads issue at k-sub s and flush at the END of k-sub s+1 — the
+// load->flush gap spans TWO 32-MFMA blocks (~2k cycles) so HBM/L3 latency
+// is fully covered.
 #define VH_RING_LOOP(SA, SB, NSUB)                                             \
   {                                                                            \
     bf16x8 va0[2], vb0[2];                                                     \
@@ -418,13 +421,17 @@ struct TRegStage8 {
       SA.flush(slotA(1), va0);                                                 \
       SB.flush(slotB(1), vb0);                                                 \
     }                                                                          \
+    bf16x8 vap[2], vbp[2], van[2], vbn[2];                                     \
+    if ((NSUB) > 2) {                                                          \
+      SA.load(vap, 2 * KSUB);                                                  \
+      SB.load(vbp, 2 * KSUB);                                                  \
+    }                                                                          \
     __syncthreads();                                                           \
     for (int s = 0; s < (NSUB); ++s) {                                         \
-      bf16x8 va[2], vb[2];                                                     \
-      const bool more = s + 2 < (NSUB);                                        \
+      const bool more = s + 3 < (NSUB);                                        \
       if (more) {                                                              \
-        SA.load(va, (int64_t)(s + 2) * KSUB);                                  \
-        SB.load(vb, (int64_t)(s + 2) * KSUB);                                  \
+        SA.load(van, (int64_t)(s + 3) * KSUB);                                 \
+        SB.load(vbn, (int64_t)(s + 3) * KSUB);                                 \
       }                                                                        \
       {                                                                        \
         const bf16_t* TA = slotA(s);                                           \
@@ -432,9 +439,13 @@ struct TRegStage8 {
         VH_MFMA_PHASE8(TA, TB, 0, 0)                                           \
         VH_MFMA_PHASE8(TA, TB, 64, 4)                                          \
       }                                                                        \
-      if (more) {                                                              \
-        SA.flush(slotA(s + 2), va);                                            \
-        SB.flush(slotB(s + 2), vb);                                            \
+      if (s + 2 < (NSUB)) {                                                    \
+        SA.flush(slotA(s + 2), vap);                                           \
+        SB.flush(slotB(s + 2), vbp);                                           \
+      }                                                                        \
+      _Pragma("unroll") for (int q = 0; q < 2; ++q) {                          \
+        vap[q] = van[q];                                                       \
+        vbp[q] = vbn[q];                                                       \
       }                                                                        \
       __syncthreads();                                                         \
     }                                                                          \
