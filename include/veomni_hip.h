@@ -95,6 +95,19 @@ int vh_group_gemm_mn_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                           const int64_t* cumsum, int G, int64_t M, int64_t N,
                           void* stream);
 
+/* Transpose-pad: [rows, C] (rows grouped by cumsum) -> [C, padded_total]
+ * with each group's rows zero-padded to a 64-multiple (padded_cumsum,
+ * host-computed ceil64 cumsum). Regions past the last group are untouched. */
+int vh_transpose_pad_bf16(const uint16_t* src, uint16_t* dst,
+                          const int64_t* cumsum, const int64_t* padded_cumsum,
+                          int G, int64_t C, int64_t padded_total, void* stream);
+
+/* wgrad over transpose-padded operands: C[g] = A'[:, pg] @ B'[:, pg]^T,
+ * A' [M, PR], B' [N, PR]; group g owns padded-row range pg (64-multiples). */
+int vh_group_gemm_wg256_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                             const int64_t* padded_cumsum, int G, int64_t M,
+                             int64_t N, int64_t PR, void* stream);
+
 /* ---- Fused MoE elementwise --------------------------------------------- */
 
 /* Fused epilogue: act = silu(gate) * up * w_row, where gate/up are the two

@@ -288,3 +288,25 @@ def test_group_gemm_nk8_ragged(lib):
             torch.testing.assert_close(c[start:end].float(), ref, rtol=3e-2, atol=3e-2,
                                        msg=lambda m: f"g={g} trans_b={trans_b} N={N} K={K}: {m}")
             start = end
+
+
+def test_group_gemm_wgrad_transpose_path(lib):
+    """Large-rows wgrad goes through transpose-pad + wg256; parity vs fp32
+    per-group matmul with ragged + empty groups."""
+    torch.manual_seed(13)
+    counts = [1000, 0, 1500, 37, 1559]
+    G, rows = len(counts), sum(counts)
+    cumsum = torch.tensor(counts).cumsum(0).cuda()
+    a = bf(torch.randn(rows, 1536) * 0.3).cuda()
+    b = bf(torch.randn(rows, 2048) * 0.3).cuda()
+    c = lib.group_gemm_mn(a, b, cumsum, G)
+    start = 0
+    for g in range(G):
+        end = int(cumsum[g])
+        if end == start:
+            assert torch.all(c[g] == 0), f"group {g} not zero-filled"
+            continue
+        ref = a[start:end].float().t() @ b[start:end].float()
+        torch.testing.assert_close(c[g].float(), ref, rtol=3e-2, atol=3e-2,
+                                   msg=lambda m: f"group {g}: {m}")
+        start = end

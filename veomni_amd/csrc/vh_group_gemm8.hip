@@ -765,3 +765,197 @@ extern "C" int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B,
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ============================================================================
+// Transpose-pad + wg256: the fast wgrad path.
+//
+// vh_transpose_pad_bf16 turns [rows, C] (rows grouped by expert via cumsum)
+// into [C, padded_rows] where each group's row-range is zero-padded to a
+// 64-multiple (padded_cumsum, host-computed). One memory-bound LDS-tiled
+// transpose (~2x tensor bytes) buys both wgrad operands a K-contiguous
+// layout, so the wgrad itself runs the nk256 glds structure (wg256) instead
+// of the 2x-slower transposed-register staging.
+// ============================================================================
+
+namespace {
+
+__global__ void k_transpose_pad(const bf16_t* __restrict__ src,
+                                bf16_t* __restrict__ dst,
+                                const int64_t* __restrict__ cumsum,
+                                const int64_t* __restrict__ padded_cumsum,
+                                int G, int64_t C, int64_t padded_total) {
+  // block: 64 padded-rows x 64 cols tile; 256 threads. [64][72] image:
+  // 144-B rows, 16-B aligned, odd 16-B slot count breaks conflicts.
+  __shared__ __attribute__((aligned(16))) bf16_t t2[64][72];
+
+  int64_t chunk = blockIdx.x;          // which 64-row padded chunk
+  int64_t col0 = (int64_t)blockIdx.y * 64;
+  int64_t p0 = chunk * 64;
+  // find group: padded chunks never straddle groups (64-multiples)
+  int g = 0;
+  while (g < G && p0 >= padded_cumsum[g]) ++g;
+  if (g >= G) return;
+  int64_t pstart = (g > 0) ? padded_cumsum[g - 1] : 0;
+  int64_t sstart = (g > 0) ? cumsum[g - 1] : 0;
+  int64_t kcount = cumsum[g] - sstart;
+  int64_t local0 = p0 - pstart;        // offset within the group
+
+  // read 64 src rows (coalesced: 8 threads x 16 B per row)
+  const int tid = threadIdx.x;
+  for (int r = tid / 8; r < 64; r += 32) {
+    int64_t lr = local0 + r;
+    const int cpos = (tid % 8) * 8;
+    bf16x8 v = {};
+    if (lr < kcount && col0 + cpos < C) {
+      v = *reinterpret_cast<const bf16x8*>(src + (sstart + lr) * C + col0 + cpos);
+    }
+    *reinterpret_cast<bf16x8*>(&t2[r][cpos]) = v;
+  }
+  __syncthreads();
+  // write transposed: dst[col][p0 + r], 16 B per thread along padded rows
+  for (int c = tid / 8; c < 64; c += 32) {
+    int rpos = (tid % 8) * 8;
+    if (col0 + c < C) {
+      bf16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = t2[rpos + j][c];
+      *reinterpret_cast<bf16x8*>(dst + (col0 + c) * padded_total + p0 + rpos) = v;
+    }
+  }
+}
+
+// wg256: C[g][M][N] = A'[:, pg] @ B'[:, pg]^T where A' [M, PR], B' [N, PR]
+// are the transpose-padded operands; group g owns padded-row range pg.
+// Same staging/compute structure as k_group_gemm_nk256.
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_wg256(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ padded_cumsum, int G,
+    int64_t M, int64_t N, int64_t PR, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 32768; };
+  auto tb = [&](int buf) { return sm + 16384 + buf * 32768; };
+
+  const int gid = blockIdx.y;
+  const int64_t p_start = (gid > 0) ? padded_cumsum[gid - 1] : 0;
+  const int64_t klen = padded_cumsum[gid] - p_start;  // 64-multiple
+  const int bm = blockIdx.x / tiles_n;
+  const int bn = blockIdx.x % tiles_n;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  bf16_t* Cg = C + (int64_t)gid * M * N;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  if (klen > 0) {
+    const bf16_t* Ag = A + p_start;
+    const bf16_t* Bg = B + p_start;
+    KStage64 sa, sb;
+    sa.init(Ag, PR, [&](int r) -> int64_t {
+      int64_t gm = (int64_t)bm * BM8 + r;
+      return gm % M;
+    }, tid);
+    sb.init(Bg, PR, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+
+    const int nk = (int)(klen / BK64);
+    sa.stage(ta(0), 0);
+    sb.stage(tb(0), 0);
+    __syncthreads();
+    int cur = 0;
+    for (int t = 0; t < nk; ++t) {
+      if (t + 1 < nk) {
+        sa.stage(ta(cur ^ 1), (int64_t)(t + 1) * BK64);
+        sb.stage(tb(cur ^ 1), (int64_t)(t + 1) * BK64);
+      }
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16frag af[4], bfr[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) bfr[j] = frag_read64(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + i * 16, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + 64 + i * 16, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+  const int col_in = lane & 15;
+  const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                    row_base_in + rr;
+        int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+        if (m < M && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+      }
+}
+
+}  // namespace
+
+extern "C" int vh_transpose_pad_bf16(const uint16_t* src, uint16_t* dst,
+                                     const int64_t* cumsum,
+                                     const int64_t* padded_cumsum, int G,
+                                     int64_t C, int64_t padded_total,
+                                     void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(C % 8 == 0, "C %% 8 != 0");
+  VH_CHECK(padded_total % 64 == 0, "padded_total %% 64 != 0");
+  dim3 grid((uint32_t)(padded_total / 64), (uint32_t)((C + 63) / 64));
+  hipLaunchKernelGGL(k_transpose_pad, grid, dim3(256), 0, s,
+                     reinterpret_cast<const bf16_t*>(src),
+                     reinterpret_cast<bf16_t*>(dst), cumsum, padded_cumsum, G,
+                     C, padded_total);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+extern "C" int vh_group_gemm_wg256_bf16(const uint16_t* A, const uint16_t* B,
+                                        uint16_t* C,
+                                        const int64_t* padded_cumsum, int G,
+                                        int64_t M, int64_t N, int64_t PR,
+                                        void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(PR % 64 == 0, "PR %% 64 != 0");
+  int tiles_m = (int)((M + BM8 - 1) / BM8);
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  dim3 grid(tiles_m * tiles_n, G);
+  hipLaunchKernelGGL(k_group_gemm_wg256, grid, dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), padded_cumsum, G, M, N, PR,
+                     tiles_m, tiles_n);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

@@ -53,6 +53,9 @@ def get_lib() -> ctypes.CDLL:
     _sig(lib, "vh_group_gemm_nk256_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
          c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_mn_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
+    _sig(lib, "vh_transpose_pad_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
+    _sig(lib, "vh_group_gemm_wg256_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
+         c_i64, c_i64, c_p)
     _sig(lib, "vh_moe_silu_mul_weighted_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
     _sig(lib, "vh_moe_silu_mul_weighted_bwd_bf16", c_p, c_p, c_p, c_p, c_p,
          c_i64, c_i64, c_int, c_p)
@@ -193,12 +196,37 @@ def group_gemm_nk(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
 
 def group_gemm_mn(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
                   G: int) -> torch.Tensor:
-    """C[g] = A_g^T @ B_g; A [rows, M], B [rows, N] -> C [G, M, N]."""
+    """C[g] = A_g^T @ B_g; A [rows, M], B [rows, N] -> C [G, M, N].
+
+    Large shapes take the transpose-pad + wg256 path: one memory-bound
+    transpose of each operand buys the wgrad a K-contiguous glds kernel
+    (~2x the TF/s of transposed register staging). The padded total is the
+    host-known bound rows + 64*G, so no device->host sync is needed."""
     assert a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16
+    rows = a.shape[0]
     M, N = a.shape[1], b.shape[1]
     c = torch.empty(G, M, N, dtype=a.dtype, device=a.device)
     cs = cumsum.to(torch.int64).contiguous()
-    with _prof("group_gemm_mn", 2.0 * a.shape[0] * M * N):
+    if M >= 512 and N >= 512 and rows >= 2048:
+        counts = torch.diff(cs, prepend=cs.new_zeros(1))
+        pad_cs = ((counts + 63) // 64 * 64).cumsum(0).contiguous()
+        PR = rows + 64 * G
+        at = torch.empty(M, PR, dtype=a.dtype, device=a.device)
+        bt = torch.empty(N, PR, dtype=a.dtype, device=a.device)
+        lib = get_lib()
+        with _prof("wgrad_transpose", 2.0 * rows * (M + N) * 2):
+            check(lib.vh_transpose_pad_bf16(dptr(a.contiguous()), dptr(at), dptr(cs),
+                                            dptr(pad_cs), G, M, PR, cur_stream()),
+                  "vh_transpose_pad(a)")
+            check(lib.vh_transpose_pad_bf16(dptr(b.contiguous()), dptr(bt), dptr(cs),
+                                            dptr(pad_cs), G, N, PR, cur_stream()),
+                  "vh_transpose_pad(b)")
+        with _prof("group_gemm_mn", 2.0 * rows * M * N):
+            check(lib.vh_group_gemm_wg256_bf16(dptr(at), dptr(bt), dptr(c),
+                                               dptr(pad_cs), G, M, N, PR,
+                                               cur_stream()), "vh_group_gemm_wg256")
+        return c
+    with _prof("group_gemm_mn", 2.0 * rows * M * N):
         check(get_lib().vh_group_gemm_mn_bf16(
             dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, M, N,
             cur_stream()), "vh_group_gemm_mn")
