@@ -108,10 +108,8 @@ def main():
 
     if world > 1:
         dist.init_process_group("nccl")
-    else:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29555")
-        dist.init_process_group("gloo", rank=0, world_size=1)
+    # world == 1: no process group at all — gloo's C++ banner pollutes
+    # stdout and would break the single-JSON-line contract.
 
     import veomni_amd.ops  # registrations
     from veomni_amd.ops import HIP_OPS_CONFIG, hip_lib
@@ -161,7 +159,7 @@ def main():
 
     for i in range(args.warmup):
         loss = one_step()
-        log(f"warmup {i}: loss {float(loss):.4f}")
+        log(f"warmup {i}: loss {float(loss.detach()):.4f}")
 
     # profile one extra (untimed) step for the per-kernel roofline leg
     hip_lib.profile_enable(True)
@@ -248,7 +246,8 @@ def main():
         }
         print(json.dumps(out), flush=True)
 
-    dist.destroy_process_group()
+    if world > 1:
+        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -148,7 +148,15 @@ def init_parallel_state(
     ep_size | dp_shard_sp_size (ref :598-627).
     """
     global _PARALLEL_STATE
-    assert dist.is_initialized(), "init_process_group first"
+    if not dist.is_initialized():
+        # single-process fast path (no process group, no meshes): the FSDP2
+        # wrap short-circuits at fsdp_size == 1 and SP/EP are disabled.
+        assert ulysses_size == 1 and ep_size == 1, "multi-dim parallel needs a process group"
+        _PARALLEL_STATE = ParallelState(
+            world_size=1, dp_size=1, dp_mode=dp_mode,
+            device_type=device_type or ("cuda" if torch.cuda.is_available() else "cpu"),
+        )
+        return _PARALLEL_STATE
     world_size = dist.get_world_size()
     if device_type is None:
         device_type = "cuda" if torch.cuda.is_available() else "cpu"
