@@ -199,7 +199,7 @@ def main():
         ach = fl / (rec["ms_avg"] / 1000.0) / 1e12
         roofline = {"bound": "mfma", "achieved": round(ach, 1), "peak": 2500.0,
                     "unit": "TFLOP/s", "frac": round(ach / 2500.0, 4), "traffic": None,
-                    "kernel": "vh_group_gemm_nk_bf16"}
+                    "kernel": "vh_group_gemm_nk*_bf16 (shape-dispatched)"}
     elif "ce_fwd" in prof:
         rec = prof["ce_fwd"]
         by = sum(rec["work"]) / rec["count"]
