@@ -128,7 +128,10 @@ def main():
     log(f"building {preset} on cuda:{local_rank} (ep={ep_size}, ws={world})")
     t_build = time.time()
     model = build_model(preset, dtype=torch.bfloat16, device="cuda")
-    model.use_checkpoint = True
+    # 288 GB HBM3E: dense llama-8b at N=1 holds full activations comfortably
+    # (no forward recompute). The 30B MoE keeps checkpointing (scattered
+    # expert activations are ~1 GB/layer/rank).
+    model.use_checkpoint = cfg.is_moe
     model = build_parallelize_model(model)
     def make_opt(fused):
         return torch.optim.AdamW(model.parameters(), lr=1e-5, betas=(0.9, 0.95),
