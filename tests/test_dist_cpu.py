@@ -220,3 +220,48 @@ def _ep_slice(rank, ws):
 
 def test_ep_param_slice():
     spawn(_ep_slice)
+
+
+# ------------------------------------------------------------- async ulysses
+def _async_ulysses(rank, ws):
+    from veomni_amd.distributed.parallel_state import init_parallel_state, set_parallel_state
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch, sp_collate
+
+    bind_ops("eager")
+    torch.manual_seed(0)
+
+    # sync SP run
+    init_parallel_state(ulysses_size=ws)
+    model = build_model("tiny-dense")
+    full = synthetic_batch(512, 64, seed=11)
+    batch = sp_collate(full)
+    loss_sync, _ = model(**batch)
+    loss_sync.backward()
+    g_sync = torch.nn.utils.get_total_norm(
+        [p.grad for p in model.parameters() if p.grad is not None]
+    ).detach().clone()
+    model.zero_grad(set_to_none=True)
+
+    # async SP run — same model, same batch
+    init_parallel_state(ulysses_size=ws, async_ulysses=True)
+    loss_async, _ = model(**batch)
+    loss_async.backward()
+    g_async = torch.nn.utils.get_total_norm(
+        [p.grad for p in model.parameters() if p.grad is not None]
+    ).detach().clone()
+
+    torch.testing.assert_close(loss_async.detach(), loss_sync.detach(), rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(g_async, g_sync, rtol=1e-4, atol=1e-5)
+
+    # and both match the no-SP single-process run on the full batch
+    set_parallel_state(None)
+    ref = build_model("tiny-dense")
+    loss_ref, _ = ref(**full)
+    torch.testing.assert_close(loss_async.detach().float(), loss_ref.detach().float(),
+                               rtol=5e-3, atol=5e-4)
+
+
+def test_async_ulysses_equivalence():
+    spawn(_async_ulysses)

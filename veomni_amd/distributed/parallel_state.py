@@ -30,6 +30,7 @@ class ParallelState:
     ulysses_size: int = 1
     ep_size: int = 1
     dp_mode: str = "fsdp2"
+    async_ulysses: bool = False
     device_type: str = "cpu"
     device_mesh: Optional[DeviceMesh] = None      # (dp_shard, ulysses)
     ep_device_mesh: Optional[DeviceMesh] = None   # (ep_fsdp, ep)
@@ -138,6 +139,7 @@ def init_parallel_state(
     ulysses_size: int = 1,
     ep_size: int = 1,
     dp_mode: str = "fsdp2",
+    async_ulysses: bool = False,
     device_type: Optional[str] = None,
 ) -> ParallelState:
     """Build the device meshes and register the global state.
@@ -184,6 +186,7 @@ def init_parallel_state(
         ulysses_size=ulysses_size,
         ep_size=ep_size,
         dp_mode=dp_mode,
+        async_ulysses=async_ulysses,
         device_type=device_type,
         device_mesh=mesh,
         ep_device_mesh=ep_mesh,

@@ -215,3 +215,68 @@ class ReduceLoss(torch.autograd.Function):
 
 def reduce_sequence_parallel_loss(loss: Tensor, num_valid_tokens: Tensor, group=None) -> Tensor:
     return ReduceLoss.apply(loss, num_valid_tokens, group)
+
+
+# ---------------------------------------------------------------- async path
+# Async Ulysses (parity target: async_ulysses.py:48-212 fwd): the three QKV
+# all-to-alls are launched per-tensor with async_op=True so the exchange of q
+# overlaps the k/v projections (and all three overlap on the comm stream).
+# Backward stays the synchronous mirrored a2a this round.
+_ASYNC_WORK: dict = {}
+
+
+class _A2AStartSeqHeads(torch.autograd.Function):
+    """Start [.., S/sp, h, ..] -> [.., S, h/sp, ..] async (scatter over the
+    head dim, gather over the LEADING seq dim — the buffer is directly the
+    gathered layout, no post-concat needed)."""
+
+    @staticmethod
+    def forward(ctx, group, x: Tensor, seq_dim: int, head_dim: int) -> Tensor:
+        ctx.group = group
+        ctx.seq_dim = seq_dim
+        ctx.head_dim = head_dim
+        ws = dist.get_world_size(group)
+        assert seq_dim == 0 and head_dim == 1, "async path expects [S, h, D] tensors"
+        g, s = x.shape[0], x.shape[1]
+        xs = (
+            x.reshape([g, ws, s // ws] + list(x.shape[2:]))
+            .transpose(0, 1)
+            .reshape([g * ws, s // ws] + list(x.shape[2:]))
+            .contiguous()
+        )
+        out = torch.empty_like(xs)
+        work = dist.all_to_all_single(out, xs, group=group, async_op=True)
+        _ASYNC_WORK[id(out)] = work
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_output: Tensor):
+        return (None, all_to_all_tensor(grad_output, ctx.seq_dim, ctx.head_dim, ctx.group), None, None)
+
+
+class _A2AWait(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor) -> Tensor:
+        work = _ASYNC_WORK.pop(id(x), None)
+        if work is not None:
+            work.wait()
+        return x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, grad_output: Tensor):
+        return grad_output
+
+
+def gather_seq_scatter_heads_async(x: Tensor, group=None) -> Tensor:
+    """Start the seq->head exchange; returns the (not yet valid) buffer."""
+    group = get_ulysses_sequence_parallel_group() if group is None else group
+    if not group:
+        return x
+    return _A2AStartSeqHeads.apply(group, x, 0, 1)
+
+
+def wait_gathered(x: Tensor, group=None) -> Tensor:
+    group = get_ulysses_sequence_parallel_group() if group is None else group
+    if not group:
+        return x
+    return _A2AWait.apply(x)

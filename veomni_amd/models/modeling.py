@@ -133,9 +133,12 @@ def repeat_kv(hidden_states: torch.Tensor, n_rep: int) -> torch.Tensor:
 
 def sdpa_attention(module, query, key, value, attention_mask, dropout=0.0,
                    scaling=None, **kwargs):
-    """Eager/SDPA attention on [B,h,S,D] (HF sdpa semantics)."""
-    key = repeat_kv(key, module.num_key_value_groups)
-    value = repeat_kv(value, module.num_key_value_groups)
+    """Eager/SDPA attention on [B,h,S,D] (HF sdpa semantics). GQA repeat is
+    derived from the shapes (under Ulysses the per-rank head counts differ
+    from the module's config attrs)."""
+    n_rep = query.shape[1] // key.shape[1]
+    key = repeat_kv(key, n_rep)
+    value = repeat_kv(value, n_rep)
     out = F.scaled_dot_product_attention(
         query, key, value, attn_mask=None, dropout_p=dropout, scale=scaling, is_causal=True
     )
@@ -177,12 +180,83 @@ class Attention(nn.Module):
         q, k, v = q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2)
         cos, sin = position_embeddings
         q, k = apply_rotary_pos_emb(q, k, cos, sin)
-        if veomni_attention.use_non_eager_impl:
+        ps = get_parallel_state()
+        if ps.ulysses_enabled and ps.async_ulysses:
+            # async Ulysses (ref async_ulysses.py:48-212): per-tensor async
+            # a2a launched as each projection finishes; RoPE commutes with
+            # the seq gather (elementwise per position), so it runs on the
+            # local slice. KV heads repeat first when sp > kv heads.
+            out = self._async_ulysses_attention(q, k, v)
+        elif veomni_attention.use_non_eager_impl:
+            # the bound attention kernel is SP-aware (does the Ulysses
+            # exchange itself, ref attention/flash.py:236-299)
             out, _ = veomni_attention(self, q, k, v, None, dropout=0.0, scaling=self.scaling)
+        elif ps.ulysses_enabled:
+            # eager under SP still needs the sync Ulysses exchange
+            out = self._sync_ulysses_attention(q, k, v)
         else:
             out, _ = sdpa_attention(self, q, k, v, None, dropout=0.0, scaling=self.scaling)
         out = out.reshape(*input_shape, -1).contiguous()
         return self.o_proj(out)
+
+    def _sync_ulysses_attention(self, q, k, v):
+        from ..distributed.sequence_parallel import (
+            gather_heads_scatter_seq,
+            gather_seq_scatter_heads,
+        )
+
+        ps = get_parallel_state()
+        group = ps.ulysses_group
+        sp = ps.ulysses_size
+        assert q.shape[0] == 1, "ulysses expects packed batch (B == 1)"
+        kv = k.shape[1]
+        if sp > kv:
+            rep = sp // kv
+            k = torch.repeat_interleave(k, dim=1, repeats=rep)
+            v = torch.repeat_interleave(v, dim=1, repeats=rep)
+        qs = gather_seq_scatter_heads(q.squeeze(0).transpose(0, 1).contiguous(),
+                                      seq_dim=0, head_dim=1, group=group)
+        ks = gather_seq_scatter_heads(k.squeeze(0).transpose(0, 1).contiguous(),
+                                      seq_dim=0, head_dim=1, group=group)
+        vs = gather_seq_scatter_heads(v.squeeze(0).transpose(0, 1).contiguous(),
+                                      seq_dim=0, head_dim=1, group=group)
+        out, _ = sdpa_attention(self, qs.transpose(0, 1)[None], ks.transpose(0, 1)[None],
+                                vs.transpose(0, 1)[None], None, dropout=0.0,
+                                scaling=self.scaling)  # [1, S, h/sp, D]
+        out = gather_heads_scatter_seq(out.squeeze(0), head_dim=1, seq_dim=0, group=group)
+        return out[None]
+
+    def _async_ulysses_attention(self, q, k, v):
+        from ..distributed.sequence_parallel import (
+            gather_heads_scatter_seq,
+            gather_seq_scatter_heads_async,
+            wait_gathered,
+        )
+
+        ps = get_parallel_state()
+        sp = ps.ulysses_size
+        group = ps.ulysses_group
+        # [B, h, S_loc, D] -> [S_loc, h, D] (B == 1 packed path)
+        assert q.shape[0] == 1, "async ulysses expects packed batch (B == 1)"
+        kv = k.shape[1]
+        if sp > kv:
+            rep = sp // kv
+            k = torch.repeat_interleave(k, dim=1, repeats=rep)
+            v = torch.repeat_interleave(v, dim=1, repeats=rep)
+        qs = q.squeeze(0).transpose(0, 1)  # [S_loc, h, D]
+        ks = k.squeeze(0).transpose(0, 1)
+        vs = v.squeeze(0).transpose(0, 1)
+        qb = gather_seq_scatter_heads_async(qs.contiguous(), group=group)
+        kb = gather_seq_scatter_heads_async(ks.contiguous(), group=group)
+        vb = gather_seq_scatter_heads_async(vs.contiguous(), group=group)
+        qg = wait_gathered(qb, group=group).transpose(0, 1)[None]  # [1, h/sp, S, D]
+        kg = wait_gathered(kb, group=group).transpose(0, 1)[None]
+        vg = wait_gathered(vb, group=group).transpose(0, 1)[None]
+        out, _ = sdpa_attention(self, qg, kg, vg, None, dropout=0.0,
+                                scaling=self.scaling)  # [1, S, h/sp, D]
+        out = out.squeeze(0)
+        out = gather_heads_scatter_seq(out, head_dim=1, seq_dim=0, group=group)
+        return out[None]
 
 
 class MLP(nn.Module):
