@@ -95,6 +95,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", type=str, default=None)
     ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--batch", type=int, default=0, help="micro-batch per rank (0 = auto)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -121,6 +122,9 @@ def main():
 
     preset = args.model or ("llama3-8b" if n_gpus == 1 else "qwen3-moe-30b")
     cfg = PRESETS[preset]
+    # auto micro-batch: 288 GB HBM holds llama-8b activations for 4x4096
+    # tokens without checkpointing; the 30B MoE stays at 1 (ckpt on).
+    mbs = args.batch or (1 if cfg.is_moe else 4)
     ep_size = n_gpus if (cfg.is_moe and n_gpus > 1) else 1
     init_parallel_state(ep_size=ep_size, device_type="cuda")
     bind_ops(HIP_OPS_CONFIG)
@@ -150,7 +154,7 @@ def main():
     log(f"built in {time.time() - t_build:.1f}s; mem {torch.cuda.memory_allocated()/2**30:.1f} GiB")
 
     seq = args.seq_len
-    batch = synthetic_batch(cfg.vocab_size, seq, seed=42 + rank, device="cuda")
+    batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs, seed=42 + rank, device="cuda")
 
     def one_step():
         loss, _ = model(**batch)
@@ -187,11 +191,11 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dt = float(t)
 
-    tokens_per_step = seq * n_gpus
+    tokens_per_step = seq * mbs * n_gpus
     toks_per_s = tokens_per_step * args.steps / dt
     ms_per_step = dt / args.steps * 1000.0
 
-    flops = step_flops(cfg, seq, seq) * n_gpus  # per step whole job
+    flops = step_flops(cfg, seq, seq) * mbs * n_gpus  # per step whole job
     mfu = flops * args.steps / dt / (n_gpus * 2.5e15)
 
     # roofline: dominant hand-written kernel
@@ -240,7 +244,7 @@ def main():
                 "workload": f"{preset} FSDP2 bf16 seq{seq}"
                 + (f" EP{ep_size}" if ep_size > 1 else ""),
                 "model": preset,
-                "global_batch": n_gpus,
+                "global_batch": mbs * n_gpus,
                 "seq_len": seq,
                 "parallelism": f"dp{world}" + (f"_ep{ep_size}" if ep_size > 1 else ""),
             },

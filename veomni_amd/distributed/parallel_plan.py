@@ -39,6 +39,7 @@ class ParallelPlan:
             return
         ep_rank = ps.ep_rank
         ep_size = ps.ep_size
+        model._ep_fqns = set()  # read by the checkpointer (EP-rank-keyed save)
         named = dict(model.named_parameters())
         for fqn, param in named.items():
             for pattern, dim in plan.items():
@@ -52,6 +53,7 @@ class ParallelPlan:
                 new_param = nn.Parameter(local, requires_grad=param.requires_grad)
                 new_param._ep_param = True
                 _set_param_by_fqn(model, fqn, new_param)
+                model._ep_fqns.add(fqn)
                 break
 
 
