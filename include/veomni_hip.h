@@ -164,6 +164,15 @@ int vh_silu_mul_bwd_bf16(const uint16_t* dy, const uint16_t* gate,
                          const uint16_t* up, uint16_t* dgate, uint16_t* dup,
                          int64_t n, void* stream);
 
+/* ---- Flash attention (causal, GQA, D = 128) ----------------------------- */
+
+/* Forward: O [B,Hq,S,128] bf16, LSE [B,Hq,S] fp32; S % 128 == 0.
+ * Replaces the external flash_attn wheel behind the reference's attention
+ * slot (attention/flash.py:153-301) for the packed causal path. */
+int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
+                     uint16_t* O, float* LSE, int B, int Hq, int Hkv,
+                     int64_t S, float scale, void* stream);
+
 /* ---- Fused chunked cross-entropy ---------------------------------------- */
 
 /* Per-row softmax CE over a bf16 logits chunk:

@@ -20,6 +20,7 @@ SOURCES = [
     "vh_group_gemm.hip",
     "vh_group_gemm8.hip",
     "vh_norms.hip",
+    "vh_attention.hip",
     "vh_ce.hip",
 ]
 
