@@ -101,34 +101,44 @@ __global__ __launch_bounds__(256, 2) void k_attn_fwd(
   float m_run = -1e30f;
   float l_run = 0.f;
 
+  // persistent per-lane staging addresses (advance by KB*DH per tile)
+  const bf16_t* ksrc[4];
+  int klds[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int base = i * 4096 + wave * 1024;
+    int o = base + lane * 16;
+    int row = o >> 8;
+    int colb = o & 255;
+    ksrc[i] = Kb + (int64_t)row * DH + (kswz(row, colb) >> 1);
+    klds[i] = base >> 1;
+  }
+  const int v_kv = tid & 63;
+  const int v_d0 = (tid >> 6) * 8;     // 4 d-chunks per thread (stride 32 rows)
+  const bf16_t* vsrc = Vb + (int64_t)v_kv * DH + v_d0;
+
   const int t_max = (int)(((int64_t)qb * QB + QB - 1) / KB);  // inclusive
   for (int t = 0; t <= t_max; ++t) {
     // ---- stage K tile via glds (4 instructions; swizzled source)
     {
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        int base = i * 4096 + wave * 1024;
-        int o = base + lane * 16;
-        int row = o >> 8;
-        int colb = o & 255;
-        const bf16_t* g = Kb + ((int64_t)t * KB + row) * DH + (kswz(row, colb) >> 1);
-        glds16a(g, kt + (base >> 1));
-      }
+      for (int i = 0; i < 4; ++i) glds16a(ksrc[i], kt + klds[i]);
       // ---- stage V transposed: thread t loads 16 B (8 d at one kv) and
-      // scatters 8 2-B writes into [d][kv]
+      // scatters 8 2-B writes into [d][kv]; 4 units serialised to bound
+      // register liveness
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        int unit = tid + u * 256;        // 1024 units of [1 kv][8 d]
-        int kv = unit & 63;
-        int d0 = (unit >> 6) * 8;
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(
-            Vb + ((int64_t)t * KB + kv) * DH + d0);
+        int d0 = v_d0 + u * 32;
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(vsrc + u * 32);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int row = d0 + j;
-          vt[(row * 128 + vswz(row, kv * 2)) >> 1] = v.v[j];
+          vt[(row * 128 + vswz(row, v_kv * 2)) >> 1] = v.v[j];
         }
       }
+#pragma unroll
+      for (int i = 0; i < 4; ++i) ksrc[i] += KB * DH;
+      vsrc += KB * DH;
     }
     __syncthreads();
 
@@ -168,12 +178,16 @@ __global__ __launch_bounds__(256, 2) void k_attn_fwd(
       mt = fmaxf(mt, xor32(mt));
       float m_new = fmaxf(m_run, mt);
       float alpha = __expf(m_run - m_new);
-      // ---- exponentiate + partial row sum
+      // ---- exponentiate + partial row sum, packing pairs to bf16 on the
+      // fly (frees the fp32 score registers early)
       float psum = 0.f;
+      uint32_t pk[8];  // 4 quads x 2 dwords of packed bf16 pairs
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        p[r] = __expf(p[r] - m_new);
-        psum += p[r];
+      for (int r = 0; r < 16; r += 2) {
+        float e0 = __expf(p[r] - m_new);
+        float e1 = __expf(p[r + 1] - m_new);
+        psum += e0 + e1;
+        pk[r >> 1] = (uint32_t)f2bf(e0) | ((uint32_t)f2bf(e1) << 16);
       }
       psum += xor32(psum);
       l_run = l_run * alpha + psum;
@@ -192,32 +206,31 @@ __global__ __launch_bounds__(256, 2) void k_attn_fwd(
         for (int d = 0; d < 4; ++d) oacc[d][r] *= a_r;
       }
 
-      // ---- pack P to bf16 quads and exchange halves for the A-fragment
-      // quad qd (regs 4qd..4qd+3) covers kv = 8*qd + 4*half .. +4
-      uint32_t pk[8];  // 4 quads x 2 dwords of packed bf16 pairs
-#pragma unroll
-      for (int qd = 0; qd < 4; ++qd) {
-        pk[2 * qd] = (uint32_t)f2bf(p[4 * qd + 0]) | ((uint32_t)f2bf(p[4 * qd + 1]) << 16);
-        pk[2 * qd + 1] = (uint32_t)f2bf(p[4 * qd + 2]) | ((uint32_t)f2bf(p[4 * qd + 3]) << 16);
-      }
-      uint32_t pr[8];
-#pragma unroll
-      for (int i = 0; i < 8; ++i) pr[i] = __shfl_xor((int)pk[i], 32, 64);
-
-      // assemble A-fragments: chunk m covers kv 16m..16m+15
-      //   half0: [own quad 2m | partner quad 2m]   (kv 8m*2.. : own 0-3, partner 4-7)
-      //   half1: [partner quad 2m+1 | own quad 2m+1]
+      // ---- exchange packed quads for the P^T A-fragment. Chunk m covers
+      // kv 16m..16m+15: half0 = [own quad 2m | partner quad 2m],
+      // half1 = [partner quad 2m+1 | own quad 2m+1]. Each lane only needs
+      // the partner's MATCHING quad: exchange its own counterpart (the
+      // shuffle is symmetric, both sides send the quad the other needs:
+      // half0 sends quad 2m+1's slot? no — both halves hold quads 0..3 of
+      // DIFFERENT kv sets; the partner's quad with the SAME index is the
+      // one required, so a plain xor-32 shuffle of quad pairs suffices).
       bf16frag pa[2];
 #pragma unroll
       for (int mch = 0; mch < 2; ++mch) {
-        uint32_t w0, w1, w2, w3;
-        if (half == 0) {
-          w0 = pk[4 * mch];     w1 = pk[4 * mch + 1];
-          w2 = pr[4 * mch];     w3 = pr[4 * mch + 1];
-        } else {
-          w0 = pr[4 * mch + 2]; w1 = pr[4 * mch + 3];
-          w2 = pk[4 * mch + 2]; w3 = pk[4 * mch + 3];
-        }
+        // convergent exchange: each half sends the quad the partner needs
+        // (half0 sends quad 2m, half1 sends quad 2m+1); select BEFORE the
+        // collective so every lane executes the same shuffles.
+        // half0 needs the partner's quad 2m (so half1 sends quad 2m);
+        // half1 needs the partner's quad 2m+1 (so half0 sends quad 2m+1)
+        uint32_t s0 = half ? pk[4 * mch] : pk[4 * mch + 2];
+        uint32_t s1 = half ? pk[4 * mch + 1] : pk[4 * mch + 3];
+        uint32_t o0 = (uint32_t)__shfl_xor((int)s0, 32, 64);
+        uint32_t o1 = (uint32_t)__shfl_xor((int)s1, 32, 64);
+        // half0 frag = [own 2m | partner 2m]; half1 = [partner 2m+1 | own 2m+1]
+        uint32_t w0 = half ? o0 : pk[4 * mch];
+        uint32_t w1 = half ? o1 : pk[4 * mch + 1];
+        uint32_t w2 = half ? pk[4 * mch + 2] : o0;
+        uint32_t w3 = half ? pk[4 * mch + 3] : o1;
         uint4 u{w0, w1, w2, w3};
         pa[mch] = __builtin_bit_cast(bf16frag, u);
       }
