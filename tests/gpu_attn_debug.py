@@ -47,7 +47,7 @@ def ref_attn(q, k, v, scale):
 def main():
     torch.manual_seed(0)
     dev = "cuda"
-    for (B, Hq, Hkv, S) in [(1, 1, 1, 128), (2, 4, 2, 256), (1, 8, 2, 1024)]:
+    for (B, Hq, Hkv, S) in [(1, 1, 1, 256), (2, 4, 2, 256), (1, 8, 2, 1024)]:
         q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
         k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
         v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
