@@ -18,7 +18,7 @@
 //     O += mfma(A=P^T, B=V^T).
 //   Epilogue: O / l via lane broadcasts; LSE = m + log(l) saved for bwd.
 //
-// LDS: Q 8x8 KiB + 2x(K 16 + V^T 16) = 128 KiB -> 1 block/CU, 2 waves/SIMD.
+// LDS: 2x(K 16 + V^T 16) = 64 KiB; Q lives in registers (tile-invariant).
 
 #include "vh_common.h"
 
@@ -57,12 +57,11 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
     float* __restrict__ LSE, int B, int Hq, int Hkv, int64_t S, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* qt = reinterpret_cast<bf16_t*>(smem);              // 8 x 8 KiB
   auto kt = [&](int buf) {                                   // 2 x 16 KiB
-    return reinterpret_cast<bf16_t*>(smem + 65536 + buf * 16384);
+    return reinterpret_cast<bf16_t*>(smem + buf * 16384);
   };
   auto vt = [&](int buf) {                                   // 2 x 16 KiB
-    return reinterpret_cast<bf16_t*>(smem + 98304 + buf * 16384);
+    return reinterpret_cast<bf16_t*>(smem + 32768 + buf * 16384);
   };
 
   const int qb = blockIdx.x;
@@ -85,19 +84,13 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 
   const int64_t q_global = (int64_t)qb * QB + wave * WQ + col;
 
-  // ---- stage this wave's 32 Q rows into LDS once (glds, kswz image)
-  {
-    bf16_t* qw = qt + wave * 4096;
+  // ---- Q fragments straight into registers (tile-invariant: chunk c =
+  // Q[q = col][c*16 + half*8 .. +8))
+  bf16frag qreg[8];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int base = i * 1024;
-      int o = base + lane * 16;
-      int row = o >> 8;
-      int colb = o & 255;
-      const bf16_t* g = Qb + ((int64_t)qb * QB + wave * WQ + row) * DH +
-                        (kswz(row, colb) >> 1);
-      glds16a(g, qw + (base >> 1));
-    }
+  for (int c = 0; c < 8; ++c) {
+    qreg[c] = *reinterpret_cast<const bf16frag*>(
+        Qb + q_global * DH + c * 16 + half * 8);
   }
 
   // ---- persistent staging addresses
@@ -153,12 +146,13 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     }
 
     const bool diag = ((int64_t)(t + 1) * KB) > ((int64_t)qb * QB + wave * WQ);
+    // causal skip: every kv in this tile is beyond every q of this wave
+    const bool live = (int64_t)t * KB <= (int64_t)qb * QB + wave * WQ + (WQ - 1);
     const bf16_t* ktc = kt(cur);
     const bf16_t* vtc = vt(cur);
-    const char* qw = reinterpret_cast<const char*>(qt) + wave * 8192;
 
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
+    for (int sub = 0; sub < 2 && live; ++sub) {
       // ---- S^T = K·Q^T over 8 d-chunks
       f32x16 sacc = f32x16{};
 #pragma unroll
@@ -167,9 +161,7 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
         int colb = (c * 16 + half * 8) * 2;
         bf16frag kf = *reinterpret_cast<const bf16frag*>(
             reinterpret_cast<const char*>(ktc) + krow * 256 + kswz(krow, colb));
-        bf16frag qf = *reinterpret_cast<const bf16frag*>(
-            qw + col * 256 + kswz(col, colb));
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, sacc, 0, 0, 0);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qreg[c], sacc, 0, 0, 0);
       }
 
       // ---- scale + causal mask + per-lane max
@@ -287,7 +279,7 @@ extern "C" int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(S % QB == 0, "S %% 256 != 0 (pad the sequence)");
   VH_CHECK(Hq % Hkv == 0, "Hq %% Hkv != 0");
   dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
-  hipLaunchKernelGGL(k_attn_fwd, grid, dim3(512), 131072, s,
+  hipLaunchKernelGGL(k_attn_fwd, grid, dim3(512), 65536, s,
                      reinterpret_cast<const bf16_t*>(Q),
                      reinterpret_cast<const bf16_t*>(K),
                      reinterpret_cast<const bf16_t*>(V),
