@@ -111,3 +111,37 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def probe():
+    import ctypes
+    lib = L.get_lib()
+    fn = lib.vh_attn_fwd_probe_bf16
+    fn.restype = ctypes.c_int
+    fn.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 + [ctypes.c_int64, ctypes.c_float, ctypes.c_int, ctypes.c_void_p]
+    dev = "cuda"
+    B, Hq, Hkv, S = 1, 32, 8, 4096
+    import math
+    scale = 1.0 / math.sqrt(128)
+    q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    o = torch.empty_like(q)
+    lse = torch.empty(B, Hq, S, dtype=torch.float32, device=dev)
+    fl = 2 * 2 * S * S * Hq * 128 * 0.5
+    for mode, name in [(0, "full"), (1, "no-softmax"), (2, "no-PV"), (3, "no-QK")]:
+        for _ in range(3):
+            fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), lse.data_ptr(),
+               B, Hq, Hkv, S, scale, mode, L.cur_stream())
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), lse.data_ptr(),
+               B, Hq, Hkv, S, scale, mode, L.cur_stream())
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        print(f"probe {name}: {dt*1e3:.3f} ms ({fl/dt/1e12:.0f} TF/s-equiv)", flush=True)
+
+
+if os.environ.get("VH_ATTN_PROBE"):
+    probe()

@@ -48,10 +48,18 @@ __device__ __forceinline__ int vswz(int row, int colb) {
   return colb ^ ((((row >> 1) ^ (row >> 3)) & 7) << 4);
 }
 
-__device__ __forceinline__ float xor32(float v) {
-  return __shfl_xor(v, 32, 64);
+// 32-lane-distance exchange via v_permlane32_swap (1 VALU op) instead of
+// __shfl_xor's ds_bpermute (~50-cycle LDS round trip) — guide T12.
+__device__ __forceinline__ uint32_t swap32_u(uint32_t v, int half) {
+  auto r = __builtin_amdgcn_permlane32_swap(v, v, false, false);
+  return half ? (uint32_t)r[0] : (uint32_t)r[1];
 }
 
+__device__ __forceinline__ float xor32h(float v, int half) {
+  return __builtin_bit_cast(float, swap32_u(__builtin_bit_cast(uint32_t, v), half));
+}
+
+template <int MODE>
 __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
@@ -83,6 +91,7 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
   float* Lb = LSE + ((int64_t)b * Hq + hq) * S;
 
   const int64_t q_global = (int64_t)qb * QB + wave * WQ + col;
+  const float scale2 = scale * 1.4426950408889634f;  // log2(e)
 
   // ---- Q fragments straight into registers (tile-invariant: chunk c =
   // Q[q = col][c*16 + half*8 .. +8))
@@ -155,36 +164,65 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     for (int sub = 0; sub < 2 && live; ++sub) {
       // ---- S^T = K·Q^T over 8 d-chunks
       f32x16 sacc = f32x16{};
+      if (MODE != 3) {
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        int krow = sub * 32 + col;
-        int colb = (c * 16 + half * 8) * 2;
-        bf16frag kf = *reinterpret_cast<const bf16frag*>(
-            reinterpret_cast<const char*>(ktc) + krow * 256 + kswz(krow, colb));
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qreg[c], sacc, 0, 0, 0);
+        for (int c = 0; c < 8; ++c) {
+          int krow = sub * 32 + col;
+          int colb = (c * 16 + half * 8) * 2;
+          bf16frag kf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(ktc) + krow * 256 + kswz(krow, colb));
+          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qreg[c], sacc, 0, 0, 0);
+        }
       }
 
-      // ---- scale + causal mask + per-lane max
+      if (MODE == 3) {
+        // keep sacc live without QK cost accounting (probe only)
+        asm volatile("" :: "v"(sacc[0]));
+      }
+      // ---- scale + causal mask in the exp2 domain (scale2 = scale*log2e
+      // folded into the score multiply; v_exp_f32 IS 2^x, so exp2-domain
+      // bookkeeping drops one multiply per element)
       float p[16];
-      float mt = -INFINITY;
+      const int kv_lim = (int)(q_global - (int64_t)t * KB) - sub * 32 - 4 * half;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float sc = sacc[r] * scale;
-        if (diag) {
-          int64_t kv_g = (int64_t)t * KB + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-          if (kv_g > q_global) sc = -INFINITY;
-        }
+        float sc = sacc[r] * scale2;
+        if (diag && ((r & 3) + 8 * (r >> 2)) > kv_lim) sc = -INFINITY;
         p[r] = sc;
-        mt = fmaxf(mt, sc);
       }
-      mt = fmaxf(mt, xor32(mt));
+      float m8[8], m4[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) m8[i] = fmaxf(p[2 * i], p[2 * i + 1]);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) m4[i] = fmaxf(m8[2 * i], m8[2 * i + 1]);
+      float mt = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
+      mt = fmaxf(mt, xor32h(mt, half));
 
+      if (MODE == 1) {
+        // probe: skip ALL softmax VALU/shuffles; fabricate pa from sacc bits
+        bf16frag pa1[2];
+        uint4 u1{__builtin_bit_cast(uint32_t, sacc[0]), __builtin_bit_cast(uint32_t, sacc[1]),
+                 __builtin_bit_cast(uint32_t, sacc[2]), __builtin_bit_cast(uint32_t, sacc[3])};
+        pa1[0] = __builtin_bit_cast(bf16frag, u1);
+        pa1[1] = pa1[0];
+#pragma unroll
+        for (int mch = 0; mch < 2; ++mch)
+#pragma unroll
+          for (int d = 0; d < 4; ++d) {
+            int vrow = d * 32 + col;
+            int colb2 = (sub * 32 + mch * 16 + half * 8) * 2;
+            bf16frag vf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(vtc) + vrow * 128 + vswz(vrow, colb2));
+            oacc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1[mch], vf, oacc[d], 0, 0, 0);
+          }
+        continue;
+      }
       // ---- defer-max (T13): rescale only when some lane's max moved past
       // the threshold (wave-uniform decision via ballot)
       bool need = mt > m_run + DEFER_THR;
       if (__builtin_amdgcn_ballot_w64(need) != 0ull) {
         float m_new = fmaxf(m_run, mt);
-        float alpha = __expf(m_run - m_new);
+        float alpha = __builtin_exp2f(m_run - m_new);
         m_run = m_new;
         l_run *= alpha;
 #pragma unroll
@@ -196,17 +234,24 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
         }
       }
 
-      // ---- exponentiate + pack pairs + partial row sum
-      float psum = 0.f;
+      // ---- exponentiate + pack pairs + tree row sum
+      float s8[8];
       uint32_t pk[8];
 #pragma unroll
       for (int r = 0; r < 16; r += 2) {
-        float e0 = __expf(p[r] - m_run);
-        float e1 = __expf(p[r + 1] - m_run);
-        psum += e0 + e1;
-        pk[r >> 1] = (uint32_t)f2bf(e0) | ((uint32_t)f2bf(e1) << 16);
+        float e0 = __builtin_exp2f(p[r] - m_run);
+        float e1 = __builtin_exp2f(p[r + 1] - m_run);
+        s8[r >> 1] = e0 + e1;
+        // native bf16 converts (v_cvt) — the integer-emulated RNE rounding
+        // was ~6 VALU ops per element on the hot path
+        uint16_t b0 = __builtin_bit_cast(uint16_t, (__bf16)e0);
+        uint16_t b1 = __builtin_bit_cast(uint16_t, (__bf16)e1);
+        pk[r >> 1] = (uint32_t)b0 | ((uint32_t)b1 << 16);
       }
-      psum += xor32(psum);
+      float s4a = (s8[0] + s8[1]) + (s8[2] + s8[3]);
+      float s4b = (s8[4] + s8[5]) + (s8[6] + s8[7]);
+      float psum = s4a + s4b;
+      psum += xor32h(psum, half);
       l_run += psum;
 
       // ---- partner-quad exchange -> P^T A-fragments
@@ -215,8 +260,8 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
       for (int mch = 0; mch < 2; ++mch) {
         uint32_t s0 = half ? pk[4 * mch] : pk[4 * mch + 2];
         uint32_t s1 = half ? pk[4 * mch + 1] : pk[4 * mch + 3];
-        uint32_t o0 = (uint32_t)__shfl_xor((int)s0, 32, 64);
-        uint32_t o1 = (uint32_t)__shfl_xor((int)s1, 32, 64);
+        uint32_t o0 = swap32_u(s0, half);
+        uint32_t o1 = swap32_u(s1, half);
         uint32_t w0 = half ? o0 : pk[4 * mch];
         uint32_t w1 = half ? o1 : pk[4 * mch + 1];
         uint32_t w2 = half ? pk[4 * mch + 2] : o0;
@@ -226,16 +271,20 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
       }
 
       // ---- O += P^T · V
+      if (MODE != 2) {
 #pragma unroll
-      for (int mch = 0; mch < 2; ++mch) {
+        for (int mch = 0; mch < 2; ++mch) {
 #pragma unroll
-        for (int d = 0; d < 4; ++d) {
-          int vrow = d * 32 + col;
-          int colb = (sub * 32 + mch * 16 + half * 8) * 2;
-          bf16frag vf = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(vtc) + vrow * 128 + vswz(vrow, colb));
-          oacc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[mch], vf, oacc[d], 0, 0, 0);
+          for (int d = 0; d < 4; ++d) {
+            int vrow = d * 32 + col;
+            int colb = (sub * 32 + mch * 16 + half * 8) * 2;
+            bf16frag vf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(vtc) + vrow * 128 + vswz(vrow, colb));
+            oacc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[mch], vf, oacc[d], 0, 0, 0);
+          }
         }
+      } else {
+        asm volatile("" :: "v"(pa[0]), "v"(pa[1]));
       }
     }
 
@@ -265,7 +314,8 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     }
   }
   if (half == 0) {
-    Lb[q_global] = m_run + __logf(l_run);
+    // back to the natural-log domain: LSE = ln2 * (m2 + log2(l))
+    Lb[q_global] = 0.6931471805599453f * (m_run + __log2f(l_run));
   }
 }
 
@@ -279,11 +329,29 @@ extern "C" int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(S % QB == 0, "S %% 256 != 0 (pad the sequence)");
   VH_CHECK(Hq % Hkv == 0, "Hq %% Hkv != 0");
   dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
-  hipLaunchKernelGGL(k_attn_fwd, grid, dim3(512), 65536, s,
+  hipLaunchKernelGGL(k_attn_fwd<0>, grid, dim3(512), 65536, s,
                      reinterpret_cast<const bf16_t*>(Q),
                      reinterpret_cast<const bf16_t*>(K),
                      reinterpret_cast<const bf16_t*>(V),
                      reinterpret_cast<bf16_t*>(O), LSE, B, Hq, Hkv, S, scale);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* ablation probe: mode 1 = no softmax VALU, 2 = no PV, 3 = no QK (timing only) */
+extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
+                                      const uint16_t* V, uint16_t* O,
+                                      float* LSE, int B, int Hq, int Hkv,
+                                      int64_t S, float scale, int mode,
+                                      void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
+#define VH_AM(M_)                                                                hipLaunchKernelGGL(k_attn_fwd<M_>, grid, dim3(512), 65536, s,                                     reinterpret_cast<const bf16_t*>(Q),                                            reinterpret_cast<const bf16_t*>(K),                                            reinterpret_cast<const bf16_t*>(V),                                            reinterpret_cast<bf16_t*>(O), LSE, B, Hq, Hkv, S, scale)
+  if (mode == 1) VH_AM(1);
+  else if (mode == 2) VH_AM(2);
+  else if (mode == 3) VH_AM(3);
+  else VH_AM(0);
+#undef VH_AM
   VH_HIP(hipGetLastError());
   return 0;
 }
