@@ -173,6 +173,20 @@ int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                      uint16_t* O, float* LSE, int B, int Hq, int Hkv,
                      int64_t S, float scale, void* stream);
 
+/* Backward preprocess: delta[r] = rowsum(dO[r] * O[r]), lse2[r] = LSE[r]*log2e
+ * over rows = B*Hq*S flattened. */
+int vh_attn_bwd_pre_bf16(const uint16_t* dO, const uint16_t* O,
+                         const float* LSE, float* delta, float* lse2,
+                         int64_t rows, void* stream);
+
+/* Backward: dQacc [B,Hq,S,128] fp32 (caller zero-fills; atomically
+ * accumulated), dK/dV [B,Hq,S,128] bf16 written per Q-head — the host sums
+ * GQA head groups down to [B,Hkv,S,128]. delta/lse2 from the preprocess. */
+int vh_attn_bwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
+                     const uint16_t* dO, const float* delta, const float* lse2,
+                     float* dQacc, uint16_t* dK, uint16_t* dV, int B, int Hq,
+                     int Hkv, int64_t S, float scale, void* stream);
+
 /* ---- Fused chunked cross-entropy ---------------------------------------- */
 
 /* Per-row softmax CE over a bf16 logits chunk:

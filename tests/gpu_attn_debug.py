@@ -79,6 +79,56 @@ def main():
     err = (o.float() - oref).abs().max().item()
     print(f"spike: max|dO|={err:.4g} ok={err < 3e-2}", flush=True)
 
+    # ---- backward parity vs torch autograd (fp32 reference)
+    for (B, Hq, Hkv, S) in [(1, 1, 1, 128), (1, 1, 1, 256), (2, 4, 2, 256), (1, 8, 2, 1024)]:
+        q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+        k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+        v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+        do = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+        scale = 1.0 / math.sqrt(128)
+        o, lse = attn_fwd(q, k, v, scale)
+        dq, dk, dv_ = L.attn_bwd(q, k, v, o, lse, do, scale)
+        qf = q.float().requires_grad_(True)
+        kf = k.float().requires_grad_(True)
+        vf = v.float().requires_grad_(True)
+        oref, _ = ref_attn(qf, kf, vf, scale)
+        oref.backward(do.float())
+        edq = (dq.float() - qf.grad).abs().max().item()
+        edk = (dk.float() - kf.grad).abs().max().item()
+        edv = (dv_.float() - vf.grad).abs().max().item()
+        sc = qf.grad.abs().max().item()
+        ok = edq < 0.06 * max(sc, 1.0) and edk < 0.06 * max(kf.grad.abs().max().item(), 1.0) \
+            and edv < 0.06 * max(vf.grad.abs().max().item(), 1.0)
+        print(f"bwd B{B} Hq{Hq} Hkv{Hkv} S{S}: max|ddq|={edq:.4g} max|ddk|={edk:.4g} "
+              f"max|ddv|={edv:.4g} (|dq|max={sc:.3g}) ok={ok}", flush=True)
+        if not ok:
+            print("dq[0,0,:2,:6]    ", dq.float()[0, 0, :2, :6].tolist())
+            print("dqref[0,0,:2,:6] ", qf.grad[0, 0, :2, :6].tolist())
+            print("dk[0,0,:2,:6]    ", dk.float()[0, 0, :2, :6].tolist())
+            print("dkref[0,0,:2,:6] ", kf.grad[0, 0, :2, :6].tolist())
+            print("dv[0,0,:2,:6]    ", dv_.float()[0, 0, :2, :6].tolist())
+            print("dvref[0,0,:2,:6] ", vf.grad[0, 0, :2, :6].tolist())
+            return
+
+    # backward timing at llama shape
+    B, Hq, Hkv, S = 1, 32, 8, 4096
+    scale = 1.0 / math.sqrt(128)
+    q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    do = torch.randn_like(q)
+    o, lse = attn_fwd(q, k, v, scale)
+    for _ in range(3):
+        L.attn_bwd(q, k, v, o, lse, do, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        L.attn_bwd(q, k, v, o, lse, do, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 10
+    flb = 3 * 2 * 2 * S * S * Hq * 128 * 0.5
+    print(f"bwd llama shape: {dt*1e3:.3f} ms  {flb/dt/1e12:.0f} TF/s", flush=True)
+
     # timing at llama shape
     B, Hq, Hkv, S = 1, 32, 8, 4096
     q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)

@@ -355,3 +355,361 @@ extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ============================================================================
+// Flash attention backward (bf16, causal, GQA, D = 128).
+//
+// FA2 column-block structure: grid (S/128, B*Hq); a block (4 waves, 256 thr)
+// owns 128 kv rows (wave w the 32-row slice w) and loops over 32-row q tiles
+// from the causal diagonal to S. Both score orientations are recomputed so
+// no cross-lane transpose is needed (the MFMA A-operand always comes from
+// the forward's cheap reg->chunk pack-exchange):
+//   or1 (q in regs, kv = lane):  S1 = mfma(K, Q), dP1 = mfma(V, dO)
+//       -> P1, dS1 packs feed  dV += mfma(P1^T, dO^T),  dK += mfma(dS1^T, Q^T)
+//   or2 (kv in regs, q = lane): S2 = mfma(Q, K), dP2 = mfma(dO, V)
+//       -> dS2 pack feeds      dQ += mfma(dS2, K^T)
+// P = exp2(S*scale2 - lse2[q]); dS = P * (dP - delta[q]) * scale.
+// dQ partials are block-reduced in LDS fp32 and atomically added to a fp32
+// buffer; dK/dV are written per Q-head and the host sums GQA groups.
+// delta = rowsum(dO*O), lse2 = LSE*log2e from vh_attn_bwd_pre_bf16.
+// ============================================================================
+
+namespace {
+
+// swizzle for 64-B rows ([128][32] q-side transposed tiles): 4 16-B slots
+// per row; bank row = 256 B = 4 tile rows -> slot = (row>>2)&3 (the vswz
+// 8-slot map would overflow a 64-B row and alias across rows).
+__device__ __forceinline__ int qswz(int row, int colb) {
+  return colb ^ (((row >> 2) & 3) << 4);
+}
+
+__global__ void k_attn_bwd_pre(const bf16_t* __restrict__ dO,
+                               const bf16_t* __restrict__ O,
+                               const float* __restrict__ LSE,
+                               float* __restrict__ delta,
+                               float* __restrict__ lse2, int64_t rows) {
+  int wave = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  int lane = threadIdx.x & (kWave - 1);
+  int num_waves = (gridDim.x * blockDim.x) / kWave;
+  for (int64_t r = wave; r < rows; r += num_waves) {
+    const bf16x8* d8 = reinterpret_cast<const bf16x8*>(dO + r * DH);
+    const bf16x8* o8 = reinterpret_cast<const bf16x8*>(O + r * DH);
+    float acc = 0.f;
+    if (lane < 16) {
+      bf16x8 a = d8[lane], b = o8[lane];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += bf2f(a.v[j]) * bf2f(b.v[j]);
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+    if (lane == 0) {
+      delta[r] = acc;
+      lse2[r] = LSE[r] * 1.4426950408889634f;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256, 1) void k_attn_bwd(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
+    const float* __restrict__ delta, const float* __restrict__ lse2,
+    float* __restrict__ dQacc, bf16_t* __restrict__ dK,
+    bf16_t* __restrict__ dV, int B, int Hq, int Hkv, int64_t S, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [128][128] 32 K
+  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 32768);    // [128][128] 32 K
+  bf16_t* ktr = reinterpret_cast<bf16_t*>(smem + 65536);     // [128][128] 32 K
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 98304);     // [128][32] 8 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 106496);   // [128][32] 8 K
+  float* dqred = reinterpret_cast<float*>(smem + 114688);    // [32][128] 16 K
+  float* lsed = reinterpret_cast<float*>(smem + 131072);     // [32]
+  float* deld = reinterpret_cast<float*>(smem + 131200);     // [32]
+
+  const int kvb = blockIdx.x;          // kv block of 128 rows
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 0..3 = kv slice
+  const int half = lane >> 5;
+  const int col = lane & 31;
+
+  const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+  const bf16_t* Kb = K + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* Vb = V + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+  const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+  const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+  float* dQb = dQacc + (((int64_t)b * Hq + hq) * S) * DH;
+  bf16_t* dKb = dK + (((int64_t)b * Hq + hq) * S) * DH;  // per-HQ; host sums GQA
+  bf16_t* dVb = dV + (((int64_t)b * Hq + hq) * S) * DH;
+
+  const int64_t kv0 = (int64_t)kvb * 128;
+  const int kvrow_l = wave * 32 + col;   // this lane's kv row (or1 / K,V reads)
+
+  // ---- stage K/V rows + K^T once per block (256 thr: 8 passes of 4 KiB)
+  {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int o = i * 4096 + tid * 16;
+      int row = o >> 8;
+      int colb = o & 255;
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(krow) + row * 256 + kswz(row, colb)) =
+          *reinterpret_cast<const bf16x8*>(Kb + (kv0 + row) * DH + (colb >> 1));
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(vrow) + row * 256 + kswz(row, colb)) =
+          *reinterpret_cast<const bf16x8*>(Vb + (kv0 + row) * DH + (colb >> 1));
+    }
+    // K^T [128 d][128 kv]: 2048 units of [1 kv][8 d] / 256 thr = 8 each
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      int unit = tid + u * 256;
+      int kv = unit & 127;
+      int d0 = (unit >> 7) * 8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(Kb + (kv0 + kv) * DH + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int row = d0 + j;
+        ktr[(row * 256 + kswz(row, kv * 2)) >> 1] = v.v[j];
+      }
+    }
+  }
+
+  f32x16 dv_acc[4], dk_acc[4];
+#pragma unroll
+  for (int d = 0; d < 4; ++d) {
+    dv_acc[d] = f32x16{};
+    dk_acc[d] = f32x16{};
+  }
+
+  const float scale2 = scale * 1.4426950408889634f;
+  const int qt0 = (int)(kv0 / 32);
+  const int qtn = (int)(S / 32);
+
+  for (int qt = qt0; qt < qtn; ++qt) {
+    const int64_t q0 = (int64_t)qt * 32;
+    // ---- stage Q^T / dO^T (512 units of [1 q][8 d] / 256 thr = 2 each)
+    {
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        int unit = tid + u * 256;
+        int q = unit & 31;
+        int d0 = (unit >> 5) * 8;
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0 + q) * DH + d0);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0 + q) * DH + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = d0 + j;
+          qtr[(row * 64 + qswz(row, q * 2)) >> 1] = vq.v[j];
+          dotr[(row * 64 + qswz(row, q * 2)) >> 1] = vd.v[j];
+        }
+      }
+      if (tid < 32) {
+        lsed[tid] = lseb[q0 + tid];
+        deld[tid] = delb[q0 + tid];
+      }
+      for (int i = tid; i < 32 * DH; i += 256) dqred[i] = 0.f;
+    }
+    __syncthreads();
+
+    // this wave has work only when some of its kv rows are <= some q
+    const bool live = (q0 + 31) >= (kv0 + wave * 32);
+    const bool diag = (q0 < kv0 + 128);
+
+    if (live) {
+      // per-wave Q/dO row fragments at q = q0+col (or1 B-operands and or2
+      // A-operands: either way lane l holds row q0 + (l&31))
+      bf16frag qrow[8], dorow[8];
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        qrow[c] = *reinterpret_cast<const bf16frag*>(Qb + (q0 + col) * DH + c * 16 + half * 8);
+        dorow[c] = *reinterpret_cast<const bf16frag*>(dOb + (q0 + col) * DH + c * 16 + half * 8);
+      }
+      const float lse_l = lsed[col];
+      const float del_l = deld[col];
+
+      // ---- phase 1 (or1): C = [kv regs][q lanes] -> dS1 -> dQ.
+      // s1 = mfma(K, Q): rows = this wave's 32 kv, cols = q (lane scalars
+      // lse/delta, the forward's own orientation).
+      f32x16 s1 = f32x16{}, dp1 = f32x16{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int colb = (c * 16 + half * 8) * 2;
+        bf16frag kf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        bf16frag vf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(vrow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qrow[c], s1, 0, 0, 0);
+        dp1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dorow[c], dp1, 0, 0, 0);
+      }
+      uint32_t dg1[8];
+#pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        float g[2];
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+          int r2 = r + rr;
+          int kvl = (r2 & 3) + 8 * (r2 >> 2) + 4 * half + wave * 32;
+          bool masked = diag && ((int64_t)kv0 + kvl > q0 + col);
+          float pp = masked ? 0.f : __builtin_exp2f(s1[r2] * scale2 - lse_l);
+          g[rr] = pp * (dp1[r2] - del_l) * scale;
+        }
+        dg1[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)g[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)g[1]) << 16);
+      }
+      bf16frag da1[2];
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+        uint32_t a0 = half ? dg1[4 * mch] : dg1[4 * mch + 2];
+        uint32_t a1 = half ? dg1[4 * mch + 1] : dg1[4 * mch + 3];
+        uint32_t b0 = swap32_u(a0, half);
+        uint32_t b1 = swap32_u(a1, half);
+        uint4 u{half ? b0 : dg1[4 * mch], half ? b1 : dg1[4 * mch + 1],
+                half ? dg1[4 * mch + 2] : b0, half ? dg1[4 * mch + 3] : b1};
+        da1[mch] = __builtin_bit_cast(bf16frag, u);
+      }
+      // dQ[q][d] += dS1^T(pack: A[q][kv-chunk]) x K^T-tile(B[kv][d])
+#pragma unroll
+      for (int dblk = 0; dblk < 4; ++dblk) {
+        f32x16 dq = f32x16{};
+#pragma unroll
+        for (int mch = 0; mch < 2; ++mch) {
+          int trow = dblk * 32 + col;
+          int colb = (wave * 32 + mch * 16 + half * 8) * 2;
+          bf16frag ktf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(ktr) + trow * 256 + kswz(trow, colb));
+          dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1[mch], ktf, dq, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
+          atomicAdd(&dqred[qr * DH + dblk * 32 + col], dq[r]);
+        }
+      }
+
+      // ---- phase 2 (or2): C = [q regs][kv lanes] -> P2, dS2 -> dV, dK.
+      // s2 = mfma(Q, K): rows = q tile, cols = this wave's 32 kv.
+      f32x16 s2 = f32x16{}, dp2 = f32x16{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int colb = (c * 16 + half * 8) * 2;
+        bf16frag kf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        bf16frag vf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(vrow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qrow[c], kf, s2, 0, 0, 0);
+        dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dorow[c], vf, dp2, 0, 0, 0);
+      }
+      uint32_t pk2[8], dg2[8];
+#pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        float pv[2], gv[2];
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+          int r2 = r + rr;
+          int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
+          bool masked = diag && ((int64_t)kv0 + wave * 32 + col > q0 + qrm);
+          float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lsed[qrm]);
+          pv[rr] = pp;
+          gv[rr] = pp * (dp2[r2] - deld[qrm]) * scale;
+        }
+        pk2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[1]) << 16);
+        dg2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[1]) << 16);
+      }
+      bf16frag pa2[2], da2[2];
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+        uint32_t a0 = half ? pk2[4 * mch] : pk2[4 * mch + 2];
+        uint32_t a1 = half ? pk2[4 * mch + 1] : pk2[4 * mch + 3];
+        uint32_t b0 = swap32_u(a0, half);
+        uint32_t b1 = swap32_u(a1, half);
+        uint4 u{half ? b0 : pk2[4 * mch], half ? b1 : pk2[4 * mch + 1],
+                half ? pk2[4 * mch + 2] : b0, half ? pk2[4 * mch + 3] : b1};
+        pa2[mch] = __builtin_bit_cast(bf16frag, u);
+        uint32_t c0 = half ? dg2[4 * mch] : dg2[4 * mch + 2];
+        uint32_t c1 = half ? dg2[4 * mch + 1] : dg2[4 * mch + 3];
+        uint32_t e0 = swap32_u(c0, half);
+        uint32_t e1 = swap32_u(c1, half);
+        uint4 u2{half ? e0 : dg2[4 * mch], half ? e1 : dg2[4 * mch + 1],
+                 half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
+        da2[mch] = __builtin_bit_cast(bf16frag, u2);
+      }
+      // dV[kv][d] += P2^T(pack: A[kv][q-chunk]) x dO^T-tile(B[q][d]);
+      // dK[kv][d] += dS2^T x Q^T-tile
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+#pragma unroll
+        for (int dblk = 0; dblk < 4; ++dblk) {
+          int trow = dblk * 32 + col;
+          int colb = (mch * 16 + half * 8) * 2;
+          bf16frag dof = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(dotr) + trow * 64 + qswz(trow, colb));
+          bf16frag qf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(qtr) + trow * 64 + qswz(trow, colb));
+          dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
+          dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < 32 * DH; i += 256) {
+      float vsum = dqred[i];
+      if (vsum != 0.f) atomicAdd(&dQb[q0 * DH + i], vsum);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: dK/dV bf16 for this wave's kv rows (C layout: kv in regs,
+  // d = lane col + 32*dblk)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int kvr = (r & 3) + 8 * (r >> 2) + 4 * half;
+    int64_t kvg = kv0 + wave * 32 + kvr;
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      dKb[kvg * DH + d * 32 + col] = f2bf(dk_acc[d][r]);
+      dVb[kvg * DH + d * 32 + col] = f2bf(dv_acc[d][r]);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int vh_attn_bwd_pre_bf16(const uint16_t* dO, const uint16_t* O,
+                                    const float* LSE, float* delta,
+                                    float* lse2, int64_t rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  int blocks = (int)((rows + 3) / 4);
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(k_attn_bwd_pre, dim3(blocks), dim3(256), 0, s,
+                     reinterpret_cast<const bf16_t*>(dO),
+                     reinterpret_cast<const bf16_t*>(O), LSE, delta, lse2, rows);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+extern "C" int vh_attn_bwd_bf16(const uint16_t* Q, const uint16_t* K,
+                                const uint16_t* V, const uint16_t* dO,
+                                const float* delta, const float* lse2,
+                                float* dQacc, uint16_t* dK, uint16_t* dV,
+                                int B, int Hq, int Hkv, int64_t S, float scale,
+                                void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(S % 128 == 0, "S %% 128 != 0");
+  dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
+  hipLaunchKernelGGL(k_attn_bwd, grid, dim3(256), 131328, s,
+                     reinterpret_cast<const bf16_t*>(Q),
+                     reinterpret_cast<const bf16_t*>(K),
+                     reinterpret_cast<const bf16_t*>(V),
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2, dQacc,
+                     reinterpret_cast<bf16_t*>(dK),
+                     reinterpret_cast<bf16_t*>(dV), B, Hq, Hkv, S, scale);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

@@ -66,6 +66,11 @@ def get_lib() -> ctypes.CDLL:
     _sig(lib, "vh_silu_mul_bf16", c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_silu_mul_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_ce_fwd_bf16", c_p, c_p, c_p, c_p, c_i64, c_i64, c_f32, c_i64, c_p)
+    _sig(lib, "vh_attn_fwd_bf16", c_p, c_p, c_p, c_p, c_p, c_int, c_int, c_int,
+         c_i64, c_f32, c_p)
+    _sig(lib, "vh_attn_bwd_pre_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
+    _sig(lib, "vh_attn_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p,
+         c_int, c_int, c_int, c_i64, c_f32, c_p)
     _LIB = lib
     return lib
 
@@ -326,3 +331,51 @@ def ce_fwd(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float,
                                        grad_scale, ignore_index, cur_stream()),
               "vh_ce_fwd")
     return loss_rows, dlogits
+
+
+def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float):
+    """Causal GQA flash attention forward: q [B,Hq,S,128] bf16 contiguous,
+    k/v [B,Hkv,S,128]. Returns (o, lse)."""
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    assert D == 128 and S % 128 == 0, (S, D)
+    o = torch.empty_like(q)
+    lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+    with _prof("attn_fwd", 2.0 * 2 * S * S * Hq * D * 0.5 * B):
+        check(get_lib().vh_attn_fwd_bf16(dptr(q), dptr(k), dptr(v), dptr(o),
+                                         dptr(lse), B, Hq, Hkv, S, scale,
+                                         cur_stream()), "vh_attn_fwd")
+    return o, lse
+
+
+def attn_bwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+             o: torch.Tensor, lse: torch.Tensor, do: torch.Tensor,
+             scale: float):
+    """Backward for attn_fwd. Returns (dq bf16 [B,Hq,S,D],
+    dk/dv bf16 [B,Hkv,S,D] with GQA head groups summed)."""
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    rows = B * Hq * S
+    do = do.contiguous()
+    delta = torch.empty(rows, dtype=torch.float32, device=q.device)
+    lse2 = torch.empty(rows, dtype=torch.float32, device=q.device)
+    lib = get_lib()
+    with _prof("attn_bwd", 3.0 * 2 * 2 * S * S * Hq * D * 0.5 * B):
+        check(lib.vh_attn_bwd_pre_bf16(dptr(do), dptr(o), dptr(lse.contiguous()),
+                                       dptr(delta), dptr(lse2), rows,
+                                       cur_stream()), "vh_attn_bwd_pre")
+        dqacc = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
+        dkh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
+        dvh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
+        check(lib.vh_attn_bwd_bf16(dptr(q), dptr(k), dptr(v), dptr(do),
+                                   dptr(delta), dptr(lse2), dptr(dqacc),
+                                   dptr(dkh), dptr(dvh), B, Hq, Hkv, S, scale,
+                                   cur_stream()), "vh_attn_bwd")
+        dq = dqacc.to(torch.bfloat16)
+        rep = Hq // Hkv
+        if rep > 1:
+            dk = dkh.view(B, Hkv, rep, S, D).float().sum(2).to(torch.bfloat16)
+            dv = dvh.view(B, Hkv, rep, S, D).float().sum(2).to(torch.bfloat16)
+        else:
+            dk, dv = dkh, dvh
+    return dq, dk, dv
