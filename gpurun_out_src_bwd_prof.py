@@ -1,0 +1,17 @@
+import sys, os, math, time
+sys.path.insert(0, "/root/repo")
+import torch
+from veomni_amd.ops import hip_lib as L
+dev = "cuda"
+B, Hq, Hkv, S = 1, 32, 8, 4096
+scale = 1.0 / math.sqrt(128)
+torch.manual_seed(0)
+q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+do = torch.randn_like(q)
+o, lse = L.attn_fwd(q, k, v, scale)
+for _ in range(6):
+    L.attn_bwd(q, k, v, o, lse, do, scale)
+torch.cuda.synchronize()
+print("done")

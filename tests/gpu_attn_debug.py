@@ -80,7 +80,7 @@ def main():
     print(f"spike: max|dO|={err:.4g} ok={err < 3e-2}", flush=True)
 
     # ---- backward parity vs torch autograd (fp32 reference)
-    for (B, Hq, Hkv, S) in [(1, 1, 1, 128), (1, 1, 1, 256), (2, 4, 2, 256), (1, 8, 2, 1024)]:
+    for (B, Hq, Hkv, S) in [(1, 1, 1, 256), (1, 2, 1, 512), (2, 4, 2, 256), (1, 8, 2, 1024)]:
         q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
         k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
         v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
@@ -195,3 +195,46 @@ def probe():
 
 if os.environ.get("VH_ATTN_PROBE"):
     probe()
+
+
+def probe_bwd():
+    import ctypes
+    lib = L.get_lib()
+    fn = lib.vh_attn_bwd_probe_bf16
+    fn.restype = ctypes.c_int
+    fn.argtypes = [ctypes.c_void_p] * 9 + [ctypes.c_int] * 3 + [ctypes.c_int64, ctypes.c_float, ctypes.c_int, ctypes.c_void_p]
+    dev = "cuda"
+    B, Hq, Hkv, S = 1, 32, 8, 4096
+    scale = 1.0 / math.sqrt(128)
+    q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    k = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    v = (torch.randn(B, Hkv, S, 128, device=dev) * 0.5).to(torch.bfloat16)
+    do = torch.randn_like(q)
+    o, lse = attn_fwd(q, k, v, scale)
+    rows = B * Hq * S
+    delta = torch.zeros(rows, dtype=torch.float32, device=dev)
+    lse2 = lse.flatten() * 1.4426950408889634
+    dqacc = torch.zeros(B, Hq, S, 128, dtype=torch.float32, device=dev)
+    dk = torch.empty(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    dv = torch.empty(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    flb = 3 * 2 * 2 * S * S * Hq * 128 * 0.5
+    names = {0: "full", 1: "no-dQ", 2: "no-phase2(dV/dK)", 3: "no-softmax", 4: "no-qt-staging", 5: "no-qrow-loads", 6: "no-global-flush", 7: "no-lds-dsadd", 8: "no-dq-mfma"}
+    for mode in [0, 1, 2, 3, 4, 5, 6, 7, 8]:
+        for _ in range(2):
+            rc = fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+                    delta.data_ptr(), lse2.data_ptr(), dqacc.data_ptr(),
+                    dk.data_ptr(), dv.data_ptr(), B, Hq, Hkv, S, scale, mode, L.cur_stream())
+            assert rc == 0, lib.vh_last_error()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(5):
+            fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+               delta.data_ptr(), lse2.data_ptr(), dqacc.data_ptr(),
+               dk.data_ptr(), dv.data_ptr(), B, Hq, Hkv, S, scale, mode, L.cur_stream())
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 5
+        print(f"bwd probe {names[mode]}: {dt*1e3:.3f} ms ({flb/dt/1e12:.0f} TF/s-equiv)", flush=True)
+
+
+if os.environ.get("VH_ATTN_PROBE_BWD"):
+    probe_bwd()

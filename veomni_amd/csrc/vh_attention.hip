@@ -409,6 +409,11 @@ __global__ void k_attn_bwd_pre(const bf16_t* __restrict__ dO,
   }
 }
 
+// MODE (probe ablation, product path uses 0): 1 = skip the dQ path (LDS
+// tree + global flush), 2 = skip phase 2 (dV/dK), 3 = skip softmax VALU
+// (fabricated packs), 4 = skip per-tile Q^T/dO^T staging, 5 = skip qrow/dorow
+// global loads. Non-zero modes produce garbage results.
+template <int MODE>
 __global__ __launch_bounds__(256, 1) void k_attn_bwd(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -421,9 +426,12 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
   bf16_t* ktr = reinterpret_cast<bf16_t*>(smem + 65536);     // [128][128] 32 K
   bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 98304);     // [128][32] 8 K
   bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 106496);   // [128][32] 8 K
-  float* dqred = reinterpret_cast<float*>(smem + 114688);    // [32][128] 16 K
-  float* lsed = reinterpret_cast<float*>(smem + 131072);     // [32]
-  float* deld = reinterpret_cast<float*>(smem + 131200);     // [32]
+  // per-wave bf16 dQ partials: each wave owns a quarter, written with plain
+  // stores (LDS fp32 atomics measured ~600 cycles/op; a fp32 pairwise tree
+  // doubled register spans). One extra bf16 rounding per 32-kv partial.
+  bf16_t* dqred = reinterpret_cast<bf16_t*>(smem + 114688);  // [4][32][128] 32 K
+  float* lsed = reinterpret_cast<float*>(smem + 147456);     // [32]
+  float* deld = reinterpret_cast<float*>(smem + 147584);     // [32]
 
   const int kvb = blockIdx.x;          // kv block of 128 rows
   const int bh = blockIdx.y;
@@ -493,7 +501,7 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
   for (int qt = qt0; qt < qtn; ++qt) {
     const int64_t q0 = (int64_t)qt * 32;
     // ---- stage Q^T / dO^T (512 units of [1 q][8 d] / 256 thr = 2 each)
-    {
+    if (MODE != 4) {
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
         int unit = tid + u * 256;
@@ -512,7 +520,6 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
         lsed[tid] = lseb[q0 + tid];
         deld[tid] = delb[q0 + tid];
       }
-      for (int i = tid; i < 32 * DH; i += 256) dqred[i] = 0.f;
     }
     __syncthreads();
 
@@ -520,14 +527,25 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
     const bool live = (q0 + 31) >= (kv0 + wave * 32);
     const bool diag = (q0 < kv0 + 128);
 
+    // per-wave Q/dO row fragments at q = q0+col (or1 B-operands and or2
+    // A-operands: either way lane l holds row q0 + (l&31)); phase 2 reloads
+    // them from L2 so their 64 registers do not span the reduction section.
+    bf16_t* myq = dqred + wave * (32 * DH);
     if (live) {
-      // per-wave Q/dO row fragments at q = q0+col (or1 B-operands and or2
-      // A-operands: either way lane l holds row q0 + (l&31))
       bf16frag qrow[8], dorow[8];
+      if (MODE != 5) {
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        qrow[c] = *reinterpret_cast<const bf16frag*>(Qb + (q0 + col) * DH + c * 16 + half * 8);
-        dorow[c] = *reinterpret_cast<const bf16frag*>(dOb + (q0 + col) * DH + c * 16 + half * 8);
+        for (int c = 0; c < 8; ++c) {
+          qrow[c] = *reinterpret_cast<const bf16frag*>(Qb + (q0 + col) * DH + c * 16 + half * 8);
+          dorow[c] = *reinterpret_cast<const bf16frag*>(dOb + (q0 + col) * DH + c * 16 + half * 8);
+        }
+      } else {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          uint4 z{(uint32_t)(c + col), (uint32_t)half, 3u, 4u};
+          qrow[c] = __builtin_bit_cast(bf16frag, z);
+          dorow[c] = qrow[c];
+        }
       }
       const float lse_l = lsed[col];
       const float del_l = deld[col];
@@ -547,6 +565,10 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
         dp1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dorow[c], dp1, 0, 0, 0);
       }
       uint32_t dg1[8];
+      if (MODE == 3) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) dg1[i] = __builtin_bit_cast(uint32_t, s1[i]);
+      } else
 #pragma unroll
       for (int r = 0; r < 16; r += 2) {
         float g[2];
@@ -572,27 +594,80 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
                 half ? dg1[4 * mch + 2] : b0, half ? dg1[4 * mch + 3] : b1};
         da1[mch] = __builtin_bit_cast(bf16frag, u);
       }
-      // dQ[q][d] += dS1^T(pack: A[q][kv-chunk]) x K^T-tile(B[kv][d])
+      // dQ[q][d] += dS1^T(pack: A[q][kv-chunk]) x K^T-tile(B[kv][d]);
+      // each dblk's 16 values go straight to this wave's LDS quarter so the
+      // registers die immediately.
 #pragma unroll
-      for (int dblk = 0; dblk < 4; ++dblk) {
+      for (int dblk = 0; dblk < 4 && MODE != 1; ++dblk) {
         f32x16 dq = f32x16{};
+        if (MODE != 8) {
 #pragma unroll
-        for (int mch = 0; mch < 2; ++mch) {
-          int trow = dblk * 32 + col;
-          int colb = (wave * 32 + mch * 16 + half * 8) * 2;
-          bf16frag ktf = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(ktr) + trow * 256 + kswz(trow, colb));
-          dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1[mch], ktf, dq, 0, 0, 0);
+          for (int mch = 0; mch < 2; ++mch) {
+            int trow = dblk * 32 + col;
+            int colb = (wave * 32 + mch * 16 + half * 8) * 2;
+            bf16frag ktf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(ktr) + trow * 256 + kswz(trow, colb));
+            dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1[mch], ktf, dq, 0, 0, 0);
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r)
+            dq[r] = __builtin_bit_cast(float, dg1[r >> 1]) + (float)dblk;
         }
+        if (MODE == 7) {
+          asm volatile("" :: "v"(dq[0]), "v"(dq[15]));
+        } else {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
-          atomicAdd(&dqred[qr * DH + dblk * 32 + col], dq[r]);
+          for (int r = 0; r < 16; ++r) {
+            int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
+            myq[qr * DH + dblk * 32 + col] = f2bf(dq[r]);
+          }
         }
       }
+    }
 
-      // ---- phase 2 (or2): C = [q regs][kv lanes] -> P2, dS2 -> dV, dK.
-      // s2 = mfma(Q, K): rows = q tile, cols = this wave's 32 kv.
+    // ---- cross-wave dQ flush: non-live waves zero their quarter, then the
+    // block sums the 4 bf16 quarters in fp32 and atomically adds to global.
+    if (MODE != 1 && MODE != 7) {
+      if (!live) {
+#pragma unroll
+        for (int dblk = 0; dblk < 4; ++dblk)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
+            myq[qr * DH + dblk * 32 + col] = (bf16_t)0;
+          }
+      }
+      __syncthreads();
+      if (MODE != 6) {
+        for (int i = tid; i < 32 * DH; i += 256) {
+          float vsum = bf2f(dqred[i]) + bf2f(dqred[32 * DH + i]) +
+                       bf2f(dqred[2 * 32 * DH + i]) + bf2f(dqred[3 * 32 * DH + i]);
+          if (vsum != 0.f) atomicAdd(&dQb[q0 * DH + i], vsum);
+        }
+      }
+    } else {
+      __syncthreads();
+    }
+
+    // ---- phase 2 (or2): C = [q regs][kv lanes] -> P2, dS2 -> dV, dK.
+    // s2 = mfma(Q, K): rows = q tile, cols = this wave's 32 kv.
+    if (live && MODE != 2) {
+      bf16frag qrow[8], dorow[8];
+      if (MODE != 5) {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          qrow[c] = *reinterpret_cast<const bf16frag*>(Qb + (q0 + col) * DH + c * 16 + half * 8);
+          dorow[c] = *reinterpret_cast<const bf16frag*>(dOb + (q0 + col) * DH + c * 16 + half * 8);
+        }
+      } else {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          uint4 z{(uint32_t)(c + col), (uint32_t)half, 3u, 4u};
+          qrow[c] = __builtin_bit_cast(bf16frag, z);
+          dorow[c] = qrow[c];
+        }
+      }
       f32x16 s2 = f32x16{}, dp2 = f32x16{};
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
@@ -605,6 +680,13 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
         dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dorow[c], vf, dp2, 0, 0, 0);
       }
       uint32_t pk2[8], dg2[8];
+      if (MODE == 3) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          pk2[i] = __builtin_bit_cast(uint32_t, s2[i]);
+          dg2[i] = __builtin_bit_cast(uint32_t, dp2[i]);
+        }
+      } else
 #pragma unroll
       for (int r = 0; r < 16; r += 2) {
         float pv[2], gv[2];
@@ -658,11 +740,6 @@ __global__ __launch_bounds__(256, 1) void k_attn_bwd(
       }
     }
     __syncthreads();
-    for (int i = tid; i < 32 * DH; i += 256) {
-      float vsum = dqred[i];
-      if (vsum != 0.f) atomicAdd(&dQb[q0 * DH + i], vsum);
-    }
-    __syncthreads();
   }
 
   // ---- epilogue: dK/dV bf16 for this wave's kv rows (C layout: kv in regs,
@@ -703,13 +780,46 @@ extern "C" int vh_attn_bwd_bf16(const uint16_t* Q, const uint16_t* K,
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
-  hipLaunchKernelGGL(k_attn_bwd, grid, dim3(256), 131328, s,
+  hipLaunchKernelGGL(k_attn_bwd<0>, grid, dim3(256), 147712, s,
                      reinterpret_cast<const bf16_t*>(Q),
                      reinterpret_cast<const bf16_t*>(K),
                      reinterpret_cast<const bf16_t*>(V),
                      reinterpret_cast<const bf16_t*>(dO), delta, lse2, dQacc,
                      reinterpret_cast<bf16_t*>(dK),
                      reinterpret_cast<bf16_t*>(dV), B, Hq, Hkv, S, scale);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+extern "C" int vh_attn_bwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
+                                      const uint16_t* V, const uint16_t* dO,
+                                      const float* delta, const float* lse2,
+                                      float* dQacc, uint16_t* dK, uint16_t* dV,
+                                      int B, int Hq, int Hkv, int64_t S,
+                                      float scale, int mode, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
+#define VH_BWD_LAUNCH(M)                                                      \
+  hipLaunchKernelGGL(k_attn_bwd<M>, grid, dim3(256), 147712, s,               \
+                     reinterpret_cast<const bf16_t*>(Q),                      \
+                     reinterpret_cast<const bf16_t*>(K),                      \
+                     reinterpret_cast<const bf16_t*>(V),                      \
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2, dQacc, \
+                     reinterpret_cast<bf16_t*>(dK),                           \
+                     reinterpret_cast<bf16_t*>(dV), B, Hq, Hkv, S, scale)
+  switch (mode) {
+    case 0: VH_BWD_LAUNCH(0); break;
+    case 1: VH_BWD_LAUNCH(1); break;
+    case 2: VH_BWD_LAUNCH(2); break;
+    case 3: VH_BWD_LAUNCH(3); break;
+    case 4: VH_BWD_LAUNCH(4); break;
+    case 5: VH_BWD_LAUNCH(5); break;
+    case 6: VH_BWD_LAUNCH(6); break;
+    case 7: VH_BWD_LAUNCH(7); break;
+    case 8: VH_BWD_LAUNCH(8); break;
+    default: return 1;
+  }
+#undef VH_BWD_LAUNCH
   VH_HIP(hipGetLastError());
   return 0;
 }
