@@ -374,8 +374,9 @@ def attn_bwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         dq = dqacc.to(torch.bfloat16)
         rep = Hq // Hkv
         if rep > 1:
-            dk = dkh.view(B, Hkv, rep, S, D).float().sum(2).to(torch.bfloat16)
-            dv = dvh.view(B, Hkv, rep, S, D).float().sum(2).to(torch.bfloat16)
+            # torch reduces bf16 sums in fp32 internally (acc_type)
+            dk = dkh.view(B, Hkv, rep, S, D).sum(2)
+            dv = dvh.view(B, Hkv, rep, S, D).sum(2)
         else:
             dk, dv = dkh, dvh
     return dq, dk, dv
