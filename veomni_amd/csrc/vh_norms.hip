@@ -134,6 +134,160 @@ extern "C" int vh_rmsnorm_fwd_bf16(const uint16_t* x, const uint16_t* w,
 // --------------------------------------------------------------- RMSNorm bwd
 // dx = rs * (g - x * rs^2/H * dot(g, x)),  g = dy * w  (fp32)
 // dw += dy * bf16(x * rs)   accumulated per-block in LDS, then one atomic pass.
+// dw accumulates in per-lane registers (CHUNKS = Hv/64 column chunks per
+// lane) across all of the wave's rows, with one global fp32 atomicAdd pass
+// at kernel end. The previous version did 8 LDS fp32 atomicAdds per bf16x8
+// per row; ds_add_f32 under 4-wave same-address contention measures ~600
+// cycles/op on gfx950 (see the attention-backward probe notes in DESIGN.md)
+// and capped the kernel at ~1 TB/s.
+template <int CHUNKS>
+__global__ __launch_bounds__(256, 2) void k_rmsnorm_bwd_reg(
+    const bf16x8* __restrict__ dy, const bf16x8* __restrict__ x,
+    const bf16x8* __restrict__ w, const float* __restrict__ rstd,
+    bf16x8* __restrict__ dx, float* __restrict__ dw_scratch, int64_t T,
+    int64_t Hv, float invH) {
+  int wave = threadIdx.x / kWave;
+  int lane = threadIdx.x & (kWave - 1);
+  int waves_per_block = blockDim.x / kWave;
+
+  float dw_acc[CHUNKS][8];
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dw_acc[k][j] = 0.f;
+
+  for (int64_t r = blockIdx.x * waves_per_block + wave; r < T;
+       r += (int64_t)gridDim.x * waves_per_block) {
+    const bf16x8* dyr = dy + r * Hv;
+    const bf16x8* xr = x + r * Hv;
+    float rs = rstd[r];
+    // the whole row in registers: 2*CHUNKS b128 loads in flight (~4 KiB of
+    // MLP per wave); reused for the dx pass, so dy/x are read from HBM once.
+    bf16x8 dv[CHUNKS], xv[CHUNKS];
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) dv[k] = dyr[lane + k * kWave];
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) xv[k] = xr[lane + k * kWave];
+    float dot = 0.f;
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) {
+      bf16x8 wv = w[lane + k * kWave];  // L2-resident, shared across waves
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf2f(dv[k].v[j]) * bf2f(wv.v[j]) * bf2f(xv[k].v[j]);
+    }
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1) dot += __shfl_down(dot, off, kWave);
+    dot = __shfl(dot, 0, kWave);
+    float kf = dot * rs * rs * invH;
+    bf16x8* dxr = dx + r * Hv;
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) {
+      bf16x8 wv = w[lane + k * kWave];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bf2f(xv[k].v[j]);
+        float g = bf2f(dv[k].v[j]) * bf2f(wv.v[j]);
+        o.v[j] = f2bf(rs * (g - xf * kf));
+        float xhat = bf2f(f2bf(xf * rs));  // the value forward multiplied by w
+        dw_acc[k][j] += bf2f(dv[k].v[j]) * xhat;
+      }
+      dxr[lane + k * kWave] = o;
+    }
+  }
+  // per-wave partial row (plain coalesced stores; a tiny reduce kernel sums
+  // them — same-address fp32 atomic chains from ~2k waves measured ~400 us)
+  int64_t wid = (int64_t)blockIdx.x * waves_per_block + wave;
+  float* out = dw_scratch + wid * (Hv * 8);
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k) {
+    float4* o4 = reinterpret_cast<float4*>(out + (lane + k * kWave) * 8);
+    o4[0] = float4{dw_acc[k][0], dw_acc[k][1], dw_acc[k][2], dw_acc[k][3]};
+    o4[1] = float4{dw_acc[k][4], dw_acc[k][5], dw_acc[k][6], dw_acc[k][7]};
+  }
+}
+
+// big-H variant (H = 4096): no row stash — dy/x re-read in the dx pass so
+// register pressure stays at ~dw_acc + a few loads (4 waves/SIMD); traffic
+// is 5 passes instead of 3 but streams at full rate.
+template <int CHUNKS>
+__global__ __launch_bounds__(256, 2) void k_rmsnorm_bwd_big(
+    const bf16x8* __restrict__ dy, const bf16x8* __restrict__ x,
+    const bf16x8* __restrict__ w, const float* __restrict__ rstd,
+    bf16x8* __restrict__ dx, float* __restrict__ dw_scratch, int64_t T,
+    int64_t Hv, float invH) {
+  int wave = threadIdx.x / kWave;
+  int lane = threadIdx.x & (kWave - 1);
+  int waves_per_block = blockDim.x / kWave;
+
+  float dw_acc[CHUNKS][8];
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dw_acc[k][j] = 0.f;
+
+  for (int64_t r = blockIdx.x * waves_per_block + wave; r < T;
+       r += (int64_t)gridDim.x * waves_per_block) {
+    const bf16x8* dyr = dy + r * Hv;
+    const bf16x8* xr = x + r * Hv;
+    float rs = rstd[r];
+    float dot = 0.f;
+#pragma unroll 4
+    for (int k = 0; k < CHUNKS; ++k) {
+      bf16x8 d = dyr[lane + k * kWave], xv = xr[lane + k * kWave];
+      bf16x8 wv = w[lane + k * kWave];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf2f(d.v[j]) * bf2f(wv.v[j]) * bf2f(xv.v[j]);
+    }
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1) dot += __shfl_down(dot, off, kWave);
+    dot = __shfl(dot, 0, kWave);
+    float kf = dot * rs * rs * invH;
+    bf16x8* dxr = dx + r * Hv;
+#pragma unroll 4
+    for (int k = 0; k < CHUNKS; ++k) {
+      bf16x8 d = dyr[lane + k * kWave], xv = xr[lane + k * kWave];
+      bf16x8 wv = w[lane + k * kWave];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bf2f(xv.v[j]);
+        float g = bf2f(d.v[j]) * bf2f(wv.v[j]);
+        o.v[j] = f2bf(rs * (g - xf * kf));
+        float xhat = bf2f(f2bf(xf * rs));
+        dw_acc[k][j] += bf2f(d.v[j]) * xhat;
+      }
+      dxr[lane + k * kWave] = o;
+    }
+  }
+  int64_t wid = (int64_t)blockIdx.x * waves_per_block + wave;
+  float* out = dw_scratch + wid * (Hv * 8);
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k) {
+    float4* o4 = reinterpret_cast<float4*>(out + (lane + k * kWave) * 8);
+    o4[0] = float4{dw_acc[k][0], dw_acc[k][1], dw_acc[k][2], dw_acc[k][3]};
+    o4[1] = float4{dw_acc[k][4], dw_acc[k][5], dw_acc[k][6], dw_acc[k][7]};
+  }
+}
+
+// dw[i] += sum over W partial rows; grid (H/256, SPLIT) with SPLIT-way
+// atomic combine (contention depth SPLIT, not W).
+__global__ void k_dw_reduce(const float* __restrict__ scratch,
+                            float* __restrict__ dw, int64_t H, int W,
+                            int split) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= H) return;
+  int per = (W + split - 1) / split;
+  int w0 = blockIdx.y * per;
+  int w1 = w0 + per > W ? W : w0 + per;
+  float acc = 0.f;
+  for (int r = w0; r < w1; ++r) acc += scratch[(int64_t)r * H + i];
+  if (acc != 0.f) atomicAdd(&dw[i], acc);
+}
+
+// generic-H fallback (LDS fp32 atomics) for shapes outside H % 512 == 0
 __global__ void k_rmsnorm_bwd(const bf16x8* __restrict__ dy,
                               const bf16x8* __restrict__ x,
                               const bf16x8* __restrict__ w,
@@ -197,12 +351,55 @@ extern "C" int vh_rmsnorm_bwd_bf16(const uint16_t* dy, const uint16_t* x,
   int blocks = (int)((T + 3) / 4);
   if (blocks > 1024) blocks = 1024;
   if (blocks < 1) blocks = 1;
-  hipLaunchKernelGGL(k_rmsnorm_bwd, dim3(blocks), dim3(256),
-                     (size_t)(H * sizeof(float)), s,
-                     reinterpret_cast<const bf16x8*>(dy),
-                     reinterpret_cast<const bf16x8*>(x),
-                     reinterpret_cast<const bf16x8*>(w), rstd,
-                     reinterpret_cast<bf16x8*>(dx), dw, T, Hv, 1.0f / (float)H);
+  int blocks_reg = blocks > 512 ? 512 : blocks;
+  int reg_chunks = 0;
+  switch (H) {
+    case 512: reg_chunks = 1; break;
+    case 1024: reg_chunks = 2; break;
+    case 2048: reg_chunks = 4; break;
+    case 4096: reg_chunks = 8; break;
+  }
+  if (reg_chunks) {
+    int waves_total = blocks_reg * 4;
+    float* scratch = nullptr;
+    VH_HIP(hipMallocAsync(&scratch, (size_t)waves_total * H * sizeof(float), s));
+#define VH_RMS_BWD_REG(C)                                                     \
+  hipLaunchKernelGGL(k_rmsnorm_bwd_reg<C>, dim3(blocks_reg), dim3(256), 0, s, \
+                     reinterpret_cast<const bf16x8*>(dy),                     \
+                     reinterpret_cast<const bf16x8*>(x),                      \
+                     reinterpret_cast<const bf16x8*>(w), rstd,                \
+                     reinterpret_cast<bf16x8*>(dx), scratch, T, Hv,           \
+                     1.0f / (float)H)
+    switch (reg_chunks) {
+      case 1: VH_RMS_BWD_REG(1); break;
+      case 2: VH_RMS_BWD_REG(2); break;
+      case 4: VH_RMS_BWD_REG(4); break;
+      case 8:
+        hipLaunchKernelGGL(k_rmsnorm_bwd_big<8>, dim3(blocks_reg), dim3(256),
+                           0, s, reinterpret_cast<const bf16x8*>(dy),
+                           reinterpret_cast<const bf16x8*>(x),
+                           reinterpret_cast<const bf16x8*>(w), rstd,
+                           reinterpret_cast<bf16x8*>(dx), scratch, T, Hv,
+                           1.0f / (float)H);
+        break;
+    }
+#undef VH_RMS_BWD_REG
+    int split = 16;
+    hipLaunchKernelGGL(k_dw_reduce, dim3((uint32_t)((H + 255) / 256), split),
+                       dim3(256), 0, s, scratch, dw, H, waves_total, split);
+    VH_HIP(hipFreeAsync(scratch, s));
+    VH_HIP(hipGetLastError());
+    return 0;
+  }
+  {
+      hipLaunchKernelGGL(k_rmsnorm_bwd, dim3(blocks), dim3(256),
+                         (size_t)(H * sizeof(float)), s,
+                         reinterpret_cast<const bf16x8*>(dy),
+                         reinterpret_cast<const bf16x8*>(x),
+                         reinterpret_cast<const bf16x8*>(w), rstd,
+                         reinterpret_cast<bf16x8*>(dx), dw, T, Hv,
+                         1.0f / (float)H);
+  }
   VH_HIP(hipGetLastError());
   return 0;
 }
