@@ -187,6 +187,15 @@ int vh_attn_bwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                      float* dQacc, uint16_t* dK, uint16_t* dV, int B, int Hq,
                      int Hkv, int64_t S, float scale, void* stream);
 
+/* Split backward (v2, the dispatched path): dk/dv kernel (block owns 128 kv
+ * rows) + dq kernel (block owns 128 q rows, dQ written once as bf16 — no
+ * atomics, no fp32 accumulator). dK/dV still per Q-head (host sums GQA). */
+int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
+                      const uint16_t* dO, const float* delta,
+                      const float* lse2, uint16_t* dQ, uint16_t* dK,
+                      uint16_t* dV, int B, int Hq, int Hkv, int64_t S,
+                      float scale, void* stream);
+
 /* ---- Fused chunked cross-entropy ---------------------------------------- */
 
 /* Per-row softmax CE over a bf16 logits chunk:

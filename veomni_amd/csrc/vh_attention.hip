@@ -823,3 +823,361 @@ extern "C" int vh_attn_bwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ============================================================================
+// Split backward (v2): two kernels, registers only, no atomics.
+//   k_attn_bwd_dkv — block owns 128 kv rows (wave = 32-kv slice), loops q
+//     tiles from the diagonal; single or2 softmax; dK/dV in registers.
+//     80 KB LDS -> 2 blocks/CU (2 waves/SIMD).
+//   k_attn_bwd_dq — block owns 128 q rows (wave = 32-q slice), loops kv
+//     tiles up to the diagonal; single or1 softmax; dQ in registers, written
+//     once as bf16 (this block is the only contributor). 24 KB LDS.
+// Scores are recomputed in both kernels (the monolithic variant's cross-wave
+// dQ reduction measured 2-4 ms of pure LDS/atomic overhead per call).
+// ============================================================================
+
+namespace {
+
+__global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
+    const float* __restrict__ delta, const float* __restrict__ lse2,
+    bf16_t* __restrict__ dK, bf16_t* __restrict__ dV, int B, int Hq, int Hkv,
+    int64_t S, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [128][128] 32 K
+  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 32768);    // [128][128] 32 K
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 65536);     // [128][32] 8 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 73728);    // [128][32] 8 K
+
+  const int kvb = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 32-row kv slice
+  const int half = lane >> 5;
+  const int col = lane & 31;
+
+  const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+  const bf16_t* Kb = K + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* Vb = V + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+  const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+  const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+  bf16_t* dKb = dK + (((int64_t)b * Hq + hq) * S) * DH;  // per-Hq; host sums
+  bf16_t* dVb = dV + (((int64_t)b * Hq + hq) * S) * DH;
+
+  const int64_t kv0 = (int64_t)kvb * 128;
+  const int kvrow_l = wave * 32 + col;
+
+  // stage K/V rows once (256 thr: 8 passes of 4 KiB)
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int o = i * 4096 + tid * 16;
+    int row = o >> 8;
+    int colb = o & 255;
+    *reinterpret_cast<bf16x8*>(
+        reinterpret_cast<char*>(krow) + row * 256 + kswz(row, colb)) =
+        *reinterpret_cast<const bf16x8*>(Kb + (kv0 + row) * DH + (colb >> 1));
+    *reinterpret_cast<bf16x8*>(
+        reinterpret_cast<char*>(vrow) + row * 256 + kswz(row, colb)) =
+        *reinterpret_cast<const bf16x8*>(Vb + (kv0 + row) * DH + (colb >> 1));
+  }
+
+  f32x16 dv_acc[4], dk_acc[4];
+#pragma unroll
+  for (int d = 0; d < 4; ++d) {
+    dv_acc[d] = f32x16{};
+    dk_acc[d] = f32x16{};
+  }
+
+  const float scale2 = scale * 1.4426950408889634f;
+  const int qt0 = (int)(kv0 / 32);
+  const int qtn = (int)(S / 32);
+
+  for (int qt = qt0; qt < qtn; ++qt) {
+    const int64_t q0 = (int64_t)qt * 32;
+    // stage Q^T / dO^T (512 units / 256 thr = 2 each)
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int unit = tid + u * 256;
+      int q = unit & 31;
+      int d0 = (unit >> 5) * 8;
+      bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0 + q) * DH + d0);
+      bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0 + q) * DH + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int row = d0 + j;
+        qtr[(row * 64 + qswz(row, q * 2)) >> 1] = vq.v[j];
+        dotr[(row * 64 + qswz(row, q * 2)) >> 1] = vd.v[j];
+      }
+    }
+    __syncthreads();
+
+    const bool live = (q0 + 31) >= (kv0 + wave * 32);
+    const bool diag = (q0 < kv0 + 128);
+
+    if (live) {
+      // or2: C = [q regs][kv lanes]; A-operands are the lane's Q/dO rows,
+      // loaded per chunk (k = q0+col) so only 8 regs are live per iteration.
+      f32x16 s2 = f32x16{}, dp2 = f32x16{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int colb = (c * 16 + half * 8) * 2;
+        bf16frag qc = *reinterpret_cast<const bf16frag*>(
+            Qb + (q0 + col) * DH + c * 16 + half * 8);
+        bf16frag dc = *reinterpret_cast<const bf16frag*>(
+            dOb + (q0 + col) * DH + c * 16 + half * 8);
+        bf16frag kf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        bf16frag vf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(vrow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+        s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qc, kf, s2, 0, 0, 0);
+        dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dc, vf, dp2, 0, 0, 0);
+      }
+      uint32_t pk2[8], dg2[8];
+#pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        float pv[2], gv[2];
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+          int r2 = r + rr;
+          int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
+          bool masked = diag && ((int64_t)kv0 + wave * 32 + col > q0 + qrm);
+          float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lseb[q0 + qrm]);
+          pv[rr] = pp;
+          gv[rr] = pp * (dp2[r2] - delb[q0 + qrm]) * scale;
+        }
+        pk2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[1]) << 16);
+        dg2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[1]) << 16);
+      }
+      bf16frag pa2[2], da2[2];
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+        uint32_t a0 = half ? pk2[4 * mch] : pk2[4 * mch + 2];
+        uint32_t a1 = half ? pk2[4 * mch + 1] : pk2[4 * mch + 3];
+        uint32_t b0 = swap32_u(a0, half);
+        uint32_t b1 = swap32_u(a1, half);
+        uint4 u{half ? b0 : pk2[4 * mch], half ? b1 : pk2[4 * mch + 1],
+                half ? pk2[4 * mch + 2] : b0, half ? pk2[4 * mch + 3] : b1};
+        pa2[mch] = __builtin_bit_cast(bf16frag, u);
+        uint32_t c0 = half ? dg2[4 * mch] : dg2[4 * mch + 2];
+        uint32_t c1 = half ? dg2[4 * mch + 1] : dg2[4 * mch + 3];
+        uint32_t e0 = swap32_u(c0, half);
+        uint32_t e1 = swap32_u(c1, half);
+        uint4 u2{half ? e0 : dg2[4 * mch], half ? e1 : dg2[4 * mch + 1],
+                 half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
+        da2[mch] = __builtin_bit_cast(bf16frag, u2);
+      }
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+#pragma unroll
+        for (int dblk = 0; dblk < 4; ++dblk) {
+          int trow = dblk * 32 + col;
+          int colb = (mch * 16 + half * 8) * 2;
+          bf16frag dof = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(dotr) + trow * 64 + qswz(trow, colb));
+          bf16frag qf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(qtr) + trow * 64 + qswz(trow, colb));
+          dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
+          dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int kvr = (r & 3) + 8 * (r >> 2) + 4 * half;
+    int64_t kvg = kv0 + wave * 32 + kvr;
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      dKb[kvg * DH + d * 32 + col] = f2bf(dk_acc[d][r]);
+      dVb[kvg * DH + d * 32 + col] = f2bf(dv_acc[d][r]);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
+    const float* __restrict__ delta, const float* __restrict__ lse2,
+    bf16_t* __restrict__ dQ, int B, int Hq, int Hkv, int64_t S, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [32][128] 8 K
+  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 8192);     // [32][128] 8 K
+  bf16_t* ktr = reinterpret_cast<bf16_t*>(smem + 16384);     // [128][32] 8 K
+
+  const int qb = blockIdx.x;           // q block of 128 rows
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 32-row q slice
+  const int half = lane >> 5;
+  const int col = lane & 31;
+
+  const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+  const bf16_t* Kb = K + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* Vb = V + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+  const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+  const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+  bf16_t* dQb = dQ + (((int64_t)b * Hq + hq) * S) * DH;
+
+  const int64_t q0b = (int64_t)qb * 128;
+  const int64_t q_l = q0b + wave * 32 + col;   // this lane's q row
+
+  // the wave's Q/dO rows and lse/delta are fixed for the whole kernel
+  bf16frag qrow[8], dorow[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    qrow[c] = *reinterpret_cast<const bf16frag*>(Qb + q_l * DH + c * 16 + half * 8);
+    dorow[c] = *reinterpret_cast<const bf16frag*>(dOb + q_l * DH + c * 16 + half * 8);
+  }
+  const float lse_l = lseb[q_l];
+  const float del_l = delb[q_l];
+  const float scale2 = scale * 1.4426950408889634f;
+
+  f32x16 dq4[4];
+#pragma unroll
+  for (int d = 0; d < 4; ++d) dq4[d] = f32x16{};
+
+  const int kvtn = (int)((q0b + 128) / 32);
+  for (int kvt = 0; kvt < kvtn; ++kvt) {
+    const int64_t kvt0 = (int64_t)kvt * 32;
+    // stage K/V rows [32][128] (2 passes) + K^T [128][32] (2 units each)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int o = i * 4096 + tid * 16;
+      int row = o >> 8;
+      int colb = o & 255;
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(krow) + row * 256 + kswz(row, colb)) =
+          *reinterpret_cast<const bf16x8*>(Kb + (kvt0 + row) * DH + (colb >> 1));
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(vrow) + row * 256 + kswz(row, colb)) =
+          *reinterpret_cast<const bf16x8*>(Vb + (kvt0 + row) * DH + (colb >> 1));
+    }
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int unit = tid + u * 256;
+      int kv = unit & 31;
+      int d0 = (unit >> 5) * 8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(Kb + (kvt0 + kv) * DH + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int row = d0 + j;
+        ktr[(row * 64 + qswz(row, kv * 2)) >> 1] = v.v[j];
+      }
+    }
+    __syncthreads();
+
+    const bool live = kvt0 <= q0b + wave * 32 + 31;
+    const bool diag = (kvt0 + 31 >= q0b + wave * 32);
+
+    if (live) {
+      // or1: C = [kv regs][q lanes]; s1 = mfma(K, Q)
+      f32x16 s1 = f32x16{}, dp1 = f32x16{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int colb = (c * 16 + half * 8) * 2;
+        bf16frag kf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(krow) + col * 256 + kswz(col, colb));
+        bf16frag vf = *reinterpret_cast<const bf16frag*>(
+            reinterpret_cast<const char*>(vrow) + col * 256 + kswz(col, colb));
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qrow[c], s1, 0, 0, 0);
+        dp1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dorow[c], dp1, 0, 0, 0);
+      }
+      uint32_t dg1[8];
+#pragma unroll
+      for (int r = 0; r < 16; r += 2) {
+        float g[2];
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+          int r2 = r + rr;
+          int kvl = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
+          bool masked = diag && (kvt0 + kvl > q_l);
+          float pp = masked ? 0.f : __builtin_exp2f(s1[r2] * scale2 - lse_l);
+          g[rr] = pp * (dp1[r2] - del_l) * scale;
+        }
+        dg1[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)g[0]) |
+                      ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)g[1]) << 16);
+      }
+      bf16frag da1[2];
+#pragma unroll
+      for (int mch = 0; mch < 2; ++mch) {
+        uint32_t a0 = half ? dg1[4 * mch] : dg1[4 * mch + 2];
+        uint32_t a1 = half ? dg1[4 * mch + 1] : dg1[4 * mch + 3];
+        uint32_t b0 = swap32_u(a0, half);
+        uint32_t b1 = swap32_u(a1, half);
+        uint4 u{half ? b0 : dg1[4 * mch], half ? b1 : dg1[4 * mch + 1],
+                half ? dg1[4 * mch + 2] : b0, half ? dg1[4 * mch + 3] : b1};
+        da1[mch] = __builtin_bit_cast(bf16frag, u);
+      }
+      // dQ[q][d] += dS1^T(pack) x K^T-tile
+#pragma unroll
+      for (int dblk = 0; dblk < 4; ++dblk) {
+#pragma unroll
+        for (int mch = 0; mch < 2; ++mch) {
+          int trow = dblk * 32 + col;
+          int colb = (mch * 16 + half * 8) * 2;
+          bf16frag ktf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(ktr) + trow * 64 + qswz(trow, colb));
+          dq4[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1[mch], ktf, dq4[dblk], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // single-contributor store: this block covers every kv for its q rows
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
+    int64_t qg = q0b + wave * 32 + qr;
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      dQb[qg * DH + d * 32 + col] = f2bf(dq4[d][r]);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
+                                 const uint16_t* V, const uint16_t* dO,
+                                 const float* delta, const float* lse2,
+                                 uint16_t* dQ, uint16_t* dK, uint16_t* dV,
+                                 int B, int Hq, int Hkv, int64_t S, float scale,
+                                 void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(S % 128 == 0, "S %% 128 != 0");
+  dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
+  hipLaunchKernelGGL(k_attn_bwd_dkv, grid, dim3(256), 81920, s,
+                     reinterpret_cast<const bf16_t*>(Q),
+                     reinterpret_cast<const bf16_t*>(K),
+                     reinterpret_cast<const bf16_t*>(V),
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                     reinterpret_cast<bf16_t*>(dK),
+                     reinterpret_cast<bf16_t*>(dV), B, Hq, Hkv, S, scale);
+  VH_HIP(hipGetLastError());
+  hipLaunchKernelGGL(k_attn_bwd_dq, grid, dim3(256), 24576, s,
+                     reinterpret_cast<const bf16_t*>(Q),
+                     reinterpret_cast<const bf16_t*>(K),
+                     reinterpret_cast<const bf16_t*>(V),
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                     reinterpret_cast<bf16_t*>(dQ), B, Hq, Hkv, S, scale);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

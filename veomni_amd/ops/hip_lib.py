@@ -71,6 +71,8 @@ def get_lib() -> ctypes.CDLL:
     _sig(lib, "vh_attn_bwd_pre_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_attn_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p,
          c_int, c_int, c_int, c_i64, c_f32, c_p)
+    _sig(lib, "vh_attn_bwd2_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p,
+         c_int, c_int, c_int, c_i64, c_f32, c_p)
     _LIB = lib
     return lib
 
@@ -320,10 +322,11 @@ def silu_mul_bwd(dy: torch.Tensor, gate: torch.Tensor, up: torch.Tensor):
 
 
 def ce_fwd(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float,
-           ignore_index: int = -100):
+           ignore_index: int = -100, dlogits_out: torch.Tensor | None = None):
     rows, V = logits.shape
     loss_rows = torch.zeros(rows, dtype=torch.float32, device=logits.device)
-    dlogits = torch.empty_like(logits)
+    dlogits = dlogits_out if dlogits_out is not None else torch.empty_like(logits)
+    assert dlogits.is_contiguous() and dlogits.shape == logits.shape
     with _prof("ce_fwd", 3.0 * rows * V * 2):
         check(get_lib().vh_ce_fwd_bf16(dptr(logits.contiguous()),
                                        dptr(labels.to(torch.int64).contiguous()),
@@ -364,14 +367,13 @@ def attn_bwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         check(lib.vh_attn_bwd_pre_bf16(dptr(do), dptr(o), dptr(lse.contiguous()),
                                        dptr(delta), dptr(lse2), rows,
                                        cur_stream()), "vh_attn_bwd_pre")
-        dqacc = torch.zeros(B, Hq, S, D, dtype=torch.float32, device=q.device)
+        dq = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
         dkh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
         dvh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
-        check(lib.vh_attn_bwd_bf16(dptr(q), dptr(k), dptr(v), dptr(do),
-                                   dptr(delta), dptr(lse2), dptr(dqacc),
-                                   dptr(dkh), dptr(dvh), B, Hq, Hkv, S, scale,
-                                   cur_stream()), "vh_attn_bwd")
-        dq = dqacc.to(torch.bfloat16)
+        check(lib.vh_attn_bwd2_bf16(dptr(q), dptr(k), dptr(v), dptr(do),
+                                    dptr(delta), dptr(lse2), dptr(dq),
+                                    dptr(dkh), dptr(dvh), B, Hq, Hkv, S, scale,
+                                    cur_stream()), "vh_attn_bwd2")
         rep = Hq // Hkv
         if rep > 1:
             # torch reduces bf16 sums in fp32 internally (acc_type)
