@@ -11,11 +11,13 @@ from veomni_amd.data import synthetic_batch
 
 init_parallel_state(ep_size=1, device_type="cuda")
 bind_ops(HIP_OPS_CONFIG)
-model = build_model("llama3-8b", dtype=torch.bfloat16, device="cuda")
-model.use_checkpoint = False
+import sys as _s
+preset = _s.argv[1] if len(_s.argv) > 1 else "llama3-8b"
+model = build_model(preset, dtype=torch.bfloat16, device="cuda")
 model = build_parallelize_model(model)
 opt = torch.optim.AdamW(model.parameters(), lr=1e-5, fused=True)
-batch = synthetic_batch(model.config.vocab_size, 4096, batch=4, seed=42, device="cuda")
+batch = synthetic_batch(model.config.vocab_size, 4096, batch=(4 if preset=="llama3-8b" else 1), seed=42, device="cuda")
+model.use_checkpoint = model.config.is_moe
 def step():
     loss, _ = model(**batch)
     loss.backward()

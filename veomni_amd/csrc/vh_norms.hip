@@ -272,6 +272,71 @@ __global__ __launch_bounds__(256, 2) void k_rmsnorm_bwd_big(
   }
 }
 
+// head-dim-sized rows (H = 128, the per-head q/k norms): 16-lane groups, a
+// wave covers 4 rows per iteration; dw per lane column with a cross-subgroup
+// shuffle reduce before the scratch store.
+__global__ __launch_bounds__(256, 4) void k_rmsnorm_bwd_small128(
+    const bf16x8* __restrict__ dy, const bf16x8* __restrict__ x,
+    const bf16x8* __restrict__ w, const float* __restrict__ rstd,
+    bf16x8* __restrict__ dx, float* __restrict__ dw_scratch, int64_t T,
+    float invH) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+  const int sub = lane >> 4;   // row within the wave's quad
+  const int cl = lane & 15;    // bf16x8 column
+  const int waves_per_block = blockDim.x / kWave;
+  const int64_t gw = (int64_t)blockIdx.x * waves_per_block + wave;
+  const int64_t nw = (int64_t)gridDim.x * waves_per_block;
+
+  const bf16x8 wv = w[cl];
+  float dw_acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) dw_acc[j] = 0.f;
+
+  for (int64_t r0 = gw * 4; r0 < T; r0 += nw * 4) {
+    int64_t r = r0 + sub;
+    bool ok = r < T;
+    bf16x8 d = {}, xv = {};
+    float rs = 0.f;
+    if (ok) {
+      d = dy[r * 16 + cl];
+      xv = x[r * 16 + cl];
+      rs = rstd[r];
+    }
+    float dot = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dot += bf2f(d.v[j]) * bf2f(wv.v[j]) * bf2f(xv.v[j]);
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) dot += __shfl_xor(dot, off, kWave);
+    float kf = dot * rs * rs * invH;
+    if (ok) {
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = bf2f(xv.v[j]);
+        float g = bf2f(d.v[j]) * bf2f(wv.v[j]);
+        o.v[j] = f2bf(rs * (g - xf * kf));
+        float xhat = bf2f(f2bf(xf * rs));
+        dw_acc[j] += bf2f(d.v[j]) * xhat;
+      }
+      dx[r * 16 + cl] = o;
+    }
+  }
+  // combine the 4 row-subgroups (same columns) before storing
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    dw_acc[j] += __shfl_xor(dw_acc[j], 16, kWave);
+    dw_acc[j] += __shfl_xor(dw_acc[j], 32, kWave);
+  }
+  if (sub == 0) {
+    float* out = dw_scratch + gw * 128;
+    float4* o4 = reinterpret_cast<float4*>(out + cl * 8);
+    o4[0] = float4{dw_acc[0], dw_acc[1], dw_acc[2], dw_acc[3]};
+    o4[1] = float4{dw_acc[4], dw_acc[5], dw_acc[6], dw_acc[7]};
+  }
+}
+
 // dw[i] += sum over W partial rows; grid (H/256, SPLIT) with SPLIT-way
 // atomic combine (contention depth SPLIT, not W).
 __global__ void k_dw_reduce(const float* __restrict__ scratch,
@@ -352,6 +417,25 @@ extern "C" int vh_rmsnorm_bwd_bf16(const uint16_t* dy, const uint16_t* x,
   if (blocks > 1024) blocks = 1024;
   if (blocks < 1) blocks = 1;
   int blocks_reg = blocks > 512 ? 512 : blocks;
+  if (H == 128) {
+    int blocks_s = (int)((T + 15) / 16);
+    if (blocks_s > 1024) blocks_s = 1024;
+    if (blocks_s < 1) blocks_s = 1;
+    int waves_total = blocks_s * 4;
+    float* scratch = nullptr;
+    VH_HIP(hipMallocAsync(&scratch, (size_t)waves_total * 128 * sizeof(float), s));
+    hipLaunchKernelGGL(k_rmsnorm_bwd_small128, dim3(blocks_s), dim3(256), 0, s,
+                       reinterpret_cast<const bf16x8*>(dy),
+                       reinterpret_cast<const bf16x8*>(x),
+                       reinterpret_cast<const bf16x8*>(w), rstd,
+                       reinterpret_cast<bf16x8*>(dx), scratch, T,
+                       1.0f / 128.0f);
+    hipLaunchKernelGGL(k_dw_reduce, dim3(1, 16), dim3(128), 0, s, scratch, dw,
+                       128, waves_total, 16);
+    VH_HIP(hipFreeAsync(scratch, s));
+    VH_HIP(hipGetLastError());
+    return 0;
+  }
   int reg_chunks = 0;
   switch (H) {
     case 512: reg_chunks = 1; break;
