@@ -170,10 +170,19 @@ class Attention(nn.Module):
 
     def forward(self, hidden_states, position_embeddings, **kwargs):
         input_shape = hidden_states.shape[:-1]
-        hidden_shape = (*input_shape, -1, self.head_dim)
-        q = self.q_proj(hidden_states).view(hidden_shape)
-        k = self.k_proj(hidden_states).view(hidden_shape)
-        v = self.v_proj(hidden_states).view(hidden_shape)
+        # one fused qkv GEMM (weights concatenated at call time, so the
+        # per-projection parameters/state-dict stay reference-shaped); the
+        # per-element math is identical to three separate linears.
+        wqkv = torch.cat((self.q_proj.weight, self.k_proj.weight, self.v_proj.weight))
+        bqkv = None
+        if self.q_proj.bias is not None:
+            bqkv = torch.cat((self.q_proj.bias, self.k_proj.bias, self.v_proj.bias))
+        qkv = nn.functional.linear(hidden_states, wqkv, bqkv)
+        hv = qkv.view(*input_shape, -1, self.head_dim)
+        nh, nkv = self.num_heads, self.num_key_value_heads
+        q = hv[..., :nh, :]
+        k = hv[..., nh:nh + nkv, :]
+        v = hv[..., nh + nkv:, :]
         if self.q_norm is not None:
             q = self.q_norm(q)
             k = self.k_norm(k)

@@ -71,8 +71,32 @@ class HipSiluMul(torch.autograd.Function):
         return dg, du
 
 
+class HipFc1SiluMul(torch.autograd.Function):
+    """silu(fc1[:, :I]) * fc1[:, I:] on the fused gate|up projection output
+    (vh_moe_silu_mul_weighted with has_w=0) — one pass over [T, 2I]."""
+
+    @staticmethod
+    def forward(ctx, fc1):
+        assert fc1.dtype == torch.bfloat16
+        flat = fc1.reshape(-1, fc1.shape[-1]).contiguous()
+        out = hip_lib.silu_mul_weighted(flat, None)
+        ctx.save_for_backward(flat)
+        return out.view(*fc1.shape[:-1], fc1.shape[-1] // 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (flat,) = ctx.saved_tensors
+        dfc1, _ = hip_lib.silu_mul_weighted_bwd(
+            dy.reshape(-1, dy.shape[-1]), flat, None)
+        return dfc1.view(*dy.shape[:-1], dy.shape[-1] * 2)
+
+
 def hip_swiglu_mlp(self, x):
-    return self.down_proj(HipSiluMul.apply(self.gate_proj(x), self.up_proj(x)))
+    # single fused gate|up GEMM (one wide hipBLASLt call instead of two) and
+    # a one-pass SiLU-mul epilogue over the fused fc1
+    w12 = torch.cat((self.gate_proj.weight, self.up_proj.weight))
+    fc1 = torch.nn.functional.linear(x, w12)
+    return self.down_proj(HipFc1SiluMul.apply(fc1))
 
 
 for op_name, variant, fn, desc in [
