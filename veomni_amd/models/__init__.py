@@ -47,10 +47,27 @@ PRESETS = {
 }
 
 
-def build_model(preset: str, dtype=None, device=None) -> ForCausalLM:
+def build_model(preset: str, dtype=None, device=None, empty_init=False) -> ForCausalLM:
     import torch
 
     cfg = PRESETS[preset]
+    if empty_init:
+        # timing-only instantiation (bench cpu_baseline): meta-build +
+        # to_empty skips the (minutes-long on host, for 30B) random init;
+        # weights are filled with a small constant so the step is
+        # numerically tame but costs exactly the same FLOPs.
+        with torch.device("meta"):
+            model = ForCausalLM(cfg)
+        model = model.to_empty(device=device or "cpu")
+        with torch.no_grad():
+            for p in model.parameters():
+                p.fill_(0.01)
+            for b in model.buffers():
+                if b.is_floating_point():
+                    b.fill_(0.01)
+        if dtype is not None:
+            model = model.to(dtype)
+        return model
     with torch.device(device) if device is not None else torch.device("cpu"):
         model = ForCausalLM(cfg)
     if dtype is not None:

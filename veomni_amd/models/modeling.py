@@ -450,6 +450,8 @@ class ForCausalLM(nn.Module):
             dev = next(self.parameters()).device
         except StopIteration:
             return
+        if dev.type == "meta":
+            return  # empty-init build (bench cpu_baseline): filled later
         g = torch.Generator(device=dev).manual_seed(seed)
         for name, p in sorted(self.named_parameters(), key=lambda kv: kv[0]):
             if name.endswith("layernorm.weight") or name.endswith("norm.weight") or ".q_norm" in name or ".k_norm" in name:
