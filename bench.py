@@ -147,16 +147,19 @@ def main():
         return torch.optim.AdamW(model.parameters(), lr=1e-5, betas=(0.9, 0.95),
                                  fused=fused, foreach=None if fused else True)
 
+    fused_ok = True
     try:  # fused adamw: one kernel sweep instead of ~7 foreach passes
         opt = make_opt(True)
         for p in model.parameters():
             if p.requires_grad:
                 p.grad = torch.zeros_like(p)
+        opt.grad_scale = torch.ones((), dtype=torch.float32, device="cuda")
         opt.step()
         opt.zero_grad(set_to_none=True)
     except Exception as e:
         log(f"fused AdamW unavailable ({e}); using foreach")
         opt = make_opt(False)
+        fused_ok = False
     log(f"built in {time.time() - t_build:.1f}s; mem {torch.cuda.memory_allocated()/2**30:.1f} GiB")
 
     seq = args.seq_len
@@ -165,7 +168,8 @@ def main():
     def one_step():
         loss, _ = model(**batch)
         loss.backward()
-        model.clip_grad_norm_(1.0)
+        # fused path folds the clip coefficient into AdamW's grad_scale
+        model.clip_grad_norm_(1.0, fused_optimizer=opt if fused_ok else None)
         opt.step()
         opt.zero_grad(set_to_none=True)
         return loss

@@ -95,6 +95,11 @@ int vh_group_gemm_mn_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                           const int64_t* cumsum, int G, int64_t M, int64_t N,
                           void* stream);
 
+/* Batched expert-weight transpose [E, M, N] -> [E, N, M] bf16 (dgrad W^T;
+ * M, N 64-multiples). */
+int vh_wtranspose_bf16(const uint16_t* src, uint16_t* dst, int E, int64_t M,
+                       int64_t N, void* stream);
+
 /* Transpose-pad: [rows, C] (rows grouped by cumsum) -> [C, padded_total]
  * with each group's rows zero-padded to a 64-multiple (padded_cumsum,
  * host-computed ceil64 cumsum). Regions past the last group are untouched. */

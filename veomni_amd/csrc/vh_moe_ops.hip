@@ -218,3 +218,47 @@ extern "C" int vh_moe_silu_mul_weighted_bwd_bf16(
   VH_HIP(hipGetLastError());
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Batched expert-weight transpose: [E, M, N] -> [E, N, M] bf16.
+// The dgrad path needs W^T per step; torch's permuted-copy ran at ~1.4 TB/s
+// on [128, 1536, 2048] (strided writes). [64][72] LDS tile, both sides b128.
+// M % 64 == 0 and N % 64 == 0 (production shapes; host falls back otherwise).
+// ---------------------------------------------------------------------------
+namespace {
+__global__ void k_wtranspose(const bf16_t* __restrict__ src,
+                             bf16_t* __restrict__ dst, int64_t M, int64_t N) {
+  __shared__ __attribute__((aligned(16))) bf16_t t2[64][72];
+  const int64_t e = blockIdx.z;
+  const int64_t m0 = (int64_t)blockIdx.x * 64;
+  const int64_t n0 = (int64_t)blockIdx.y * 64;
+  const bf16_t* s = src + e * M * N;
+  bf16_t* d = dst + e * M * N;
+  const int tid = threadIdx.x;
+  for (int r = tid / 8; r < 64; r += 32) {
+    const int cpos = (tid % 8) * 8;
+    *reinterpret_cast<bf16x8*>(&t2[r][cpos]) =
+        *reinterpret_cast<const bf16x8*>(s + (m0 + r) * N + n0 + cpos);
+  }
+  __syncthreads();
+  for (int c = tid / 8; c < 64; c += 32) {
+    const int rpos = (tid % 8) * 8;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v.v[j] = t2[rpos + j][c];
+    *reinterpret_cast<bf16x8*>(d + (n0 + c) * M + m0 + rpos) = v;
+  }
+}
+}  // namespace
+
+extern "C" int vh_wtranspose_bf16(const uint16_t* src, uint16_t* dst, int E,
+                                  int64_t M, int64_t N, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(M % 64 == 0 && N % 64 == 0, "wtranspose needs 64-multiples");
+  dim3 grid((uint32_t)(M / 64), (uint32_t)(N / 64), (uint32_t)E);
+  hipLaunchKernelGGL(k_wtranspose, grid, dim3(256), 0, s,
+                     reinterpret_cast<const bf16_t*>(src),
+                     reinterpret_cast<bf16_t*>(dst), M, N);
+  VH_HIP(hipGetLastError());
+  return 0;
+}

@@ -138,7 +138,8 @@ def _local_pth_sum(params, norm_type: float) -> torch.Tensor:
 
 @torch.no_grad()
 def clip_grad_norm(model: nn.Module, max_norm: float, norm_type: float = 2.0,
-                   error_if_nonfinite: bool = False, foreach: Optional[bool] = None) -> torch.Tensor:
+                   error_if_nonfinite: bool = False, foreach: Optional[bool] = None,
+                   fused_optimizer: Optional[torch.optim.Optimizer] = None) -> torch.Tensor:
     """EP-aware global grad-norm clip (ref fsdp2/clip_grad_norm.py:86-226):
     dense p-th-power sums all-reduced over the fsdp shard group; EP sums over
     ep_fsdp then ep groups; one global clip coefficient."""
@@ -161,6 +162,15 @@ def clip_grad_norm(model: nn.Module, max_norm: float, norm_type: float = 2.0,
     if error_if_nonfinite and (torch.isnan(total_norm) or torch.isinf(total_norm)):
         raise RuntimeError("non-finite grad norm")
 
+    if fused_optimizer is not None and norm_type == 2.0:
+        # fold the clip into the fused optimizer: torch's fused AdamW divides
+        # grads by `optimizer.grad_scale` in-register (AMP-unscale plumbing,
+        # optim/adam.py:267), so the explicit _foreach_mul_ over every grad
+        # (~120 GB of traffic on a 30B model) disappears. Exact same math:
+        # grad * coef == grad / max(1, (norm + eps) / max_norm). No host sync.
+        fused_optimizer.grad_scale = torch.clamp(
+            (total_norm + 1e-6) / max_norm, min=1.0).to(torch.float32)
+        return total_norm
     clip_coef = max_norm / (total_norm + 1e-6)
     clip_coef = torch.clamp(clip_coef, max=1.0)
     grads = []

@@ -54,6 +54,7 @@ def get_lib() -> ctypes.CDLL:
          c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_mn_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
     _sig(lib, "vh_transpose_pad_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
+    _sig(lib, "vh_wtranspose_bf16", c_p, c_p, c_int, c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_wg256_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
          c_i64, c_i64, c_p)
     _sig(lib, "vh_moe_silu_mul_weighted_bf16", c_p, c_p, c_p, c_i64, c_i64, c_int, c_p)
@@ -238,6 +239,19 @@ def group_gemm_mn(a: torch.Tensor, b: torch.Tensor, cumsum: torch.Tensor,
             dptr(a.contiguous()), dptr(b.contiguous()), dptr(c), dptr(cs), G, M, N,
             cur_stream()), "vh_group_gemm_mn")
     return c
+
+
+def weight_transpose(b: torch.Tensor) -> torch.Tensor:
+    """[E, M, N] -> [E, N, M] contiguous (dgrad W^T). HIP tile kernel for
+    64-multiple shapes; torch copy otherwise."""
+    E, M, N = b.shape
+    if M % 64 or N % 64 or b.dtype != torch.bfloat16:
+        return b.transpose(1, 2).contiguous()
+    out = torch.empty(E, N, M, dtype=b.dtype, device=b.device)
+    with _prof("wtranspose", 2.0 * E * M * N * 2):
+        check(get_lib().vh_wtranspose_bf16(dptr(b.contiguous()), dptr(out), E,
+                                           M, N, cur_stream()), "vh_wtranspose")
+    return out
 
 
 def silu_mul_weighted(fc1: torch.Tensor, w_row: torch.Tensor | None) -> torch.Tensor:

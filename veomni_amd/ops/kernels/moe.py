@@ -77,7 +77,7 @@ class HipFusedMoeFunction(torch.autograd.Function):
         # once so dgrad runs the fast trans_b kernel (HBM copy ~50 us vs
         # ~2x slower K-strided dgrad staging; profiles/r01_groupgemm_microbench)
         d_weighted = hip_lib.group_gemm_nk(
-            grad_fc2_out, fc2_weight.transpose(1, 2).contiguous(), cumsum_t, trans_b=True)
+            grad_fc2_out, hip_lib.weight_transpose(fc2_weight), cumsum_t, trans_b=True)
         # wgrad fc2: [G, H, I]
         d_fc2_w = hip_lib.group_gemm_mn(grad_fc2_out, weighted, cumsum_t, G)
 
@@ -88,7 +88,7 @@ class HipFusedMoeFunction(torch.autograd.Function):
 
         # dgrad + wgrad through merged fc1 (same weight-transpose trick)
         d_scatter = hip_lib.group_gemm_nk(
-            d_fc1, fc1_1_2_weight.transpose(1, 2).contiguous(), cumsum_t, trans_b=True)
+            d_fc1, hip_lib.weight_transpose(fc1_1_2_weight), cumsum_t, trans_b=True)
         d_fc1_w = hip_lib.group_gemm_mn(d_fc1, scatter_output, cumsum_t, G)
 
         grad_hidden = hip_lib.moe_gather(d_scatter, scatter_index)
@@ -113,11 +113,11 @@ class EPMergedFc1HipGroupGemm(torch.autograd.Function):
         G = fc2_weight.shape[0]
         grad_output = grad_output.contiguous()
         d_act = hip_lib.group_gemm_nk(
-            grad_output, fc2_weight.transpose(1, 2).contiguous(), cumsum, trans_b=True)
+            grad_output, hip_lib.weight_transpose(fc2_weight), cumsum, trans_b=True)
         d_fc2_w = hip_lib.group_gemm_mn(grad_output, act, cumsum, G)
         d_fc1, _ = hip_lib.silu_mul_weighted_bwd(d_act, fc1, None)
         d_tokens = hip_lib.group_gemm_nk(
-            d_fc1, fc1_1_2_weight.transpose(1, 2).contiguous(), cumsum, trans_b=True)
+            d_fc1, hip_lib.weight_transpose(fc1_1_2_weight), cumsum, trans_b=True)
         d_fc1_w = hip_lib.group_gemm_mn(d_fc1, permute_tokens, cumsum, G)
         return d_tokens, None, d_fc1_w, d_fc2_w
 
