@@ -145,22 +145,33 @@ def main():
     # expert activations are ~1 GB/layer/rank).
     model.use_checkpoint = cfg.is_moe
     model = build_parallelize_model(model)
-    def make_opt(fused):
+    def make_opt(kind):
+        if kind == "ve":
+            from veomni_amd.optim import VeAdamW
+            return VeAdamW(model.parameters(), lr=1e-5, betas=(0.9, 0.95))
         return torch.optim.AdamW(model.parameters(), lr=1e-5, betas=(0.9, 0.95),
-                                 fused=fused, foreach=None if fused else True)
+                                 fused=kind == "fused",
+                                 foreach=None if kind == "fused" else True)
 
     fused_ok = True
-    try:  # fused adamw: one kernel sweep instead of ~7 foreach passes
-        opt = make_opt(True)
-        for p in model.parameters():
-            if p.requires_grad:
-                p.grad = torch.zeros_like(p)
-        opt.grad_scale = torch.ones((), dtype=torch.float32, device="cuda")
-        opt.step()
-        opt.zero_grad(set_to_none=True)
-    except Exception as e:
-        log(f"fused AdamW unavailable ({e}); using foreach")
-        opt = make_opt(False)
+    opt = None
+    for kind in ("ve", "fused", "foreach"):
+        try:
+            opt = make_opt(kind)
+            for p in model.parameters():
+                if p.requires_grad:
+                    p.grad = torch.zeros_like(p)
+            opt.grad_scale = torch.ones((), dtype=torch.float32, device="cuda")
+            opt.step()
+            opt.zero_grad(set_to_none=True)
+            fused_ok = kind in ("ve", "fused")  # these honor grad_scale
+            log(f"optimizer: {kind}")
+            break
+        except Exception as e:
+            log(f"{kind} AdamW unavailable ({e})")
+            opt = None
+    if opt is None:
+        opt = make_opt("foreach")
         fused_ok = False
     log(f"built in {time.time() - t_build:.1f}s; mem {torch.cuda.memory_allocated()/2**30:.1f} GiB")
     torch.cuda.reset_peak_memory_stats()

@@ -201,6 +201,18 @@ int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                       uint16_t* dV, int B, int Hq, int Hkv, int64_t S,
                       float scale, void* stream);
 
+/* ---- Fused optimizer ----------------------------------------------------- */
+
+/* One-sweep AdamW over a device pointer table (torch semantics: decoupled
+ * weight decay, bias correction; fp32 math, bf16 p/g/m/v storage). prefix is
+ * the inclusive element prefix sum [T+1] (all sizes 8-multiples); grad_scale
+ * (nullable fp32) divides grads in-register (grad-clip fold). */
+int vh_adamw_bf16(const uint64_t* p_ptrs, const uint64_t* g_ptrs,
+                  const uint64_t* m_ptrs, const uint64_t* v_ptrs,
+                  const int64_t* prefix, int T, int64_t total, float lr,
+                  float beta1, float beta2, float eps, float weight_decay,
+                  int step, const float* grad_scale, void* stream);
+
 /* ---- Fused chunked cross-entropy ---------------------------------------- */
 
 /* Per-row softmax CE over a bf16 logits chunk:

@@ -22,6 +22,7 @@ SOURCES = [
     "vh_norms.hip",
     "vh_attention.hip",
     "vh_ce.hip",
+    "vh_adamw.hip",
 ]
 
 

@@ -67,6 +67,8 @@ def get_lib() -> ctypes.CDLL:
     _sig(lib, "vh_silu_mul_bf16", c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_silu_mul_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_ce_fwd_bf16", c_p, c_p, c_p, c_p, c_i64, c_i64, c_f32, c_i64, c_p)
+    _sig(lib, "vh_adamw_bf16", c_p, c_p, c_p, c_p, c_p, c_int, c_i64, c_f32,
+         c_f32, c_f32, c_f32, c_f32, c_int, c_p, c_p)
     _sig(lib, "vh_attn_fwd_bf16", c_p, c_p, c_p, c_p, c_p, c_int, c_int, c_int,
          c_i64, c_f32, c_p)
     _sig(lib, "vh_attn_bwd_pre_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
