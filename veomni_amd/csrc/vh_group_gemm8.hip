@@ -791,9 +791,14 @@ __global__ void k_transpose_pad(const bf16_t* __restrict__ src,
   int64_t chunk = blockIdx.x;          // which 64-row padded chunk
   int64_t col0 = (int64_t)blockIdx.y * 64;
   int64_t p0 = chunk * 64;
-  // find group: padded chunks never straddle groups (64-multiples)
-  int g = 0;
-  while (g < G && p0 >= padded_cumsum[g]) ++g;
+  // find group (binary search — a linear walk was ~G dependent global loads
+  // per block): padded chunks never straddle groups (64-multiples)
+  int lo = 0, hi = G;  // first g with p0 < padded_cumsum[g]
+  while (lo < hi) {
+    int mid = (lo + hi) >> 1;
+    if (p0 >= padded_cumsum[mid]) lo = mid + 1; else hi = mid;
+  }
+  int g = lo;
   if (g >= G) return;
   int64_t pstart = (g > 0) ? padded_cumsum[g - 1] : 0;
   int64_t sstart = (g > 0) ? cumsum[g - 1] : 0;
