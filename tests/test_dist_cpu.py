@@ -344,3 +344,34 @@ def test_checkpoint_ep(tmp_path_factory):
 
     d = tempfile.mkdtemp(prefix="vh_ckpt_ep_")
     spawn(_ckpt_ep, d)
+
+
+def test_clip_fold_sets_grad_scale_without_scaling_grads():
+    """clip_grad_norm(fused_optimizer=...) must leave grads untouched and set
+    opt.grad_scale = max(1, (norm + eps) / max_norm) (the in-register fold
+    torch's fused AdamW and VeAdamW consume)."""
+    import torch
+
+    from veomni_amd.distributed.fsdp2 import clip_grad_norm
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+
+    init_parallel_state(device_type="cpu")
+    model = torch.nn.Linear(8, 8)
+    for p in model.parameters():
+        p.grad = torch.full_like(p, 2.0)
+    g_before = [p.grad.clone() for p in model.parameters()]
+
+    class _Opt:
+        pass
+
+    opt = _Opt()
+    norm = clip_grad_norm(model, max_norm=1.0, fused_optimizer=opt)
+    for p, g0 in zip(model.parameters(), g_before):
+        assert torch.equal(p.grad, g0), "grads must not be scaled in the fold path"
+    expected = max(1.0, (float(norm) + 1e-6) / 1.0)
+    assert abs(float(opt.grad_scale) - expected) < 1e-4
+    # below-threshold norm -> scale exactly 1
+    for p in model.parameters():
+        p.grad = torch.full_like(p, 1e-6)
+    clip_grad_norm(model, max_norm=1.0, fused_optimizer=opt)
+    assert float(opt.grad_scale) == 1.0
