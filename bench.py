@@ -120,6 +120,7 @@ def main():
 
     import veomni_amd.ops  # registrations
     from veomni_amd.ops import HIP_OPS_CONFIG, hip_lib
+    from veomni_amd.distributed.loss_utils import mean_global_loss
     from veomni_amd.distributed.fsdp2 import build_parallelize_model
     from veomni_amd.distributed.parallel_state import init_parallel_state
     from veomni_amd.models import PRESETS, build_model
@@ -179,8 +180,13 @@ def main():
     seq = args.seq_len
     batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs, seed=42 + rank, device="cuda")
 
+    n_valid = (batch["labels"] != -100).sum()
+
     def one_step():
         loss, _ = model(**batch)
+        # reference trainer semantics (loss_utils.py:54-96): global per-token
+        # mean; a no-op at equal tokens/rank but keeps the parity surface hot
+        loss = mean_global_loss(loss, n_valid)
         loss.backward()
         # fused path folds the clip coefficient into AdamW's grad_scale
         model.clip_grad_norm_(1.0, fused_optimizer=opt if fused_ok else None)

@@ -375,3 +375,25 @@ def test_clip_fold_sets_grad_scale_without_scaling_grads():
         p.grad = torch.full_like(p, 1e-6)
     clip_grad_norm(model, max_norm=1.0, fused_optimizer=opt)
     assert float(opt.grad_scale) == 1.0
+
+
+def _mean_global_loss_worker(rank, world):
+    from veomni_amd.distributed.loss_utils import mean_global_loss
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+
+    init_parallel_state(device_type="cpu")
+    # uneven valid-token counts: rank0 has 10 tokens at loss 2.0, rank1 has
+    # 30 tokens at loss 4.0 -> global per-token mean = (10*2 + 30*4)/40 = 3.5
+    n = [10, 30][rank]
+    loss = torch.tensor([2.0, 4.0][rank])
+    out = mean_global_loss(loss, n)
+    # after FSDP averages grads (sum/world), the effective loss is
+    # sum_r out_r / world  (the fsdp_size factor cancels the division)
+    gathered = [torch.zeros(()) for _ in range(world)]
+    dist.all_gather(gathered, out)
+    eff = sum(g.item() for g in gathered) / world
+    assert abs(eff - 3.5) < 1e-5, eff
+
+
+def test_mean_global_loss_gloo():
+    spawn(_mean_global_loss_worker)
