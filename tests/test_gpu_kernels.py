@@ -375,3 +375,35 @@ def test_veadamw_vs_torch_fused(lib):
     for a, b, s in zip(pa, pb, shapes):
         torch.testing.assert_close(a.float(), b.float(), rtol=2e-2, atol=2e-2,
                                    msg=lambda m: f"shape {s}: {m}")
+
+
+def test_veadamw_dtensor_params(lib):
+    """VeAdamW must step DTensor-sharded (FSDP2-style) params through their
+    local shards — the N>=2 bench path (1-rank mesh here)."""
+    import torch.distributed as dist
+    from veomni_amd.optim import VeAdamW
+
+    if not dist.is_initialized():
+        import os
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import Shard, distribute_tensor
+
+    mesh = init_device_mesh("cuda", (1,))
+    torch.manual_seed(9)
+    full = bf(torch.randn(256, 64)).cuda()
+    dt = torch.nn.Parameter(distribute_tensor(full.clone(), mesh, [Shard(0)]))
+    plain = torch.nn.Parameter(full.clone())
+    g = bf(torch.randn(256, 64)).cuda()
+    dt.grad = distribute_tensor(g.clone(), mesh, [Shard(0)])
+    plain.grad = g.clone()
+    kw = dict(lr=1e-2, betas=(0.9, 0.95), weight_decay=0.1)
+    oa = VeAdamW([dt], **kw)
+    ob = VeAdamW([plain], **kw)
+    oa.step()
+    ob.step()
+    torch.testing.assert_close(dt.data.to_local(), plain.data, rtol=0, atol=0)
