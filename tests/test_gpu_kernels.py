@@ -402,8 +402,11 @@ def test_veadamw_dtensor_params(lib):
     dt.grad = distribute_tensor(g.clone(), mesh, [Shard(0)])
     plain.grad = g.clone()
     kw = dict(lr=1e-2, betas=(0.9, 0.95), weight_decay=0.1)
-    oa = VeAdamW([dt], **kw)
-    ob = VeAdamW([plain], **kw)
-    oa.step()
-    ob.step()
-    torch.testing.assert_close(dt.data.to_local(), plain.data, rtol=0, atol=0)
+    try:
+        oa = VeAdamW([dt], **kw)
+        ob = VeAdamW([plain], **kw)
+        oa.step()
+        ob.step()
+        torch.testing.assert_close(dt.data.to_local(), plain.data, rtol=0, atol=0)
+    finally:
+        dist.destroy_process_group()
