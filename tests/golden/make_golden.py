@@ -235,6 +235,40 @@ for sp_size in (2, 4):
         golden[f"spcol/{sp_size}/{rank}/labels_out"] = sliced
         golden[f"spcol/{sp_size}/{rank}/ids_out"] = ids_s
 
+# ── Packing collator ─ ref: data/data_collator.py:219-317 ──
+_dc = importlib.import_module("veomni.data.data_collator")
+PackingCollator = _dc.PackingCollator
+DEFAULT_INFO = _dc.DEFAULT_DATA_COLLATE_INFO
+
+def _mk_pack_features(lens, seed):
+    g = torch.Generator().manual_seed(seed)
+    feats = []
+    for L in lens:
+        ids = torch.randint(0, 500, (L,), generator=g)
+        feats.append({
+            "input_ids": ids,
+            "labels": ids.clone(),
+            "attention_mask": torch.ones(L, dtype=torch.int64),
+            "position_ids": torch.arange(L),
+        })
+    return feats
+
+for name, lens, pad_to in (("plain", [5, 7, 3], None), ("padded", [5, 7, 3], 32),
+                           ("single", [9], 16)):
+    col = object.__new__(PackingCollator)
+    col.collate_infos = DEFAULT_INFO.copy()
+    col.pad_to_length = pad_to if pad_to else False
+    col.seq_classification = False
+    col.metadata_collate_func = None
+    col.sp_enabled = False
+    out = col(_mk_pack_features(lens, seed=11))
+    for k in ("input_ids", "labels", "attention_mask", "position_ids",
+              "cu_seq_lens_q"):
+        golden[f"pack/{name}/{k}"] = out[k]
+    golden[f"pack/{name}/max_length_q"] = torch.tensor(out["max_length_q"])
+    if "tail_padding_length" in out:
+        golden[f"pack/{name}/tail"] = out["tail_padding_length"]
+
 # ── EP permute/unpermute ─ ref: distributed/moe/moe_utils.py:19-99 ──
 from veomni.distributed.moe.moe_utils import (  # noqa: E402
     generate_weights_idx,
