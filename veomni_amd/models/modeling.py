@@ -303,6 +303,24 @@ class TopKRouter(nn.Module):
         return router_logits, top_v.to(router_logits.dtype), top_i
 
 
+class _EagerEPExperts:
+    """Eager expert compute for the EP dispatch path (plain autograd ops;
+    routing weights applied later by unpermute, matching the reference's
+    post-down_proj weighting)."""
+
+    @staticmethod
+    def apply(permute_tokens, cumsum, gate_up_w, down_w):
+        outs = []
+        start = 0
+        for g in range(gate_up_w.shape[0]):
+            end = int(cumsum[g])
+            cur = permute_tokens[start:end]
+            gate, up = F.linear(cur, gate_up_w[g]).chunk(2, dim=-1)
+            outs.append(F.linear(F.silu(gate) * up, down_w[g]))
+            start = end
+        return torch.cat(outs, dim=0)
+
+
 class Experts(nn.Module):
     """Merged gate_up [E,2I,H] + down [E,H,I] expert bank (HF v5 layout)."""
 
@@ -321,6 +339,16 @@ class Experts(nn.Module):
     def forward(self, hidden_states, top_k_index, top_k_weights):
         if veomni_moe_experts_forward.use_non_eager_impl:
             return veomni_moe_experts_forward(self, hidden_states, top_k_index, top_k_weights)
+        if get_parallel_state().ep_enabled:
+            # eager under EP still routes through the dispatch/combine path
+            # (the reference's eager-EP equally uses the EP classes; the
+            # local expert bank only holds E/ep slices, so the plain loop
+            # below would index out of range)
+            from ..distributed.moe import dispatch_to_ep_class
+
+            return dispatch_to_ep_class(
+                _EagerEPExperts, self.num_experts, top_k_weights, top_k_index,
+                hidden_states, self.gate_up_proj, self.down_proj)
         # eager loop — weights applied AFTER down_proj (ref :266-294)
         final = torch.zeros_like(hidden_states)
         with torch.no_grad():
