@@ -428,6 +428,7 @@ def _fsdp_ep_moe_equivalence(rank, ws):
 
     model = build_parallelize_model(model, param_dtype=torch.float32,
                                     reduce_dtype=torch.float32)
+    model.use_checkpoint = True   # gradient ckpt + FSDP2 + EP compose (bench config)
     opt = torch.optim.AdamW(model.parameters(), lr=1e-2, betas=(0.9, 0.95))
     ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2, betas=(0.9, 0.95))
 
