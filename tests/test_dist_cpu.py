@@ -456,3 +456,37 @@ def _fsdp_ep_moe_equivalence(rank, ws):
 
 def test_fsdp2_ep_moe_equivalence():
     spawn(_fsdp_ep_moe_equivalence)
+
+
+def _hsdp_equivalence(rank, ws):
+    """HSDP (dp_replicate=ws, dp_shard=1) over the 2-D mesh: pure replication
+    with gradient all-reduce — loss/grad-norm match a single-process run."""
+    from veomni_amd.distributed.fsdp2 import build_parallelize_model
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch
+
+    init_parallel_state(dp_replicate_size=ws)
+    bind_ops("eager")
+    model = build_model("tiny-dense")
+    ref = build_model("tiny-dense")
+
+    model = build_parallelize_model(model, param_dtype=torch.float32,
+                                    reduce_dtype=torch.float32)
+    batch = synthetic_batch(512, 64, seed=13)
+    loss, _ = model(**batch)
+    loss.backward()
+    gn = model.clip_grad_norm_(1e9)
+
+    rloss, _ = ref(**batch)
+    rloss.backward()
+    rgn = torch.nn.utils.get_total_norm(
+        [p.grad for p in ref.parameters() if p.grad is not None])
+    torch.testing.assert_close(loss.detach().float(), rloss.detach().float(),
+                               rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(gn.float(), rgn.float(), rtol=1e-3, atol=1e-4)
+
+
+def test_hsdp_equivalence():
+    spawn(_hsdp_equivalence)

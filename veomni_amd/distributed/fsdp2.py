@@ -149,8 +149,10 @@ def clip_grad_norm(model: nn.Module, max_norm: float, norm_type: float = 2.0,
     ep = [p for p in params if getattr(p, "_ep_param", False)]
 
     total = _local_pth_sum(dense, norm_type)
-    if ps.fsdp_size > 1:
-        dist.all_reduce(total, group=ps.fsdp_mesh.get_group())
+    # sum each DISTINCT shard once: HSDP replicas hold identical grads, so
+    # the reduce runs over the shard_sp group only
+    if ps.fsdp_shard_size > 1:
+        dist.all_reduce(total, group=ps.fsdp_shard_group)
     if ep:
         ep_sum = _local_pth_sum(ep, norm_type)
         if ps.ep_fsdp_size > 1:

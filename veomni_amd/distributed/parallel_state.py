@@ -27,6 +27,7 @@ _PARALLEL_STATE: Optional["ParallelState"] = None
 class ParallelState:
     world_size: int = 1
     dp_size: int = 1           # dp_shard size
+    dp_replicate_size: int = 1  # HSDP outer replication (ref :153-172)
     ulysses_size: int = 1
     ep_size: int = 1
     dp_mode: str = "fsdp2"
@@ -34,7 +35,8 @@ class ParallelState:
     device_type: str = "cpu"
     device_mesh: Optional[DeviceMesh] = None      # (dp_shard, ulysses)
     ep_device_mesh: Optional[DeviceMesh] = None   # (ep_fsdp, ep)
-    _fsdp_mesh: Optional[DeviceMesh] = None       # flattened dp_shard_sp
+    _fsdp_mesh: Optional[DeviceMesh] = None       # dp_shard_sp (2-D for HSDP)
+    _shard_sp_mesh: Optional[DeviceMesh] = None    # always the flat shard dim
     extra_parallel_names: tuple = ("ep",)
 
     # ---------------------------------------------------------------- flags
@@ -52,6 +54,10 @@ class ParallelState:
     def ep_enabled(self) -> bool:
         return self.ep_size > 1
 
+    @property
+    def dp_replicate_enabled(self) -> bool:
+        return self.dp_replicate_size > 1
+
     def extra_parallel_enabled(self, name: str) -> bool:
         return name == "ep" and self.ep_enabled
 
@@ -62,7 +68,21 @@ class ParallelState:
 
     @property
     def fsdp_size(self) -> int:
+        # total data-parallel extent FSDP divides gradients by (HSDP
+        # includes the replicate dim; ref :242-243 world/(pp*tp))
+        return self.dp_replicate_size * self.dp_size * self.ulysses_size
+
+    @property
+    def fsdp_shard_size(self) -> int:
+        # distinct-shard extent (excludes HSDP replication): the grad-norm
+        # p-th-power sum must count each shard once
         return self.dp_size * self.ulysses_size
+
+    @property
+    def fsdp_shard_group(self):
+        if self._shard_sp_mesh is None:
+            return None
+        return self._shard_sp_mesh.get_group()
 
     @property
     def ep_fsdp_mesh(self) -> Optional[DeviceMesh]:
@@ -141,6 +161,7 @@ def init_parallel_state(
     dp_mode: str = "fsdp2",
     async_ulysses: bool = False,
     device_type: Optional[str] = None,
+    dp_replicate_size: int = 1,
 ) -> ParallelState:
     """Build the device meshes and register the global state.
 
@@ -163,16 +184,29 @@ def init_parallel_state(
     if device_type is None:
         device_type = "cuda" if torch.cuda.is_available() else "cpu"
     if dp_size is None:
-        assert world_size % ulysses_size == 0
-        dp_size = world_size // ulysses_size
-    assert dp_size * ulysses_size == world_size, (dp_size, ulysses_size, world_size)
+        assert world_size % (ulysses_size * dp_replicate_size) == 0
+        dp_size = world_size // (ulysses_size * dp_replicate_size)
+    assert dp_replicate_size * dp_size * ulysses_size == world_size, (
+        dp_replicate_size, dp_size, ulysses_size, world_size)
     dp_shard_sp = dp_size * ulysses_size
     assert dp_shard_sp % ep_size == 0, f"ep_size {ep_size} must divide dp_shard*sp {dp_shard_sp}"
+    assert dp_replicate_size == 1 or ep_size == 1, \
+        "HSDP + EP needs a replicate-aware expert mesh (round-2 scope)"
 
-    mesh = init_device_mesh(
-        device_type, (dp_size, ulysses_size), mesh_dim_names=("dp_shard", "ulysses")
-    )
-    fsdp_mesh = mesh["dp_shard", "ulysses"]._flatten(mesh_dim_name="dp_shard_sp")
+    if dp_replicate_size > 1:
+        # HSDP: outer replicate dim; FSDP consumes the 2-D
+        # (dp_replicate, dp_shard_sp) mesh (ref fsdp_mesh property :218-231)
+        mesh = init_device_mesh(
+            device_type, (dp_replicate_size, dp_size, ulysses_size),
+            mesh_dim_names=("dp_replicate", "dp_shard", "ulysses"))
+        shard_sp_mesh = mesh["dp_shard", "ulysses"]._flatten(mesh_dim_name="dp_shard_sp")
+        fsdp_mesh = mesh["dp_replicate", "dp_shard_sp"]
+    else:
+        mesh = init_device_mesh(
+            device_type, (dp_size, ulysses_size), mesh_dim_names=("dp_shard", "ulysses")
+        )
+        fsdp_mesh = mesh["dp_shard", "ulysses"]._flatten(mesh_dim_name="dp_shard_sp")
+        shard_sp_mesh = fsdp_mesh
 
     ep_mesh = None
     if ep_size > 1:
@@ -183,6 +217,7 @@ def init_parallel_state(
     _PARALLEL_STATE = ParallelState(
         world_size=world_size,
         dp_size=dp_size,
+        dp_replicate_size=dp_replicate_size,
         ulysses_size=ulysses_size,
         ep_size=ep_size,
         dp_mode=dp_mode,
@@ -191,6 +226,7 @@ def init_parallel_state(
         device_mesh=mesh,
         ep_device_mesh=ep_mesh,
         _fsdp_mesh=fsdp_mesh,
+        _shard_sp_mesh=shard_sp_mesh,
     )
     return _PARALLEL_STATE
 
