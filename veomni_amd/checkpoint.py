@@ -1,16 +1,17 @@
 """Distributed checkpoint save/resume for the training step.
 
 Parity target: /root/reference/veomni/checkpoint/dcp_checkpointer.py:431-678
-(`DistributedCheckpointer.save/load` over torch DCP) with the EP-dim
-handling of :111-430 reduced to round-1 scope: EP-sliced parameters are
-saved under EP-rank-qualified keys, so resume requires the SAME ep_size
-(the reference's SpecInfo-based cross-topology resharding is §8f follow-up;
-dense / FSDP2 state resharding is handled by DCP itself via DTensor).
+(`DistributedCheckpointer.save/load` over torch DCP) including the EP-dim
+handling of :111-430: EP-sliced parameters (and their optimizer moments) are
+re-expressed as DTensors over the full (ep_fsdp, ep) mesh with placements
+[Shard(1), Shard(0)], so DCP stores ONE global expert tensor per fqn and its
+load-time resharding makes resume work across DIFFERENT ep_size topologies
+(the reference's SpecInfo mechanism, restated through DTensor metadata).
 
 Works in three regimes:
   - single process, no process group (bench N=1): DCP no-dist path;
   - FSDP2 over gloo/RCCL: DTensor-sharded save/load;
-  - EP: per-EP-rank expert keys + shared dense keys.
+  - EP: globally-shaped expert tensors, any-topology resume.
 """
 
 from __future__ import annotations
@@ -28,30 +29,43 @@ from torch.distributed.checkpoint.state_dict import (
     set_model_state_dict,
     set_state_dict,
 )
+from torch.distributed.tensor import DTensor, Shard
 
 from .distributed.parallel_state import get_parallel_state
-
-
-def _ep_suffix() -> str:
-    ps = get_parallel_state()
-    return f"__ep{ps.ep_rank}" if ps.ep_enabled else ""
 
 
 def _ep_fqns(model) -> set:
     return set(getattr(model, "_ep_fqns", set()))
 
 
-def _qualify(sd: dict, ep_fqns: set, suffix: str, param_level: bool) -> dict:
-    """Rename EP-owned keys with the EP-rank suffix. Optimizer state dicts
-    nest under 'state'/<fqn>/<slot>; model dicts are flat fqn->tensor."""
-    if not suffix:
-        return sd
-    if param_level:
-        return {(k + suffix if k in ep_fqns else k): v for k, v in sd.items()}
-    out = dict(sd)
-    if "state" in out:
-        out["state"] = {(k + suffix if k in ep_fqns else k): v
-                        for k, v in out["state"].items()}
+def _raw(t: torch.Tensor) -> torch.Tensor:
+    return t.to_local() if isinstance(t, DTensor) else t
+
+
+def _wrap_global(t: torch.Tensor) -> torch.Tensor:
+    """Local EP(+ep_fsdp) shard -> DTensor with the TRUE global expert shape
+    over the (ep_fsdp, ep) mesh: dim 0 split over ep, dim 1 over ep_fsdp."""
+    ps = get_parallel_state()
+    raw = _raw(t)
+    if raw.ndim == 0:
+        return t
+    return DTensor.from_local(raw, ps.ep_device_mesh, [Shard(1), Shard(0)],
+                              run_check=False)
+
+
+def _map_ep(state: dict, ep_fqns: set, fn) -> dict:
+    """Apply fn to every EP-owned tensor in a model state dict (flat) or an
+    optimizer state dict (nested under 'state'/<fqn>/<slot>)."""
+    out = dict(state)
+    for k in list(out.keys()):
+        if k == "state" and isinstance(out[k], dict):
+            out["state"] = {
+                fqn: ({slot: (fn(v) if torch.is_tensor(v) else v)
+                       for slot, v in slots.items()} if fqn in ep_fqns else slots)
+                for fqn, slots in out["state"].items()
+            }
+        elif k in ep_fqns and torch.is_tensor(out[k]):
+            out[k] = fn(out[k])
     return out
 
 
@@ -62,11 +76,14 @@ def save_checkpoint(path: str, model, optimizer: Optional[torch.optim.Optimizer]
         msd, osd = get_state_dict(model, optimizer, options=options)
     else:
         msd, osd = get_model_state_dict(model, options=options), None
-    suffix = _ep_suffix()
     fqns = _ep_fqns(model)
-    state = {"model": _qualify(msd, fqns, suffix, param_level=True)}
+    if fqns and get_parallel_state().ep_enabled:
+        msd = _map_ep(msd, fqns, _wrap_global)
+        if osd is not None:
+            osd = _map_ep(osd, fqns, _wrap_global)
+    state = {"model": msd}
     if osd is not None:
-        state["optim"] = _qualify(osd, fqns, suffix, param_level=False)
+        state["optim"] = osd
     if extra and (not dist.is_initialized() or dist.get_rank() == 0):
         torch.save(extra, os.path.join(path, "extra.pt") if os.path.isdir(path) else path + ".extra.pt")
     os.makedirs(path, exist_ok=True)
@@ -79,23 +96,41 @@ def load_checkpoint(path: str, model, optimizer: Optional[torch.optim.Optimizer]
         msd, osd = get_state_dict(model, optimizer, options=options)
     else:
         msd, osd = get_model_state_dict(model, options=options), None
-    suffix = _ep_suffix()
     fqns = _ep_fqns(model)
-    state = {"model": _qualify(msd, fqns, suffix, param_level=True)}
-    if osd is not None:
-        state["optim"] = _qualify(osd, fqns, suffix, param_level=False)
+    ep_on = bool(fqns) and get_parallel_state().ep_enabled
+
+    # templates: EP entries re-expressed globally (fresh buffers, so the load
+    # may reshard freely); everything else loads in its native sharding
+    def tmpl(t):
+        return _wrap_global(_raw(t).clone()) if t.ndim != 0 else t
+
+    lmsd = _map_ep(msd, fqns, tmpl) if ep_on else msd
+    losd = (_map_ep(osd, fqns, tmpl) if ep_on else osd) if osd is not None else None
+    state = {"model": lmsd}
+    if losd is not None:
+        state["optim"] = losd
     dcp.load(state, checkpoint_id=path)
-    # un-qualify back to the live fqns before applying
-    if suffix:
-        state["model"] = {k[: -len(suffix)] if k.endswith(suffix) else k: v
-                          for k, v in state["model"].items()}
-        if osd is not None and "state" in state["optim"]:
-            state["optim"]["state"] = {
-                k[: -len(suffix)] if k.endswith(suffix) else k: v
-                for k, v in state["optim"]["state"].items()
-            }
+
+    if ep_on:
+        # copy the resharded EP locals back into the native-sharded entries
+        def restore(dst_sd, src_sd):
+            for k in fqns:
+                if k in dst_sd and torch.is_tensor(dst_sd[k]):
+                    _raw(dst_sd[k]).copy_(_raw(src_sd[k]))
+            if "state" in dst_sd and "state" in src_sd:
+                for fqn in fqns:
+                    if fqn in dst_sd["state"]:
+                        for slot, v in dst_sd["state"][fqn].items():
+                            if torch.is_tensor(v) and v.ndim != 0:
+                                _raw(v).copy_(_raw(src_sd["state"][fqn][slot]))
+                            elif torch.is_tensor(v):
+                                v.copy_(src_sd["state"][fqn][slot])
+        restore(msd, lmsd)
+        if osd is not None:
+            restore(osd, losd)
+
     if optimizer is not None:
-        set_state_dict(model, optimizer, model_state_dict=state["model"],
-                       optim_state_dict=state["optim"], options=options)
+        set_state_dict(model, optimizer, model_state_dict=(lmsd if not ep_on else msd),
+                       optim_state_dict=(losd if not ep_on else osd), options=options)
     else:
-        set_model_state_dict(model, state["model"], options=options)
+        set_model_state_dict(model, (lmsd if not ep_on else msd), options=options)
