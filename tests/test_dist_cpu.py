@@ -33,11 +33,19 @@ def _run(rank, world_size, fn, port, args):
         dist.destroy_process_group()
 
 
-def spawn(fn, *args, world_size=WORLD):
-    import random
+def _free_port():
+    import socket
 
-    port = random.randint(20000, 40000)
-    mp.spawn(_run, args=(world_size, fn, port, args), nprocs=world_size, join=True)
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def spawn(fn, *args, world_size=WORLD):
+    mp.spawn(_run, args=(world_size, fn, _free_port(), args), nprocs=world_size,
+             join=True)
 
 
 # ---------------------------------------------------------------- ulysses a2a
