@@ -253,6 +253,32 @@ def _mk_pack_features(lens, seed):
         })
     return feats
 
+def _mk_vlm_features(lens, seed):
+    g = torch.Generator().manual_seed(seed)
+    feats = []
+    for L in lens:
+        ids = torch.randint(0, 500, (L,), generator=g)
+        npatch = int(torch.randint(2, 6, (1,), generator=g))
+        feats.append({
+            "input_ids": ids,
+            "labels": ids.clone(),
+            "attention_mask": torch.ones(L, dtype=torch.int64),
+            "position_ids": torch.arange(L),
+            "pixel_values": torch.randn(npatch, 8, generator=g),
+            "image_grid_thw": torch.tensor([[1, npatch, 1]]),
+        })
+    return feats
+
+col = object.__new__(PackingCollator)
+col.collate_infos = DEFAULT_INFO.copy()
+col.pad_to_length = False
+col.seq_classification = False
+col.metadata_collate_func = None
+col.sp_enabled = True  # skip fa-kwargs (needs full veomni utils on load side)
+vout = col(_mk_vlm_features([4, 6], seed=21))
+for k in ("input_ids", "pixel_values", "image_grid_thw"):
+    golden[f"packvlm/{k}"] = vout[k]
+
 for name, lens, pad_to in (("plain", [5, 7, 3], None), ("padded", [5, 7, 3], 32),
                            ("single", [9], 16)):
     col = object.__new__(PackingCollator)
