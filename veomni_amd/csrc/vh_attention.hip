@@ -357,20 +357,26 @@ extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
 }
 
 // ============================================================================
-// Flash attention backward (bf16, causal, GQA, D = 128).
+// Flash attention backward, monolithic variant (bf16, causal, GQA, D = 128).
 //
-// FA2 column-block structure: grid (S/128, B*Hq); a block (4 waves, 256 thr)
-// owns 128 kv rows (wave w the 32-row slice w) and loops over 32-row q tiles
-// from the causal diagonal to S. Both score orientations are recomputed so
-// no cross-lane transpose is needed (the MFMA A-operand always comes from
-// the forward's cheap reg->chunk pack-exchange):
-//   or1 (q in regs, kv = lane):  S1 = mfma(K, Q), dP1 = mfma(V, dO)
-//       -> P1, dS1 packs feed  dV += mfma(P1^T, dO^T),  dK += mfma(dS1^T, Q^T)
-//   or2 (kv in regs, q = lane): S2 = mfma(Q, K), dP2 = mfma(dO, V)
-//       -> dS2 pack feeds      dQ += mfma(dS2, K^T)
+// NOTE: this kernel is PROBE INFRASTRUCTURE (vh_attn_bwd_probe_bf16, ablation
+// modes 0-8 — the measurements in DESIGN.md came from it). The DISPATCHED
+// backward is vh_attn_bwd2_bf16 below: two split kernels (k_attn_bwd_dkv /
+// k_attn_bwd_dq) that keep all accumulators in registers with no cross-wave
+// reduction; the monolith's LDS dQ exchange is what they replaced.
+//
+// Structure: grid (S/128, B*Hq); a block (4 waves, 256 thr) owns 128 kv rows
+// (wave w the 32-row slice w) and loops 32-row q tiles from the causal
+// diagonal. Both score orientations are recomputed so no cross-lane
+// transpose is needed (the MFMA A-operand always comes from the forward's
+// reg->chunk pack-exchange):
+//   or1 (C = [kv regs][q lanes]): S1 = mfma(K, Q), dP1 = mfma(V, dO)
+//       -> pack(dS1) = A[q][kv] feeds  dQ += mfma(dS1^T-pack, K^T-tile)
+//   or2 (C = [q regs][kv lanes]): S2 = mfma(Q, K), dP2 = mfma(dO, V)
+//       -> pack(P2), pack(dS2) = A[kv][q] feed dV/dK += mfma(pack, dO^T/Q^T)
 // P = exp2(S*scale2 - lse2[q]); dS = P * (dP - delta[q]) * scale.
-// dQ partials are block-reduced in LDS fp32 and atomically added to a fp32
-// buffer; dK/dV are written per Q-head and the host sums GQA groups.
+// dQ partials go through per-wave bf16 LDS quarters + fp32 global atomics;
+// dK/dV are written per Q-head and the host sums GQA groups.
 // delta = rowsum(dO*O), lse2 = LSE*log2e from vh_attn_bwd_pre_bf16.
 // ============================================================================
 
