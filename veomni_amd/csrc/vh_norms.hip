@@ -133,13 +133,14 @@ extern "C" int vh_rmsnorm_fwd_bf16(const uint16_t* x, const uint16_t* w,
 
 // --------------------------------------------------------------- RMSNorm bwd
 // dx = rs * (g - x * rs^2/H * dot(g, x)),  g = dy * w  (fp32)
-// dw += dy * bf16(x * rs)   accumulated per-block in LDS, then one atomic pass.
-// dw accumulates in per-lane registers (CHUNKS = Hv/64 column chunks per
-// lane) across all of the wave's rows, with one global fp32 atomicAdd pass
-// at kernel end. The previous version did 8 LDS fp32 atomicAdds per bf16x8
-// per row; ds_add_f32 under 4-wave same-address contention measures ~600
-// cycles/op on gfx950 (see the attention-backward probe notes in DESIGN.md)
-// and capped the kernel at ~1 TB/s.
+// dw += dy * bf16(x * rs): accumulated in per-lane REGISTERS (CHUNKS = Hv/64
+// column chunks per lane) across the wave's rows, stored as per-wave partial
+// rows in a scratch buffer, then summed by k_dw_reduce — no fp32 atomics on
+// shared addresses anywhere. Earlier versions measured: 8 LDS fp32
+// atomicAdds per bf16x8 per row (ds_add_f32 under 4-wave same-address
+// contention is ~600 cycles/op on gfx950 — see the attention-backward probe
+// notes in DESIGN.md) capped the kernel at ~1 TB/s; a global fp32 atomicAdd
+// epilogue serialized ~2k same-address chains and was equally slow.
 template <int CHUNKS>
 __global__ __launch_bounds__(256, 2) void k_rmsnorm_bwd_reg(
     const bf16x8* __restrict__ dy, const bf16x8* __restrict__ x,
