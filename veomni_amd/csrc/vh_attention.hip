@@ -1014,6 +1014,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv(
 
 
 
+
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
