@@ -365,6 +365,28 @@ with torch.no_grad():
     logits = ref_model(input_ids=ids).logits
 golden["e2e/logits"] = logits.detach()
 
+# ── multi-step training trace (loss trajectory incl. optimizer + clip) ──
+train_model = m.Qwen3MoeForCausalLM(Qwen3MoeConfig(**tiny_cfg, attn_implementation="sdpa"))
+with torch.no_grad():
+    for name, p in sorted(train_model.named_parameters(), key=lambda kv: kv[0]):
+        gseed = torch.Generator().manual_seed(abs(hash(name)) % (2**31))
+        if "norm" in name and name.endswith("weight") and p.dim() == 1:
+            p.fill_(1.0)
+        else:
+            p.copy_(torch.randn(p.shape, generator=gseed) * 0.03)
+opt = torch.optim.AdamW(train_model.parameters(), lr=1e-3, betas=(0.9, 0.95),
+                        eps=1e-8, weight_decay=0.01)
+step_losses = []
+for st in range(3):
+    sids = torch.randint(0, 512, (1, 64), generator=torch.Generator().manual_seed(1000 + st))
+    out2 = train_model(input_ids=sids, labels=sids.clone())
+    out2.loss.backward()
+    torch.nn.utils.clip_grad_norm_(train_model.parameters(), 1.0)
+    opt.step()
+    opt.zero_grad(set_to_none=True)
+    step_losses.append(out2.loss.detach())
+golden["train3/losses"] = torch.stack(step_losses)
+
 out_path = os.path.join(os.path.dirname(__file__), "golden.pt")
 torch.save(golden, out_path)
 print(f"wrote {out_path} with {len(golden)} entries")
