@@ -42,9 +42,11 @@ def _stream(lens):
     return out
 
 
-def _run(mod_strategy, mod_loader, lens, budget, buffer_size, nmb, length, cap):
+def _run(mod_strategy, mod_loader, lens, budget, buffer_size, nmb, length,
+         cap, wsteps=0, winit=200):
     strat = mod_strategy(token_micro_bsz=budget, buffer_size=buffer_size,
-                         physical_token_cap=cap)
+                         physical_token_cap=cap, bsz_warmup_steps=wsteps,
+                         bsz_warmup_init_mbtoken=winit)
     loader = mod_loader(_stream(lens), strat, collate_fn=None,
                         num_micro_batch=nmb, length=length, drop_last=True)
     steps = []
@@ -69,14 +71,16 @@ if HAVE_HYP:
         nmb=st.integers(min_value=1, max_value=3),
         length=st.integers(min_value=3, max_value=20),
         cap=st.one_of(st.none(), st.integers(min_value=64, max_value=1024)),
+        warmup=st.sampled_from([(0, 200), (5, 64), (12, 128)]),
     )
     def test_selection_matches_reference(lens, budget, buffer_size, nmb,
-                                         length, cap):
+                                         length, cap, warmup):
         ref = _ref_module()
         from veomni_amd.data import DynamicBatchDataLoader, TextBatchingStrategy
 
+        wsteps, winit = warmup
         ours = _run(TextBatchingStrategy, DynamicBatchDataLoader, lens, budget,
-                    buffer_size, nmb, length, cap)
+                    buffer_size, nmb, length, cap, wsteps, winit)
         theirs = _run(ref.TextBatchingStrategy, ref.DynamicBatchSizeDataLoader,
-                      lens, budget, buffer_size, nmb, length, cap)
+                      lens, budget, buffer_size, nmb, length, cap, wsteps, winit)
         assert ours == theirs
