@@ -175,7 +175,7 @@ def test_ep_dispatch_combine():
 
 
 # ------------------------------------------------- FSDP2 grad-norm equivalence
-def _fsdp_equivalence(rank, ws):
+def _fsdp_equivalence(rank, ws, preset="tiny-dense"):
     from veomni_amd.distributed.fsdp2 import build_parallelize_model
     from veomni_amd.distributed.parallel_state import init_parallel_state
     from veomni_amd.models import build_model
@@ -184,8 +184,8 @@ def _fsdp_equivalence(rank, ws):
 
     init_parallel_state()
     bind_ops("eager")
-    model = build_model("tiny-dense")
-    ref = build_model("tiny-dense")  # identical seeded init
+    model = build_model(preset)
+    ref = build_model(preset)  # identical seeded init
 
     model = build_parallelize_model(model, param_dtype=torch.float32,
                                     reduce_dtype=torch.float32)
@@ -204,6 +204,12 @@ def _fsdp_equivalence(rank, ws):
 
 def test_fsdp2_equivalence():
     spawn(_fsdp_equivalence)
+
+
+def test_fsdp2_equivalence_qwen2_bias():
+    """configs[0] architecture (qwen2: attention bias) through the FSDP2
+    wrap — exercises the fused-qkv bias concat under sharding."""
+    spawn(_fsdp_equivalence, "tiny-qwen2")
 
 
 # -------------------------------------------------------------- EP param slice

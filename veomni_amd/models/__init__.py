@@ -38,6 +38,12 @@ PRESETS = {
         num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
         head_dim=32, qk_norm=False,
     ),
+    # qwen2-architecture knobs at test scale: attention bias on, no qk norm
+    "tiny-qwen2": ModelConfig(
+        name="tiny-qwen2", vocab_size=512, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=32, qk_norm=False, attention_bias=True,
+    ),
     "tiny-moe": ModelConfig(
         name="tiny-moe", vocab_size=512, hidden_size=128, intermediate_size=256,
         num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
