@@ -413,6 +413,61 @@ def test_mean_global_loss_gloo():
     spawn(_mean_global_loss_worker)
 
 
+def _mean_global_loss_sp_worker(rank, world):
+    """SP-enabled mean_global_loss vs the reference formula (loss_utils.py:68-94):
+    numerator = sp-reduced slice counts, denominator = world-sum of PER-RANK
+    slice counts (each token counted once), then /sp_size."""
+    from veomni_amd.distributed.loss_utils import mean_global_loss
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+
+    init_parallel_state(device_type="cpu", ulysses_size=2)
+    # world=4, sp=2, dp=2.  sp group A (ranks 0,1): 40 valid tokens split
+    # 10/30, token-weighted mean loss 3.5 (already sp-reduced, same value on
+    # both ranks).  sp group B (ranks 2,3): 20 tokens split 5/15, mean 1.0.
+    # Global per-token mean = (40*3.5 + 20*1.0)/60 = 8/3.
+    slice_tokens = [10, 30, 5, 15][rank]
+    loss = torch.tensor([3.5, 3.5, 1.0, 1.0][rank])
+    out = mean_global_loss(loss, slice_tokens)
+    expected = [14.0 / 3, 14.0 / 3, 2.0 / 3, 2.0 / 3][rank]
+    assert abs(out.item() - expected) < 1e-5, (rank, out.item(), expected)
+    gathered = [torch.zeros(()) for _ in range(world)]
+    dist.all_gather(gathered, out)
+    eff = sum(g.item() for g in gathered) / world
+    assert abs(eff - 8.0 / 3) < 1e-5, eff
+
+
+def test_mean_global_loss_sp_gloo():
+    spawn(_mean_global_loss_sp_worker, world_size=4)
+
+
+def test_veadamw_resume_invalidates_tables():
+    """load_state_dict swaps state tensors; the cached data_ptr
+    table must be dropped and the step counter must survive the round-trip."""
+    from veomni_amd.optim import VeAdamW
+
+    p = torch.nn.Parameter(torch.zeros(16, dtype=torch.bfloat16))
+    opt = VeAdamW([p])
+    opt.state[p]["exp_avg"] = torch.zeros(16, dtype=torch.bfloat16)
+    opt.state[p]["exp_avg_sq"] = torch.zeros(16, dtype=torch.bfloat16)
+    opt._step = 7
+    opt._tables = ("sentinel",) * 5 + ([(p, p.data, opt.state[p]["exp_avg"],
+                                         opt.state[p]["exp_avg_sq"])],)
+    assert not opt._tables_stale()
+    sd = opt.state_dict()
+    assert sd["ve_step"] == 7
+
+    p2 = torch.nn.Parameter(torch.zeros(16, dtype=torch.bfloat16))
+    opt2 = VeAdamW([p2])
+    opt2._tables = ("stale",)
+    opt2.load_state_dict(sd)
+    assert opt2._tables is None
+    assert opt2._step == 7
+    # direct state swap (the DCP set_optimizer_state_dict path) is caught by
+    # the per-step staleness check
+    opt.state[p]["exp_avg"] = torch.zeros(16, dtype=torch.bfloat16)
+    assert opt._tables_stale()
+
+
 def _fsdp_ep_moe_equivalence(rank, ws):
     """Full EP + FSDP2 wrap on a MoE model (the N>=2 bench structure:
     experts EP-sliced + fully_shard over the size-1 ep_fsdp mesh with

@@ -36,15 +36,20 @@ def mean_global_loss(loss: torch.Tensor,
     ps = get_parallel_state()
     nv = torch.as_tensor(local_valid_tokens, dtype=torch.float32,
                          device=loss.device)
-    if ps.sp_enabled:
-        dist.all_reduce(nv, group=ps.sp_group)
     if global_step_tokens is None:
+        # Denominator = world-sum of the PER-RANK counts (ref :77-78 reduces
+        # micro_batches_token_len, never the sp-reduced numerator): under SP
+        # each rank holds a seq slice, so the world sum already counts every
+        # token exactly once.
         total = nv.clone()
         if dist.is_initialized() and dist.get_world_size() > 1:
             dist.all_reduce(total)
     else:
         total = torch.as_tensor(global_step_tokens, dtype=torch.float32,
                                 device=loss.device)
+    if ps.sp_enabled:
+        # Numerator only (ref :74-75): full-sequence tokens of this sp group.
+        dist.all_reduce(nv, group=ps.sp_group)
     # fsdp divides gradients by its size; multiply back (ref :85)
     loss = loss * nv / total * float(ps.fsdp_size)
     if ps.sp_enabled:

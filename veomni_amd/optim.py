@@ -62,6 +62,30 @@ class VeAdamW(torch.optim.Optimizer):
             plist,
         )
 
+    def state_dict(self):
+        sd = super().state_dict()
+        sd["ve_step"] = self._step
+        return sd
+
+    def load_state_dict(self, state_dict):
+        state_dict = dict(state_dict)
+        self._step = int(state_dict.pop("ve_step", self._step))
+        super().load_state_dict(state_dict)
+        # state tensors were replaced — cached data_ptr tables are stale
+        self._tables = None
+
+    def _tables_stale(self):
+        # torch's load_state_dict / DCP set_optimizer_state_dict can swap the
+        # state tensors without going through our load_state_dict override;
+        # the pointer table must track the live tensors.
+        for p, lp, m, v in self._tables[5]:
+            st = self.state[p]
+            if st.get("exp_avg") is not m or st.get("exp_avg_sq") is not v:
+                return True
+            if _local(p).data_ptr() != lp.data_ptr():
+                return True
+        return False
+
     @torch.no_grad()
     def step(self, closure=None):
         assert closure is None
@@ -80,6 +104,8 @@ class VeAdamW(torch.optim.Optimizer):
                     break
             if dev is not None:
                 break
+        if self._tables is not None and self._tables_stale():
+            self._tables = None
         if self._tables is None:
             self._build_tables(dev)
         p_ptrs, m_ptrs, v_ptrs, prefix, total, plist = self._tables
