@@ -171,12 +171,17 @@ int vh_silu_mul_bwd_bf16(const uint16_t* dy, const uint16_t* gate,
 
 /* ---- Flash attention (causal, GQA, D = 128) ----------------------------- */
 
-/* Forward: O [B,Hq,S,128] bf16, LSE [B,Hq,S] fp32; S % 128 == 0.
+/* Forward: O [B,Hq,S,128] bf16, LSE [B,Hq,S] fp32; S % 256 == 0.
  * Replaces the external flash_attn wheel behind the reference's attention
- * slot (attention/flash.py:153-301) for the packed causal path. */
+ * slot (attention/flash.py:153-301) for the packed causal path.
+ * doc_start (nullable, requires B == 1): int32 [S] per-token document start
+ * indices (doc_start[t] = cu_seqlens[i] for t in document i) selecting the
+ * packed-varlen block-diagonal causal mask — the reference's flash-attn
+ * varlen cu_seqlens path (flash.py:61-91, data_collator.py:50). */
 int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                      uint16_t* O, float* LSE, int B, int Hq, int Hkv,
-                     int64_t S, float scale, void* stream);
+                     int64_t S, float scale, const int32_t* doc_start,
+                     void* stream);
 
 /* Backward preprocess: delta[r] = rowsum(dO[r] * O[r]), lse2[r] = LSE[r]*log2e
  * over rows = B*Hq*S flattened. */
@@ -184,22 +189,27 @@ int vh_attn_bwd_pre_bf16(const uint16_t* dO, const uint16_t* O,
                          const float* LSE, float* delta, float* lse2,
                          int64_t rows, void* stream);
 
-/* Backward: dQacc [B,Hq,S,128] fp32 (caller zero-fills; atomically
- * accumulated), dK/dV [B,Hq,S,128] bf16 written per Q-head — the host sums
- * GQA head groups down to [B,Hkv,S,128]. delta/lse2 from the preprocess. */
+/* Backward (monolithic PROBE variant): dQacc [B,Hq,S,128] fp32 (caller
+ * zero-fills; atomically accumulated), dK/dV [B,Hq,S,128] bf16 per Q-head
+ * (caller sums GQA groups). delta/lse2 from the preprocess. */
 int vh_attn_bwd_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                      const uint16_t* dO, const float* delta, const float* lse2,
                      float* dQacc, uint16_t* dK, uint16_t* dV, int B, int Hq,
                      int Hkv, int64_t S, float scale, void* stream);
 
-/* Split backward (v2, the dispatched path): dk/dv kernel (block owns 128 kv
- * rows) + dq kernel (block owns 128 q rows, dQ written once as bf16 — no
- * atomics, no fp32 accumulator). dK/dV still per Q-head (host sums GQA). */
+/* Split backward (the dispatched path): GQA-folded dk/dv kernel (block owns
+ * 128 kv rows of one KV head and loops the whole Hq/Hkv head group — dK/dV
+ * are [B,Hkv,S,128] bf16 written once, no per-Q-head intermediates) + dq
+ * kernel (block owns 128 q rows, dQ [B,Hq,S,128] bf16 written once — no
+ * atomics, no fp32 accumulator). doc_start/doc_end (nullable together,
+ * require B == 1): int32 [S] per-token document bounds for the packed-varlen
+ * block-diagonal causal mask (see vh_attn_fwd_bf16). */
 int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K, const uint16_t* V,
                       const uint16_t* dO, const float* delta,
                       const float* lse2, uint16_t* dQ, uint16_t* dK,
                       uint16_t* dV, int B, int Hq, int Hkv, int64_t S,
-                      float scale, void* stream);
+                      float scale, const int32_t* doc_start,
+                      const int32_t* doc_end, void* stream);
 
 /* ---- Fused optimizer ----------------------------------------------------- */
 

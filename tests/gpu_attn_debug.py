@@ -16,18 +16,7 @@ from veomni_amd.ops import hip_lib as L
 
 
 def attn_fwd(q, k, v, scale):
-    lib = L.get_lib()
-    fn = lib.vh_attn_fwd_bf16
-    fn.restype = ctypes.c_int
-    fn.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 + [ctypes.c_int64, ctypes.c_float, ctypes.c_void_p]
-    B, Hq, S, D = q.shape
-    Hkv = k.shape[1]
-    o = torch.empty_like(q)
-    lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
-    rc = fn(q.contiguous().data_ptr(), k.contiguous().data_ptr(), v.contiguous().data_ptr(),
-            o.data_ptr(), lse.data_ptr(), B, Hq, Hkv, S, scale, L.cur_stream())
-    assert rc == 0, lib.vh_last_error()
-    return o, lse
+    return L.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), scale)
 
 
 def ref_attn(q, k, v, scale):
@@ -128,6 +117,31 @@ def main():
     dt = (time.perf_counter() - t0) / 10
     flb = 3 * 2 * 2 * S * S * Hq * 128 * 0.5
     print(f"bwd llama shape: {dt*1e3:.3f} ms  {flb/dt/1e12:.0f} TF/s", flush=True)
+
+    # A/B: round-1 per-Q-head dkv (v2 probe) vs the GQA-folded v5 in the path
+    lib = L.get_lib()
+    fn2 = lib.vh_attn_bwd2_dkv2probe_bf16
+    fn2.restype = ctypes.c_int
+    fn2.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 3 + [
+        ctypes.c_int64, ctypes.c_float, ctypes.c_void_p]
+    rows = B * Hq * S
+    delta = torch.zeros(rows, dtype=torch.float32, device=dev)
+    lse2 = (lse.flatten() * 1.4426950408889634).contiguous()
+    dkh = torch.empty(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    dvh = torch.empty(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    for _ in range(3):
+        fn2(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+            delta.data_ptr(), lse2.data_ptr(), dkh.data_ptr(), dvh.data_ptr(),
+            B, Hq, Hkv, S, scale, L.cur_stream())
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        fn2(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+            delta.data_ptr(), lse2.data_ptr(), dkh.data_ptr(), dvh.data_ptr(),
+            B, Hq, Hkv, S, scale, L.cur_stream())
+    torch.cuda.synchronize()
+    dt2p = (time.perf_counter() - t0) / 10
+    print(f"dkv v2 probe (per-Q-head): {dt2p*1e3:.3f} ms", flush=True)
 
     # timing at llama shape
     B, Hq, Hkv, S = 1, 32, 8, 4096

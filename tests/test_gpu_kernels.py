@@ -348,6 +348,81 @@ def test_flash_attention_pair_vs_autograd(lib):
             assert err < 0.06 * max(m, 1.0), f"{nm}: max err {err} vs |ref|max {m}"
 
 
+def test_flash_attention_varlen_packed(lib):
+    """Packed 3-document batch through the HIP varlen path (doc_start/doc_end
+    from cu_seqlens) vs per-document fp32 attention — fwd AND grads must
+    bit-match document boundaries (no cross-document leakage).
+    Ref: attention/flash.py:61-91, data_collator.py:50."""
+    import math
+    from veomni_amd.ops.kernels.attention import (docs_from_cu_seqlens,
+                                                  hip_flash_attention)
+
+    torch.manual_seed(11)
+    S = 768
+    # misaligned boundaries (not multiples of the 32/64/128 tile sizes)
+    cu = torch.tensor([0, 200, 520, 768], dtype=torch.int32, device="cuda")
+    ds, de = docs_from_cu_seqlens(cu, S)
+    B, Hq, Hkv = 1, 4, 2
+    q = bf(torch.randn(B, Hq, S, 128) * 0.5).cuda().requires_grad_(True)
+    k = bf(torch.randn(B, Hkv, S, 128) * 0.5).cuda().requires_grad_(True)
+    v = bf(torch.randn(B, Hkv, S, 128) * 0.5).cuda().requires_grad_(True)
+    do = bf(torch.randn(B, Hq, S, 128) * 0.5).cuda()
+    scale = 1.0 / math.sqrt(128)
+    out = hip_flash_attention(q, k, v, scale, ds, de)
+    out.backward(do)
+
+    rep = Hq // Hkv
+    for i in range(cu.numel() - 1):
+        a, b = int(cu[i]), int(cu[i + 1])
+        qf = q.detach()[:, :, a:b].float().requires_grad_(True)
+        kf = k.detach()[:, :, a:b].float().requires_grad_(True)
+        vf = v.detach()[:, :, a:b].float().requires_grad_(True)
+        kk = kf.repeat_interleave(rep, dim=1)
+        vv = vf.repeat_interleave(rep, dim=1)
+        sc = torch.matmul(qf, kk.transpose(-1, -2)) * scale
+        L = b - a
+        mask = torch.triu(torch.ones(L, L, dtype=torch.bool, device="cuda"), 1)
+        p = torch.softmax(sc.masked_fill(mask, float("-inf")), dim=-1)
+        oref = torch.matmul(p, vv)
+        oref.backward(do[:, :, a:b].float())
+        torch.testing.assert_close(out[:, :, a:b].float(), oref,
+                                   rtol=0, atol=3e-2)
+        for got, ref, nm in ((q.grad[:, :, a:b], qf.grad, "dq"),
+                             (k.grad[:, :, a:b], kf.grad, "dk"),
+                             (v.grad[:, :, a:b], vf.grad, "dv")):
+            m = ref.abs().max().item()
+            err = (got.float() - ref).abs().max().item()
+            assert err < 0.06 * max(m, 1.0), \
+                f"doc {i} {nm}: max err {err} vs |ref|max {m}"
+
+
+def test_attention_slot_varlen_vs_eager_model(lib):
+    """Model-level packed batch: HIP ops (hip_flash core) vs eager on the
+    same packed 3-doc batch — the collator's cu_seq_lens kwargs consumed
+    end-to-end."""
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.ops import HIP_OPS_CONFIG
+
+    torch.manual_seed(3)
+    model = build_model("tiny-d128", dtype=torch.bfloat16, device="cuda")
+    model.eval()
+    S = 512
+    cu = torch.tensor([0, 150, 380, 512], dtype=torch.int32, device="cuda")
+    ids = torch.randint(0, model.config.vocab_size, (1, S), device="cuda")
+    pos = torch.cat([torch.arange(int(cu[i + 1]) - int(cu[i]), device="cuda")
+                     for i in range(cu.numel() - 1)])[None]
+    try:
+        with torch.no_grad():
+            bind_ops("eager")
+            ref, _ = model(ids, position_ids=pos, cu_seq_lens_q=cu)
+            bind_ops(HIP_OPS_CONFIG)
+            got, _ = model(ids, position_ids=pos, cu_seq_lens_q=cu)
+    finally:
+        bind_ops("eager")
+    torch.testing.assert_close(got.float(), ref.float(), rtol=5e-2, atol=5e-1)
+
+
 def test_veadamw_vs_torch_fused(lib):
     """One-sweep HIP AdamW vs torch fused AdamW (both bf16 state), incl. the
     grad_scale (clip-fold) path."""

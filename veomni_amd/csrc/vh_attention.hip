@@ -59,11 +59,16 @@ __device__ __forceinline__ float xor32h(float v, int half) {
   return __builtin_bit_cast(float, swap32_u(__builtin_bit_cast(uint32_t, v), half));
 }
 
-template <int MODE>
+// DOC: packed-varlen (block-diagonal causal) masking via per-token document
+// start indices (doc_start[t] = cu_seqlens[i] for t in document i; B == 1).
+// Replaces the reference's flash-attn varlen cu_seqlens path
+// (ops/kernels/attention/flash.py:61-91, kwargs from data_collator.py:50).
+template <int MODE, bool DOC>
 __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
-    float* __restrict__ LSE, int B, int Hq, int Hkv, int64_t S, float scale) {
+    float* __restrict__ LSE, const int* __restrict__ doc_start, int B, int Hq,
+    int Hkv, int64_t S, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   auto kt = [&](int buf) {                                   // 2 x 16 KiB
     return reinterpret_cast<bf16_t*>(smem + buf * 16384);
@@ -92,6 +97,15 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 
   const int64_t q_global = (int64_t)qb * QB + wave * WQ + col;
   const float scale2 = scale * 1.4426950408889634f;  // log2(e)
+
+  // varlen: this lane's document start, and the wave/block minima (doc_start
+  // is monotone non-decreasing, so the first row's value is the minimum)
+  int ds_l = 0, ds_wave = 0, t0 = 0;
+  if (DOC) {
+    ds_l = doc_start[q_global];
+    ds_wave = doc_start[(int64_t)qb * QB + wave * WQ];
+    t0 = doc_start[(int64_t)qb * QB] / KB;
+  }
 
   // ---- Q fragments straight into registers (tile-invariant: chunk c =
   // Q[q = col][c*16 + half*8 .. +8))
@@ -126,12 +140,13 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 
   const int t_max = (int)(((int64_t)qb * QB + QB - 1) / KB);  // inclusive
 
-  // prologue: stage tile 0 synchronously
+  // prologue: stage tile t0 synchronously
   {
 #pragma unroll
-    for (int i = 0; i < 2; ++i) glds16a(ksrc[i], kt(0) + klds[i]);
-    bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vsrc);
-    bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vsrc + 64);
+    for (int i = 0; i < 2; ++i)
+      glds16a(ksrc[i] + (int64_t)t0 * KB * DH, kt(0) + klds[i]);
+    bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH);
+    bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH + 64);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int r0 = v_d0 + j, r1 = v_d0 + 64 + j;
@@ -142,7 +157,7 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
   __syncthreads();
 
   int cur = 0;
-  for (int t = 0; t <= t_max; ++t) {
+  for (int t = t0; t <= t_max; ++t) {
     // ---- T14 split: issue next tile's loads before this tile's MFMAs
     bf16x8 vn0, vn1;
     const bool more = t < t_max;
@@ -156,7 +171,10 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 
     const bool diag = ((int64_t)(t + 1) * KB) > ((int64_t)qb * QB + wave * WQ);
     // causal skip: every kv in this tile is beyond every q of this wave
-    const bool live = (int64_t)t * KB <= (int64_t)qb * QB + wave * WQ + (WQ - 1);
+    // (varlen adds: or before every document of this wave)
+    const bool live =
+        ((int64_t)t * KB <= (int64_t)qb * QB + wave * WQ + (WQ - 1)) &&
+        (!DOC || (int64_t)(t + 1) * KB > ds_wave);
     const bf16_t* ktc = kt(cur);
     const bf16_t* vtc = vt(cur);
 
@@ -184,10 +202,14 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
       // bookkeeping drops one multiply per element)
       float p[16];
       const int kv_lim = (int)(q_global - (int64_t)t * KB) - sub * 32 - 4 * half;
+      const int lo_lim = DOC ? ds_l - (int)((int64_t)t * KB) - sub * 32 - 4 * half
+                             : -2147483647;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         float sc = sacc[r] * scale2;
-        if (diag && ((r & 3) + 8 * (r >> 2)) > kv_lim) sc = -INFINITY;
+        int idx = (r & 3) + 8 * (r >> 2);
+        if (diag && idx > kv_lim) sc = -INFINITY;
+        if (DOC && idx < lo_lim) sc = -INFINITY;
         p[r] = sc;
       }
       float m8[8], m4[4];
@@ -324,16 +346,27 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 extern "C" int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K,
                                 const uint16_t* V, uint16_t* O, float* LSE,
                                 int B, int Hq, int Hkv, int64_t S, float scale,
-                                void* stream) {
+                                const int32_t* doc_start, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % QB == 0, "S %% 256 != 0 (pad the sequence)");
   VH_CHECK(Hq % Hkv == 0, "Hq %% Hkv != 0");
+  VH_CHECK(doc_start == nullptr || B == 1, "varlen requires packed B == 1");
   dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
-  hipLaunchKernelGGL(k_attn_fwd<0>, grid, dim3(512), 65536, s,
-                     reinterpret_cast<const bf16_t*>(Q),
-                     reinterpret_cast<const bf16_t*>(K),
-                     reinterpret_cast<const bf16_t*>(V),
-                     reinterpret_cast<bf16_t*>(O), LSE, B, Hq, Hkv, S, scale);
+  if (doc_start) {
+    hipLaunchKernelGGL((k_attn_fwd<0, true>), grid, dim3(512), 65536, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<bf16_t*>(O), LSE, doc_start, B, Hq,
+                       Hkv, S, scale);
+  } else {
+    hipLaunchKernelGGL((k_attn_fwd<0, false>), grid, dim3(512), 65536, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<bf16_t*>(O), LSE, nullptr, B, Hq, Hkv,
+                       S, scale);
+  }
   VH_HIP(hipGetLastError());
   return 0;
 }
@@ -346,7 +379,7 @@ extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
                                       void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
-#define VH_AM(M_)                                                                hipLaunchKernelGGL(k_attn_fwd<M_>, grid, dim3(512), 65536, s,                                     reinterpret_cast<const bf16_t*>(Q),                                            reinterpret_cast<const bf16_t*>(K),                                            reinterpret_cast<const bf16_t*>(V),                                            reinterpret_cast<bf16_t*>(O), LSE, B, Hq, Hkv, S, scale)
+#define VH_AM(M_)                                                                hipLaunchKernelGGL((k_attn_fwd<M_, false>), grid, dim3(512), 65536, s,                                     reinterpret_cast<const bf16_t*>(Q),                                            reinterpret_cast<const bf16_t*>(K),                                            reinterpret_cast<const bf16_t*>(V),                                            reinterpret_cast<bf16_t*>(O), LSE, nullptr, B, Hq, Hkv, S, scale)
   if (mode == 1) VH_AM(1);
   else if (mode == 2) VH_AM(2);
   else if (mode == 3) VH_AM(3);
@@ -1217,11 +1250,215 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
   }
 }
 
+// dkv v5 (the dispatched variant): GQA-folded + register-prefetched A-operands.
+//   - grid (S/128, B*Hkv): a block owns 128 kv rows of ONE KV head and loops
+//     the whole GQA head group (rep = Hq/Hkv) — dK/dV are written ONCE to the
+//     [B,Hkv,S,D] buffers (the per-Q-head [B,Hq,S,D] intermediates and the
+//     host-side group sum of dkv v2 are gone: 1/rep the dK/dV HBM traffic),
+//     and the K/V LDS staging amortizes over the group.
+//   - the or2 A-operands (the lane's Q/dO rows) are prefetched into registers
+//     per q tile BEFORE the transposed-LDS staging, so the MFMA loop reads
+//     only registers + LDS — dkv v2 issued 16 global loads inside the MFMA
+//     loop per tile, serializing on L2 latency at 2 waves/SIMD (the dq kernel,
+//     whose MFMA loop is register/LDS-only, runs 2.4x more efficient).
+//   - DOC: packed-varlen block-diagonal causal via doc_start/doc_end
+//     (per-token document bounds; see k_attn_fwd).
+template <bool DOC>
+__global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
+    const float* __restrict__ delta, const float* __restrict__ lse2,
+    bf16_t* __restrict__ dK, bf16_t* __restrict__ dV,
+    const int* __restrict__ doc_start, const int* __restrict__ doc_end, int B,
+    int Hq, int Hkv, int64_t S, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [128][128] 32 K
+  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 32768);    // [128][128] 32 K
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 65536);     // [128][32] 8 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 73728);    // [128][32] 8 K
+
+  const int kvb = blockIdx.x;
+  const int bkh = blockIdx.y;          // b * Hkv + hkv
+  const int b = bkh / Hkv;
+  const int hkv = bkh % Hkv;
+  const int rep = Hq / Hkv;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 32-row kv slice
+  const int half = lane >> 5;
+  const int col = lane & 31;
+
+  const bf16_t* Kb = K + (((int64_t)b * Hkv + hkv) * S) * DH;
+  const bf16_t* Vb = V + (((int64_t)b * Hkv + hkv) * S) * DH;
+  bf16_t* dKb = dK + (((int64_t)b * Hkv + hkv) * S) * DH;
+  bf16_t* dVb = dV + (((int64_t)b * Hkv + hkv) * S) * DH;
+
+  const int64_t kv0 = (int64_t)kvb * 128;
+  const int kvrow_l = wave * 32 + col;
+
+  // stage K/V rows once per block (256 thr: 8 passes of 4 KiB)
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int o = i * 4096 + tid * 16;
+    int row = o >> 8;
+    int colb = o & 255;
+    *reinterpret_cast<bf16x8*>(
+        reinterpret_cast<char*>(krow) + row * 256 + kswz(row, colb)) =
+        *reinterpret_cast<const bf16x8*>(Kb + (kv0 + row) * DH + (colb >> 1));
+    *reinterpret_cast<bf16x8*>(
+        reinterpret_cast<char*>(vrow) + row * 256 + kswz(row, colb)) =
+        *reinterpret_cast<const bf16x8*>(Vb + (kv0 + row) * DH + (colb >> 1));
+  }
+
+  f32x16 dv_acc[4], dk_acc[4];
+#pragma unroll
+  for (int d = 0; d < 4; ++d) {
+    dv_acc[d] = f32x16{};
+    dk_acc[d] = f32x16{};
+  }
+
+  const float scale2 = scale * 1.4426950408889634f;
+  const int qt0 = (int)(kv0 / 32);
+  int qtn = (int)(S / 32);
+  int de_wave = 0;
+  if (DOC) {
+    // block-uniform loop end: last kv row's document end (doc_end monotone)
+    qtn = (doc_end[kv0 + 127] + 31) / 32;
+    de_wave = doc_end[kv0 + wave * 32 + 31];
+  }
+
+  for (int g = 0; g < rep; ++g) {
+    const int hq = hkv * rep + g;
+    const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+    const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+    const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+    const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+
+    for (int qt = qt0; qt < qtn; ++qt) {
+      const int64_t q0 = (int64_t)qt * 32;
+      // prefetch this lane's or2 A-operand rows (q = q0+col) into registers
+      // BEFORE the staging stores so the L2 latency hides under them
+      bf16frag qrow[8], dorow[8];
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        qrow[c] = *reinterpret_cast<const bf16frag*>(
+            Qb + (q0 + col) * DH + c * 16 + half * 8);
+        dorow[c] = *reinterpret_cast<const bf16frag*>(
+            dOb + (q0 + col) * DH + c * 16 + half * 8);
+      }
+      // stage Q^T / dO^T (512 units / 256 thr = 2 each)
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        int unit = tid + u * 256;
+        int q = unit & 31;
+        int d0 = (unit >> 5) * 8;
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0 + q) * DH + d0);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0 + q) * DH + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = d0 + j;
+          qtr[(row * 64 + qswz(row, q * 2)) >> 1] = vq.v[j];
+          dotr[(row * 64 + qswz(row, q * 2)) >> 1] = vd.v[j];
+        }
+      }
+      __syncthreads();
+
+      const bool live = ((q0 + 31) >= (kv0 + wave * 32)) &&
+                        (!DOC || q0 < de_wave);
+      const bool diag = (q0 < kv0 + 128);
+
+      if (live) {
+        f32x16 s2 = f32x16{}, dp2 = f32x16{};
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          int colb = (c * 16 + half * 8) * 2;
+          bf16frag kf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+          bf16frag vf = *reinterpret_cast<const bf16frag*>(
+              reinterpret_cast<const char*>(vrow) + kvrow_l * 256 + kswz(kvrow_l, colb));
+          s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qrow[c], kf, s2, 0, 0, 0);
+          dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dorow[c], vf, dp2, 0, 0, 0);
+        }
+        uint32_t pk2[8], dg2[8];
+#pragma unroll
+        for (int r = 0; r < 16; r += 2) {
+          float pv[2], gv[2];
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr) {
+            int r2 = r + rr;
+            int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
+            bool masked = diag && ((int64_t)kv0 + wave * 32 + col > q0 + qrm);
+            if (DOC)
+              masked = masked ||
+                       ((int64_t)kv0 + wave * 32 + col < doc_start[q0 + qrm]);
+            float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lseb[q0 + qrm]);
+            pv[rr] = pp;
+            gv[rr] = pp * (dp2[r2] - delb[q0 + qrm]) * scale;
+          }
+          pk2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[0]) |
+                        ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[1]) << 16);
+          dg2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[0]) |
+                        ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)gv[1]) << 16);
+        }
+        bf16frag pa2[2], da2[2];
+#pragma unroll
+        for (int mch = 0; mch < 2; ++mch) {
+          uint32_t a0 = half ? pk2[4 * mch] : pk2[4 * mch + 2];
+          uint32_t a1 = half ? pk2[4 * mch + 1] : pk2[4 * mch + 3];
+          uint32_t b0 = swap32_u(a0, half);
+          uint32_t b1 = swap32_u(a1, half);
+          uint4 u{half ? b0 : pk2[4 * mch], half ? b1 : pk2[4 * mch + 1],
+                  half ? pk2[4 * mch + 2] : b0, half ? pk2[4 * mch + 3] : b1};
+          pa2[mch] = __builtin_bit_cast(bf16frag, u);
+          uint32_t c0 = half ? dg2[4 * mch] : dg2[4 * mch + 2];
+          uint32_t c1 = half ? dg2[4 * mch + 1] : dg2[4 * mch + 3];
+          uint32_t e0 = swap32_u(c0, half);
+          uint32_t e1 = swap32_u(c1, half);
+          uint4 u2{half ? e0 : dg2[4 * mch], half ? e1 : dg2[4 * mch + 1],
+                   half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
+          da2[mch] = __builtin_bit_cast(bf16frag, u2);
+        }
+#pragma unroll
+        for (int mch = 0; mch < 2; ++mch) {
+#pragma unroll
+          for (int dblk = 0; dblk < 4; ++dblk) {
+            int trow = dblk * 32 + col;
+            int colb = (mch * 16 + half * 8) * 2;
+            bf16frag dof = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(dotr) + trow * 64 + qswz(trow, colb));
+            bf16frag qf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(qtr) + trow * 64 + qswz(trow, colb));
+            dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
+            dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // single-writer epilogue: this block is the only contributor to its
+  // [hkv, kv rows] slice (the whole GQA group was folded in-register)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int kvr = (r & 3) + 8 * (r >> 2) + 4 * half;
+    int64_t kvg = kv0 + wave * 32 + kvr;
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      dKb[kvg * DH + d * 32 + col] = f2bf(dk_acc[d][r]);
+      dVb[kvg * DH + d * 32 + col] = f2bf(dv_acc[d][r]);
+    }
+  }
+}
+
+template <bool DOC>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
     const float* __restrict__ delta, const float* __restrict__ lse2,
-    bf16_t* __restrict__ dQ, int B, int Hq, int Hkv, int64_t S, float scale) {
+    bf16_t* __restrict__ dQ, const int* __restrict__ doc_start, int B, int Hq,
+    int Hkv, int64_t S, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [32][128] 8 K
   bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 8192);     // [32][128] 8 K
@@ -1261,12 +1498,19 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
   const float del_l = delb[q_l];
   const float scale2 = scale * 1.4426950408889634f;
 
+  int ds_l = 0, ds_wave = 0, kvt_begin = 0;
+  if (DOC) {
+    ds_l = doc_start[q_l];
+    ds_wave = doc_start[q0b + wave * 32];
+    kvt_begin = doc_start[q0b] / 32;
+  }
+
   f32x16 dq4[4];
 #pragma unroll
   for (int d = 0; d < 4; ++d) dq4[d] = f32x16{};
 
   const int kvtn = (int)((q0b + 128) / 32);
-  for (int kvt = 0; kvt < kvtn; ++kvt) {
+  for (int kvt = kvt_begin; kvt < kvtn; ++kvt) {
     const int64_t kvt0 = (int64_t)kvt * 32;
     // stage K/V rows [32][128] (2 passes) + K^T [128][32] (2 units each)
 #pragma unroll
@@ -1295,7 +1539,8 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     }
     __syncthreads();
 
-    const bool live = kvt0 <= q0b + wave * 32 + 31;
+    const bool live = (kvt0 <= q0b + wave * 32 + 31) &&
+                      (!DOC || kvt0 + 31 >= ds_wave);
     const bool diag = (kvt0 + 31 >= q0b + wave * 32);
 
     if (live) {
@@ -1320,6 +1565,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
           int r2 = r + rr;
           int kvl = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
           bool masked = diag && (kvt0 + kvl > q_l);
+          if (DOC) masked = masked || (kvt0 + kvl < ds_l);
           float pp = masked ? 0.f : __builtin_exp2f(s1[r2] * scale2 - lse_l);
           g[rr] = pp * (dp1[r2] - del_l) * scale;
         }
@@ -1387,20 +1633,75 @@ extern "C" int vh_attn_bwd2_dkv4probe_bf16(const uint16_t* Q, const uint16_t* K,
   return 0;
 }
 
+/* Dispatched split backward (v5): GQA-folded dkv (dK/dV [B,Hkv,S,D] written
+ * once — no per-Q-head intermediates, no host group sum) + per-Q-head dq.
+ * doc_start/doc_end (nullable, B == 1) select the packed-varlen
+ * block-diagonal causal mask. */
 extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                                  const uint16_t* V, const uint16_t* dO,
                                  const float* delta, const float* lse2,
                                  uint16_t* dQ, uint16_t* dK, uint16_t* dV,
                                  int B, int Hq, int Hkv, int64_t S, float scale,
-                                 void* stream) {
+                                 const int32_t* doc_start,
+                                 const int32_t* doc_end, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(S % 128 == 0, "S %% 128 != 0");
+  VH_CHECK(Hq % Hkv == 0, "Hq %% Hkv != 0");
+  VH_CHECK((doc_start == nullptr) == (doc_end == nullptr),
+           "doc_start/doc_end must be passed together");
+  VH_CHECK(doc_start == nullptr || B == 1, "varlen requires packed B == 1");
+  dim3 grid_kv((uint32_t)(S / 128), (uint32_t)(B * Hkv));
+  dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
+  if (doc_start) {
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true>), grid_kv, dim3(256), 81920, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dK),
+                       reinterpret_cast<bf16_t*>(dV), doc_start, doc_end, B,
+                       Hq, Hkv, S, scale);
+    VH_HIP(hipGetLastError());
+    hipLaunchKernelGGL((k_attn_bwd_dq<true>), grid_q, dim3(256), 24576, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
+                       S, scale);
+  } else {
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false>), grid_kv, dim3(256), 81920, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dK),
+                       reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
+                       Hkv, S, scale);
+    VH_HIP(hipGetLastError());
+    hipLaunchKernelGGL((k_attn_bwd_dq<false>), grid_q, dim3(256), 24576, s,
+                       reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
+                       scale);
+  }
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* probe: the round-1 per-Q-head dkv (v2) kept for on-box A/B against the
+ * GQA-folded v5 (dK/dV here are per-Q-head [B,Hq,S,D]). */
+extern "C" int vh_attn_bwd2_dkv2probe_bf16(const uint16_t* Q, const uint16_t* K,
+                                           const uint16_t* V, const uint16_t* dO,
+                                           const float* delta, const float* lse2,
+                                           uint16_t* dK, uint16_t* dV, int B,
+                                           int Hq, int Hkv, int64_t S,
+                                           float scale, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
-  // dkv4 (64-kv strips / 64-q tiles, parity-green) measured 2.43 ms vs
-  // dkv's 2.35 at the llama shape: the halved barrier cadence is offset by
-  // doubled q-tile staging volume (each q tile re-staged for 2x as many kv
-  // strips). Kept compiled as the validated starting point for a round-2
-  // variant that also double-buffers; dkv stays dispatched.
   hipLaunchKernelGGL(k_attn_bwd_dkv, grid, dim3(256), 81920, s,
                      reinterpret_cast<const bf16_t*>(Q),
                      reinterpret_cast<const bf16_t*>(K),
@@ -1408,13 +1709,6 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<const bf16_t*>(dO), delta, lse2,
                      reinterpret_cast<bf16_t*>(dK),
                      reinterpret_cast<bf16_t*>(dV), B, Hq, Hkv, S, scale);
-  VH_HIP(hipGetLastError());
-  hipLaunchKernelGGL(k_attn_bwd_dq, grid, dim3(256), 24576, s,
-                     reinterpret_cast<const bf16_t*>(Q),
-                     reinterpret_cast<const bf16_t*>(K),
-                     reinterpret_cast<const bf16_t*>(V),
-                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,
-                     reinterpret_cast<bf16_t*>(dQ), B, Hq, Hkv, S, scale);
   VH_HIP(hipGetLastError());
   return 0;
 }

@@ -38,6 +38,13 @@ PRESETS = {
         num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
         head_dim=32, qk_norm=False,
     ),
+    # full-width head_dim at test scale: exercises the HIP flash pair
+    # (D=128 requirement) without a BASELINE-sized model
+    "tiny-d128": ModelConfig(
+        name="tiny-d128", vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+        head_dim=128, qk_norm=False,
+    ),
     # qwen2-architecture knobs at test scale: attention bias on, no qk norm
     "tiny-qwen2": ModelConfig(
         name="tiny-qwen2", vocab_size=512, hidden_size=128, intermediate_size=256,

@@ -132,16 +132,31 @@ def repeat_kv(hidden_states: torch.Tensor, n_rep: int) -> torch.Tensor:
 
 
 def sdpa_attention(module, query, key, value, attention_mask, dropout=0.0,
-                   scaling=None, **kwargs):
+                   scaling=None, doc_start=None, **kwargs):
     """Eager/SDPA attention on [B,h,S,D] (HF sdpa semantics). GQA repeat is
     derived from the shapes (under Ulysses the per-rank head counts differ
-    from the module's config attrs)."""
+    from the module's config attrs). doc_start (int32 [S]) applies the
+    packed-varlen block-diagonal causal mask (the reference flash-attn
+    cu_seqlens semantics, attention/flash.py:61-91) as an explicit mask —
+    this is the eager oracle for the HIP varlen kernels."""
     n_rep = query.shape[1] // key.shape[1]
     key = repeat_kv(key, n_rep)
     value = repeat_kv(value, n_rep)
-    out = F.scaled_dot_product_attention(
-        query, key, value, attn_mask=None, dropout_p=dropout, scale=scaling, is_causal=True
-    )
+    if doc_start is not None:
+        S = query.shape[2]
+        assert doc_start.numel() == S, (doc_start.numel(), S)
+        ar = torch.arange(S, device=query.device)
+        ds = doc_start.to(device=query.device, dtype=torch.long)
+        allowed = (ar[None, :] <= ar[:, None]) & (ar[None, :] >= ds[:, None])
+        out = F.scaled_dot_product_attention(
+            query, key, value, attn_mask=allowed, dropout_p=dropout,
+            scale=scaling, is_causal=False
+        )
+    else:
+        out = F.scaled_dot_product_attention(
+            query, key, value, attn_mask=None, dropout_p=dropout,
+            scale=scaling, is_causal=True
+        )
     return out.transpose(1, 2).contiguous(), None
 
 
@@ -195,20 +210,23 @@ class Attention(nn.Module):
             # a2a launched as each projection finishes; RoPE commutes with
             # the seq gather (elementwise per position), so it runs on the
             # local slice. KV heads repeat first when sp > kv heads.
-            out = self._async_ulysses_attention(q, k, v)
+            out = self._async_ulysses_attention(q, k, v, **kwargs)
         elif veomni_attention.use_non_eager_impl:
             # the bound attention kernel is SP-aware (does the Ulysses
-            # exchange itself, ref attention/flash.py:236-299)
-            out, _ = veomni_attention(self, q, k, v, None, dropout=0.0, scaling=self.scaling)
+            # exchange itself, ref attention/flash.py:236-299); varlen
+            # doc bounds ride the kwargs (ref flash.py:61-91)
+            out, _ = veomni_attention(self, q, k, v, None, dropout=0.0,
+                                      scaling=self.scaling, **kwargs)
         elif ps.ulysses_enabled:
             # eager under SP still needs the sync Ulysses exchange
-            out = self._sync_ulysses_attention(q, k, v)
+            out = self._sync_ulysses_attention(q, k, v, **kwargs)
         else:
-            out, _ = sdpa_attention(self, q, k, v, None, dropout=0.0, scaling=self.scaling)
+            out, _ = sdpa_attention(self, q, k, v, None, dropout=0.0,
+                                    scaling=self.scaling, **kwargs)
         out = out.reshape(*input_shape, -1).contiguous()
         return self.o_proj(out)
 
-    def _sync_ulysses_attention(self, q, k, v):
+    def _sync_ulysses_attention(self, q, k, v, **kwargs):
         from ..distributed.sequence_parallel import (
             gather_heads_scatter_seq,
             gather_seq_scatter_heads,
@@ -231,11 +249,11 @@ class Attention(nn.Module):
                                       seq_dim=0, head_dim=1, group=group)
         out, _ = sdpa_attention(self, qs.transpose(0, 1)[None], ks.transpose(0, 1)[None],
                                 vs.transpose(0, 1)[None], None, dropout=0.0,
-                                scaling=self.scaling)  # [1, S, h/sp, D]
+                                scaling=self.scaling, **kwargs)  # [1, S, h/sp, D]
         out = gather_heads_scatter_seq(out.squeeze(0), head_dim=1, seq_dim=0, group=group)
         return out[None]
 
-    def _async_ulysses_attention(self, q, k, v):
+    def _async_ulysses_attention(self, q, k, v, **kwargs):
         from ..distributed.sequence_parallel import (
             gather_heads_scatter_seq,
             gather_seq_scatter_heads_async,
@@ -262,7 +280,7 @@ class Attention(nn.Module):
         kg = wait_gathered(kb, group=group).transpose(0, 1)[None]
         vg = wait_gathered(vb, group=group).transpose(0, 1)[None]
         out, _ = sdpa_attention(self, qg, kg, vg, None, dropout=0.0,
-                                scaling=self.scaling)  # [1, S, h/sp, D]
+                                scaling=self.scaling, **kwargs)  # [1, S, h/sp, D]
         out = out.squeeze(0)
         out = gather_heads_scatter_seq(out, head_dim=1, seq_dim=0, group=group)
         return out[None]
@@ -390,10 +408,11 @@ class DecoderLayer(nn.Module):
         self.post_attention_layernorm = RMSNorm(config.hidden_size, eps=config.rms_norm_eps)
         self.is_moe = config.is_moe
 
-    def forward(self, hidden_states, position_embeddings):
+    def forward(self, hidden_states, position_embeddings, attn_kwargs=None):
         residual = hidden_states
         hidden_states = self.input_layernorm(hidden_states)
-        hidden_states = self.self_attn(hidden_states, position_embeddings)
+        hidden_states = self.self_attn(hidden_states, position_embeddings,
+                                       **(attn_kwargs or {}))
         hidden_states = residual + hidden_states
 
         residual = hidden_states
@@ -418,7 +437,8 @@ class Model(nn.Module):
         self.norm = RMSNorm(config.hidden_size, eps=config.rms_norm_eps)
         self.rotary_emb = RotaryEmbedding(config)
 
-    def forward(self, input_ids, position_ids=None, use_checkpoint=False):
+    def forward(self, input_ids, position_ids=None, use_checkpoint=False,
+                attn_kwargs=None):
         hidden = self.embed_tokens(input_ids)
         if position_ids is None:
             position_ids = torch.arange(input_ids.shape[1], device=input_ids.device)[None]
@@ -427,10 +447,10 @@ class Model(nn.Module):
         for layer in self.layers:
             if use_checkpoint and self.training:
                 h, rl = torch.utils.checkpoint.checkpoint(
-                    layer, hidden, pos_emb, use_reentrant=False
+                    layer, hidden, pos_emb, attn_kwargs, use_reentrant=False
                 )
             else:
-                h, rl = layer(hidden, pos_emb)
+                h, rl = layer(hidden, pos_emb, attn_kwargs)
             hidden = h
             if rl is not None:
                 router_logits.append(rl)
@@ -493,8 +513,21 @@ class ForCausalLM(nn.Module):
 
     # ------------------------------------------------------------- forward
     def forward(self, input_ids, labels=None, position_ids=None, **kwargs):
+        # packed-varlen: turn the collator's cu_seq_lens kwargs
+        # (data_collator.py:50, PackingCollator) into per-token document
+        # bounds ONCE per step; the attention slot applies the
+        # block-diagonal causal mask (ref attention/flash.py:61-91).
+        attn_kwargs = None
+        cu = kwargs.get("cu_seq_lens_q")
+        if cu is not None:
+            from ..ops.kernels.attention import docs_from_cu_seqlens
+
+            ds, de = docs_from_cu_seqlens(cu, int(cu.flatten()[-1]))
+            if ds is not None:
+                attn_kwargs = {"doc_start": ds, "doc_end": de}
         hidden, router_logits = self.model(
-            input_ids, position_ids=position_ids, use_checkpoint=self.use_checkpoint
+            input_ids, position_ids=position_ids,
+            use_checkpoint=self.use_checkpoint, attn_kwargs=attn_kwargs
         )
         if labels is None:
             logits = self.lm_head(hidden)

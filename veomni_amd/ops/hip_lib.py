@@ -70,12 +70,12 @@ def get_lib() -> ctypes.CDLL:
     _sig(lib, "vh_adamw_bf16", c_p, c_p, c_p, c_p, c_p, c_int, c_i64, c_f32,
          c_f32, c_f32, c_f32, c_f32, c_int, c_p, c_p)
     _sig(lib, "vh_attn_fwd_bf16", c_p, c_p, c_p, c_p, c_p, c_int, c_int, c_int,
-         c_i64, c_f32, c_p)
+         c_i64, c_f32, c_p, c_p)
     _sig(lib, "vh_attn_bwd_pre_bf16", c_p, c_p, c_p, c_p, c_p, c_i64, c_p)
     _sig(lib, "vh_attn_bwd_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p,
          c_int, c_int, c_int, c_i64, c_f32, c_p)
     _sig(lib, "vh_attn_bwd2_bf16", c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p, c_p,
-         c_int, c_int, c_int, c_i64, c_f32, c_p)
+         c_int, c_int, c_int, c_i64, c_f32, c_p, c_p, c_p)
     _LIB = lib
     return lib
 
@@ -352,49 +352,52 @@ def ce_fwd(logits: torch.Tensor, labels: torch.Tensor, grad_scale: float,
     return loss_rows, dlogits
 
 
-def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float):
+def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float,
+             doc_start: torch.Tensor | None = None):
     """Causal GQA flash attention forward: q [B,Hq,S,128] bf16 contiguous,
-    k/v [B,Hkv,S,128]. Returns (o, lse)."""
+    k/v [B,Hkv,S,128]. Returns (o, lse). doc_start (int32 [S], B == 1)
+    selects the packed-varlen block-diagonal mask (reference cu_seqlens
+    path, attention/flash.py:61-91)."""
     B, Hq, S, D = q.shape
     Hkv = k.shape[1]
-    assert D == 128 and S % 128 == 0, (S, D)
+    assert D == 128 and S % 256 == 0, (S, D)
+    if doc_start is not None:
+        assert B == 1 and doc_start.dtype == torch.int32 and doc_start.numel() == S
     o = torch.empty_like(q)
     lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
     with _prof("attn_fwd", 2.0 * 2 * S * S * Hq * D * 0.5 * B):
         check(get_lib().vh_attn_fwd_bf16(dptr(q), dptr(k), dptr(v), dptr(o),
                                          dptr(lse), B, Hq, Hkv, S, scale,
+                                         dptr(doc_start) if doc_start is not None else None,
                                          cur_stream()), "vh_attn_fwd")
     return o, lse
 
 
 def attn_bwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
              o: torch.Tensor, lse: torch.Tensor, do: torch.Tensor,
-             scale: float):
-    """Backward for attn_fwd. Returns (dq bf16 [B,Hq,S,D],
-    dk/dv bf16 [B,Hkv,S,D] with GQA head groups summed)."""
+             scale: float, doc_start: torch.Tensor | None = None,
+             doc_end: torch.Tensor | None = None):
+    """Backward for attn_fwd. Returns (dq bf16 [B,Hq,S,D], dk/dv bf16
+    [B,Hkv,S,D] — the GQA head group is folded inside the dkv kernel)."""
     B, Hq, S, D = q.shape
     Hkv = k.shape[1]
     rows = B * Hq * S
     do = do.contiguous()
     delta = torch.empty(rows, dtype=torch.float32, device=q.device)
     lse2 = torch.empty(rows, dtype=torch.float32, device=q.device)
+    assert (doc_start is None) == (doc_end is None)
     lib = get_lib()
     with _prof("attn_bwd", 3.0 * 2 * 2 * S * S * Hq * D * 0.5 * B):
         check(lib.vh_attn_bwd_pre_bf16(dptr(do), dptr(o), dptr(lse.contiguous()),
                                        dptr(delta), dptr(lse2), rows,
                                        cur_stream()), "vh_attn_bwd_pre")
         dq = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
-        dkh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
-        dvh = torch.empty(B, Hq, S, D, dtype=torch.bfloat16, device=q.device)
+        dk = torch.empty(B, Hkv, S, D, dtype=torch.bfloat16, device=q.device)
+        dv = torch.empty(B, Hkv, S, D, dtype=torch.bfloat16, device=q.device)
         check(lib.vh_attn_bwd2_bf16(dptr(q), dptr(k), dptr(v), dptr(do),
                                     dptr(delta), dptr(lse2), dptr(dq),
-                                    dptr(dkh), dptr(dvh), B, Hq, Hkv, S, scale,
+                                    dptr(dk), dptr(dv), B, Hq, Hkv, S, scale,
+                                    dptr(doc_start) if doc_start is not None else None,
+                                    dptr(doc_end) if doc_end is not None else None,
                                     cur_stream()), "vh_attn_bwd2")
-        rep = Hq // Hkv
-        if rep > 1:
-            # torch reduces bf16 sums in fp32 internally (acc_type)
-            dk = dkh.view(B, Hkv, rep, S, D).sum(2)
-            dv = dvh.view(B, Hkv, rep, S, D).sum(2)
-        else:
-            dk, dv = dkh, dvh
     return dq, dk, dv
