@@ -1250,20 +1250,29 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
   }
 }
 
-// dkv v5 (the dispatched variant): GQA-folded + register-prefetched A-operands.
-//   - grid (S/128, B*Hkv): a block owns 128 kv rows of ONE KV head and loops
-//     the whole GQA head group (rep = Hq/Hkv) — dK/dV are written ONCE to the
-//     [B,Hkv,S,D] buffers (the per-Q-head [B,Hq,S,D] intermediates and the
-//     host-side group sum of dkv v2 are gone: 1/rep the dK/dV HBM traffic),
-//     and the K/V LDS staging amortizes over the group.
-//   - the or2 A-operands (the lane's Q/dO rows) are prefetched into registers
-//     per q tile BEFORE the transposed-LDS staging, so the MFMA loop reads
-//     only registers + LDS — dkv v2 issued 16 global loads inside the MFMA
-//     loop per tile, serializing on L2 latency at 2 waves/SIMD (the dq kernel,
-//     whose MFMA loop is register/LDS-only, runs 2.4x more efficient).
+// dkv v6 (the dispatched variant): GQA-folded on the 64-kv-strip geometry.
+//   - grid (S/64, B*Hkv): a block owns a 64-row kv strip of ONE KV head and
+//     loops the whole GQA head group (rep = Hq/Hkv) — dK/dV are written ONCE
+//     to the [B,Hkv,S,D] buffers (no per-Q-head intermediates, no host group
+//     sum) and the K/V staging amortizes over the group. The strip split
+//     keeps the grid at >= 2 blocks/CU even at B = 1: the first fold attempt
+//     on 128-kv blocks (grid S/128 x B*Hkv = 256 blocks at the llama
+//     microbench shape) measured ~2.45 ms from half-empty wave slots vs
+//     1.78 ms for the unfolded v2 — occupancy beat traffic.
+//   - 4 waves = (2 kv slices) x (2 q subtiles of a 64-row staged q tile):
+//     each wave owns a distinct 32x32 quadrant; barrier cadence halves per
+//     unit of work vs 32-row q tiles (the dkv4 probe geometry).
+//   - PREF of the or2 A-operands (the lane's Q/dO rows): the first PREF
+//     chunks are prefetched into registers BEFORE the transposed-LDS staging
+//     stores (their L2 latency hides under the staging writes), the rest
+//     load in-loop as dkv v2/v4 did. PREF=8 costs 64 VGPRs and spills
+//     (~250 B/lane) against the 128 accumulator regs at the unified-file
+//     2-waves/SIMD budget of 256; PREF=4 fits. All three measured on-box
+//     via vh_attn_bwd2_dkv6probe_bf16.
 //   - DOC: packed-varlen block-diagonal causal via doc_start/doc_end
 //     (per-token document bounds; see k_attn_fwd).
-template <bool DOC>
+// LDS 64 KB: K/V strips 2x16 K + Q^T/dO^T tiles 2x16 K -> 2 blocks/CU.
+template <bool DOC, int PREF>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1272,12 +1281,13 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const int* __restrict__ doc_start, const int* __restrict__ doc_end, int B,
     int Hq, int Hkv, int64_t S, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [128][128] 32 K
-  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 32768);    // [128][128] 32 K
-  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 65536);     // [128][32] 8 K
-  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 73728);    // [128][32] 8 K
+  bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [64][128] 16 K
+  bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 16384);    // [64][128] 16 K
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 32768);     // [128][64] 16 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 49152);    // [128][64] 16 K
+  float* red = reinterpret_cast<float*>(smem);               // epilogue reuse
 
-  const int kvb = blockIdx.x;
+  const int kvb = blockIdx.x;          // 64-row kv strip
   const int bkh = blockIdx.y;          // b * Hkv + hkv
   const int b = bkh / Hkv;
   const int hkv = bkh % Hkv;
@@ -1285,7 +1295,9 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;           // 32-row kv slice
+  const int wave = tid >> 6;
+  const int kvslice = wave & 1;        // 32-kv slice within the strip
+  const int qsub = wave >> 1;          // 32-q subtile within the 64-q tile
   const int half = lane >> 5;
   const int col = lane & 31;
 
@@ -1294,12 +1306,12 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   bf16_t* dKb = dK + (((int64_t)b * Hkv + hkv) * S) * DH;
   bf16_t* dVb = dV + (((int64_t)b * Hkv + hkv) * S) * DH;
 
-  const int64_t kv0 = (int64_t)kvb * 128;
-  const int kvrow_l = wave * 32 + col;
+  const int64_t kv0 = (int64_t)kvb * 64;
+  const int kvrow_l = kvslice * 32 + col;
 
-  // stage K/V rows once per block (256 thr: 8 passes of 4 KiB)
+  // stage the strip's 64 K/V rows once (256 thr: 4 passes of 4 KiB each)
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < 4; ++i) {
     int o = i * 4096 + tid * 16;
     int row = o >> 8;
     int colb = o & 255;
@@ -1319,66 +1331,80 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   }
 
   const float scale2 = scale * 1.4426950408889634f;
-  const int qt0 = (int)(kv0 / 32);
-  int qtn = (int)(S / 32);
+  const int qt0 = (int)(kv0 / 64);     // 64-row q tiles
+  int qtn = (int)(S / 64);
   int de_wave = 0;
   if (DOC) {
-    // block-uniform loop end: last kv row's document end (doc_end monotone)
-    qtn = (doc_end[kv0 + 127] + 31) / 32;
-    de_wave = doc_end[kv0 + wave * 32 + 31];
+    qtn = (doc_end[kv0 + 63] + 63) / 64;          // block-uniform end
+    de_wave = doc_end[kv0 + kvslice * 32 + 31];   // wave-live bound
   }
 
-  for (int g = 0; g < rep; ++g) {
-    const int hq = hkv * rep + g;
-    const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
-    const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
-    const float* delb = delta + ((int64_t)b * Hq + hq) * S;
-    const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
-
-    for (int qt = qt0; qt < qtn; ++qt) {
-      const int64_t q0 = (int64_t)qt * 32;
-      // prefetch this lane's or2 A-operand rows (q = q0+col) into registers
-      // BEFORE the staging stores so the L2 latency hides under them
-      bf16frag qrow[8], dorow[8];
+  // single flattened loop over (head g, q tile): one live address chain —
+  // a nested g/qt loop form kept per-head pointer sets alive across the
+  // whole accumulator section and spilled ~190 extra bytes/lane
+  const int ntiles = qtn - qt0;
+  for (int it = 0; it < rep * ntiles; ++it) {
+    const int g = it / ntiles;
+    const int qt = qt0 + (it - g * ntiles);
+    {
+      const int hq = hkv * rep + g;
+      const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+      const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+      const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+      const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+      const int64_t q0t = (int64_t)qt * 64;
+      const int64_t q0 = q0t + qsub * 32;   // this wave's 32-q subtile
+      // prefetch this wave's first PREF or2 A-operand chunks (q = q0+col)
+      // BEFORE the staging stores so their load latency hides under them
+      bf16frag qrow[PREF > 0 ? PREF : 1], dorow[PREF > 0 ? PREF : 1];
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
+      for (int c = 0; c < PREF; ++c) {
         qrow[c] = *reinterpret_cast<const bf16frag*>(
             Qb + (q0 + col) * DH + c * 16 + half * 8);
         dorow[c] = *reinterpret_cast<const bf16frag*>(
             dOb + (q0 + col) * DH + c * 16 + half * 8);
       }
-      // stage Q^T / dO^T (512 units / 256 thr = 2 each)
+      // stage Q^T / dO^T [128][64] (1024 units of [1 q][8 d] / 256 thr = 4)
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
+      for (int u = 0; u < 4; ++u) {
         int unit = tid + u * 256;
-        int q = unit & 31;
-        int d0 = (unit >> 5) * 8;
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0 + q) * DH + d0);
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0 + q) * DH + d0);
+        int q = unit & 63;
+        int d0 = (unit >> 6) * 8;
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int row = d0 + j;
-          qtr[(row * 64 + qswz(row, q * 2)) >> 1] = vq.v[j];
-          dotr[(row * 64 + qswz(row, q * 2)) >> 1] = vd.v[j];
+          qtr[(row * 128 + vswz(row, q * 2)) >> 1] = vq.v[j];
+          dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
         }
       }
       __syncthreads();
 
-      const bool live = ((q0 + 31) >= (kv0 + wave * 32)) &&
+      const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
                         (!DOC || q0 < de_wave);
-      const bool diag = (q0 < kv0 + 128);
-
+      const bool diag = (q0 < kv0 + 64);
       if (live) {
         f32x16 s2 = f32x16{}, dp2 = f32x16{};
 #pragma unroll
         for (int c = 0; c < 8; ++c) {
           int colb = (c * 16 + half * 8) * 2;
+          bf16frag qc, dc;
+          if (c < PREF) {
+            qc = qrow[c < PREF ? c : 0];
+            dc = dorow[c < PREF ? c : 0];
+          } else {
+            qc = *reinterpret_cast<const bf16frag*>(
+                Qb + (q0 + col) * DH + c * 16 + half * 8);
+            dc = *reinterpret_cast<const bf16frag*>(
+                dOb + (q0 + col) * DH + c * 16 + half * 8);
+          }
           bf16frag kf = *reinterpret_cast<const bf16frag*>(
               reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
           bf16frag vf = *reinterpret_cast<const bf16frag*>(
               reinterpret_cast<const char*>(vrow) + kvrow_l * 256 + kswz(kvrow_l, colb));
-          s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qrow[c], kf, s2, 0, 0, 0);
-          dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dorow[c], vf, dp2, 0, 0, 0);
+          s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qc, kf, s2, 0, 0, 0);
+          dp2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dc, vf, dp2, 0, 0, 0);
         }
         uint32_t pk2[8], dg2[8];
 #pragma unroll
@@ -1388,10 +1414,10 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           for (int rr = 0; rr < 2; ++rr) {
             int r2 = r + rr;
             int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
-            bool masked = diag && ((int64_t)kv0 + wave * 32 + col > q0 + qrm);
+            bool masked = diag && ((int64_t)kv0 + kvslice * 32 + col > q0 + qrm);
             if (DOC)
               masked = masked ||
-                       ((int64_t)kv0 + wave * 32 + col < doc_start[q0 + qrm]);
+                       ((int64_t)kv0 + kvslice * 32 + col < doc_start[q0 + qrm]);
             float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lseb[q0 + qrm]);
             pv[rr] = pp;
             gv[rr] = pp * (dp2[r2] - delb[q0 + qrm]) * scale;
@@ -1419,16 +1445,17 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
                    half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
           da2[mch] = __builtin_bit_cast(bf16frag, u2);
         }
+        // B-frag q-chunks live in this wave's half of the 64-wide tiles
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
             int trow = dblk * 32 + col;
-            int colb = (mch * 16 + half * 8) * 2;
+            int colb = (qsub * 32 + mch * 16 + half * 8) * 2;
             bf16frag dof = *reinterpret_cast<const bf16frag*>(
-                reinterpret_cast<const char*>(dotr) + trow * 64 + qswz(trow, colb));
+                reinterpret_cast<const char*>(dotr) + trow * 128 + vswz(trow, colb));
             bf16frag qf = *reinterpret_cast<const bf16frag*>(
-                reinterpret_cast<const char*>(qtr) + trow * 64 + qswz(trow, colb));
+                reinterpret_cast<const char*>(qtr) + trow * 128 + vswz(trow, colb));
             dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
             dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
           }
@@ -1438,18 +1465,39 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     }
   }
 
-  // single-writer epilogue: this block is the only contributor to its
-  // [hkv, kv rows] slice (the whole GQA group was folded in-register)
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    int kvr = (r & 3) + 8 * (r >> 2) + 4 * half;
-    int64_t kvg = kv0 + wave * 32 + kvr;
-#pragma unroll
-    for (int d = 0; d < 4; ++d) {
-      dKb[kvg * DH + d * 32 + col] = f2bf(dk_acc[d][r]);
-      dVb[kvg * DH + d * 32 + col] = f2bf(dv_acc[d][r]);
-    }
+  // combine the two q-subtile partials per kv slice through LDS, then store.
+  // red layout: [slice 2][lane 64][64 floats] = 32 KB (reuses K/V space —
+  // every wave has passed the loop-final barrier). Written out directly per
+  // accumulator (no pointer indirection: an f32x16* into the accumulator
+  // arrays sent all 128 accumulator VGPRs to scratch).
+#define VH_DKVG_COMBINE(ACC, DST)                                             \
+  {                                                                           \
+    if (qsub == 1) {                                                          \
+      float* out = red + (kvslice * 64 + lane) * 64;                          \
+      _Pragma("unroll") for (int d = 0; d < 4; ++d)                           \
+          _Pragma("unroll") for (int r = 0; r < 16; r += 4)                   \
+              *reinterpret_cast<float4*>(out + d * 16 + r) = float4{          \
+                  ACC[d][r], ACC[d][r + 1], ACC[d][r + 2], ACC[d][r + 3]};    \
+    }                                                                         \
+    __syncthreads();                                                          \
+    if (qsub == 0) {                                                          \
+      const float* in = red + (kvslice * 64 + lane) * 64;                     \
+      _Pragma("unroll") for (int d = 0; d < 4; ++d)                           \
+          _Pragma("unroll") for (int r = 0; r < 16; ++r) ACC[d][r] +=         \
+          in[d * 16 + r];                                                     \
+      _Pragma("unroll") for (int r = 0; r < 16; ++r) {                        \
+        int kvr = (r & 3) + 8 * (r >> 2) + 4 * half;                          \
+        int64_t kvg = kv0 + kvslice * 32 + kvr;                               \
+        _Pragma("unroll") for (int d = 0; d < 4; ++d) DST[kvg * DH + d * 32 + \
+                                                          col] =             \
+            f2bf(ACC[d][r]);                                                  \
+      }                                                                       \
+    }                                                                         \
+    __syncthreads();                                                          \
   }
+  VH_DKVG_COMBINE(dv_acc, dVb)
+  VH_DKVG_COMBINE(dk_acc, dKb)
+#undef VH_DKVG_COMBINE
 }
 
 template <bool DOC>
@@ -1650,10 +1698,10 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK((doc_start == nullptr) == (doc_end == nullptr),
            "doc_start/doc_end must be passed together");
   VH_CHECK(doc_start == nullptr || B == 1, "varlen requires packed B == 1");
-  dim3 grid_kv((uint32_t)(S / 128), (uint32_t)(B * Hkv));
+  dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true>), grid_kv, dim3(256), 81920, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 4>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1670,7 +1718,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false>), grid_kv, dim3(256), 81920, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 4>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1691,8 +1739,36 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   return 0;
 }
 
+/* probe: the dispatched dkv v6 at prefetch depths 0/4/8 for on-box A/B
+ * (dK/dV [B,Hkv,S,D], no varlen). */
+extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
+                                           const uint16_t* V, const uint16_t* dO,
+                                           const float* delta, const float* lse2,
+                                           uint16_t* dK, uint16_t* dV, int B,
+                                           int Hq, int Hkv, int64_t S,
+                                           float scale, int pref, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(S % 128 == 0, "S %% 128 != 0");
+  dim3 grid((uint32_t)(S / 64), (uint32_t)(B * Hkv));
+#define VH_DKV6(P_)                                                           \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 65536,   \
+                     s, reinterpret_cast<const bf16_t*>(Q),                   \
+                     reinterpret_cast<const bf16_t*>(K),                      \
+                     reinterpret_cast<const bf16_t*>(V),                      \
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,        \
+                     reinterpret_cast<bf16_t*>(dK),                           \
+                     reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
+                     Hkv, S, scale)
+  if (pref == 0) VH_DKV6(0);
+  else if (pref == 8) VH_DKV6(8);
+  else VH_DKV6(4);
+#undef VH_DKV6
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
 /* probe: the round-1 per-Q-head dkv (v2) kept for on-box A/B against the
- * GQA-folded v5 (dK/dV here are per-Q-head [B,Hq,S,D]). */
+ * GQA-folded v6 (dK/dV here are per-Q-head [B,Hq,S,D]). */
 extern "C" int vh_attn_bwd2_dkv2probe_bf16(const uint16_t* Q, const uint16_t* K,
                                            const uint16_t* V, const uint16_t* dO,
                                            const float* delta, const float* lse2,
