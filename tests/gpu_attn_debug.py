@@ -143,6 +143,48 @@ def main():
     dt2p = (time.perf_counter() - t0) / 10
     print(f"dkv v2 probe (per-Q-head): {dt2p*1e3:.3f} ms", flush=True)
 
+    # dkv v6 prefetch-depth A/B (PREF = 0 / 4 / 8)
+    fn6 = lib.vh_attn_bwd2_dkv6probe_bf16
+    fn6.restype = ctypes.c_int
+    fn6.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 3 + [
+        ctypes.c_int64, ctypes.c_float, ctypes.c_int, ctypes.c_void_p]
+    dkv = torch.empty(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    dvv = torch.empty(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    for pref in (0, 4, 8):
+        for _ in range(3):
+            fn6(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+                delta.data_ptr(), lse2.data_ptr(), dkv.data_ptr(), dvv.data_ptr(),
+                B, Hq, Hkv, S, scale, pref, L.cur_stream())
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            fn6(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+                delta.data_ptr(), lse2.data_ptr(), dkv.data_ptr(), dvv.data_ptr(),
+                B, Hq, Hkv, S, scale, pref, L.cur_stream())
+        torch.cuda.synchronize()
+        dt6 = (time.perf_counter() - t0) / 10
+        print(f"dkv v6 pref={pref}: {dt6*1e3:.3f} ms", flush=True)
+
+    # AOTriton reference: time the torch SDPA backward at the same shape
+    qf = q.clone().requires_grad_(True)
+    kf = k.clone().requires_grad_(True)
+    vf = v.clone().requires_grad_(True)
+    import torch.nn.functional as F
+    kk4 = kf.repeat_interleave(Hq // Hkv, dim=1)
+    vv4 = vf.repeat_interleave(Hq // Hkv, dim=1)
+    oo = F.scaled_dot_product_attention(qf, kk4, vv4, is_causal=True, scale=scale)
+    for _ in range(3):
+        oo.backward(do, retain_graph=True)
+        qf.grad = kf.grad = vf.grad = None
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        oo.backward(do, retain_graph=True)
+        qf.grad = kf.grad = vf.grad = None
+    torch.cuda.synchronize()
+    dts = (time.perf_counter() - t0) / 10
+    print(f"torch sdpa bwd (incl. autograd + GQA-repeat grads): {dts*1e3:.3f} ms", flush=True)
+
     # timing at llama shape
     B, Hq, Hkv, S = 1, 32, 8, 4096
     q = (torch.randn(B, Hq, S, 128, device=dev) * 0.5).to(torch.bfloat16)
