@@ -1262,17 +1262,17 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
 //   - 4 waves = (2 kv slices) x (2 q subtiles of a 64-row staged q tile):
 //     each wave owns a distinct 32x32 quadrant; barrier cadence halves per
 //     unit of work vs 32-row q tiles (the dkv4 probe geometry).
-//   - the Q/dO tiles are staged ROW-major (kswz-swizzled) via
-//     global_load_lds — no staging registers, no transpose VALU; the or2
-//     A-operands read back as vectorized b128 LDS rows and the dv/dk
-//     B-frags as 8 scalar column reads each. Register-prefetch variants
-//     (PREF 0/4/8 of the A-operands from global) measured IDENTICAL
-//     (2.08-2.10 ms) — the stall was never the A loads but the scattered
-//     per-lane lse/delta gathers, now LDS-staged per tile (lsed/deld).
+//   - PREF of the or2 A-operands (the lane's Q/dO rows): the first PREF
+//     chunks are prefetched into registers BEFORE the transposed-LDS staging
+//     stores (their L2 latency hides under the staging writes), the rest
+//     load in-loop as dkv v2/v4 did. PREF=8 costs 64 VGPRs and spills
+//     (~250 B/lane) against the 128 accumulator regs at the unified-file
+//     2-waves/SIMD budget of 256; PREF=4 fits. All three measured on-box
+//     via vh_attn_bwd2_dkv6probe_bf16.
 //   - DOC: packed-varlen block-diagonal causal via doc_start/doc_end
 //     (per-token document bounds; see k_attn_fwd).
-// LDS 64.8 KB: K/V strips 2x16 K + Q/dO tiles 2x16 K + scalars -> 2/CU.
-template <bool DOC>
+// LDS 64 KB: K/V strips 2x16 K + Q^T/dO^T tiles 2x16 K -> 2 blocks/CU.
+template <bool DOC, int PREF>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1283,22 +1283,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [64][128] 16 K
   bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 16384);    // [64][128] 16 K
-  // Q/dO tiles: ROW-major with a 16-B row pad (stride 272 B, no XOR
-  // swizzle). The pad de-conflicts the banks (~2-way worst case) while
-  // keeping every read address = one base register + a CONSTANT immediate
-  // offset: with a kswz layout, the 128+ per-lane swizzled addresses are
-  // it-loop-invariant, LICM hoists them all, and the kernel spills
-  // ~390 B/lane against the 128 accumulator registers.
-  bf16_t* qrows = reinterpret_cast<bf16_t*>(smem + 32768);   // [64][272B] 17 K
-  bf16_t* dorows = reinterpret_cast<bf16_t*>(smem + 50176);  // [64][272B] 17 K
-  // per-q-tile softmax scalars, staged once per tile and read per element
-  // as LDS broadcasts: lseb[q0+qrm] straight from global was a SCATTERED
-  // per-lane gather (qrm depends on lane>>5) on the critical path between
-  // the s2 MFMA chain and the pack — the dq kernel, whose lse/delta are
-  // per-lane registers loaded once, runs 2.4x more efficient.
-  float* lsed = reinterpret_cast<float*>(smem + 67584);      // [64]  256 B
-  float* deld = reinterpret_cast<float*>(smem + 67840);      // [64]  256 B
-  int* dsd = reinterpret_cast<int*>(smem + 68096);           // [64]  256 B
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 32768);     // [128][64] 16 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 49152);    // [128][64] 16 K
+  // per-q-tile softmax scalars: lseb[q0+qrm]/delb[q0+qrm] read straight
+  // from global are SCATTERED per-lane gathers (qrm depends on lane>>5)
+  // sitting on the critical path between the s2 MFMA chain and the pack;
+  // stage them once per tile and read as LDS broadcasts instead.
+  float* lsed = reinterpret_cast<float*>(smem + 65536);      // [64]  256 B
+  float* deld = reinterpret_cast<float*>(smem + 65792);      // [64]  256 B
+  int* dsd = reinterpret_cast<int*>(smem + 66048);           // [64]  256 B
   float* red = reinterpret_cast<float*>(smem);               // epilogue reuse
 
   const int kvb = blockIdx.x;          // 64-row kv strip
@@ -1353,46 +1346,55 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     de_wave = doc_end[kv0 + kvslice * 32 + 31];   // wave-live bound
   }
 
-  // single flattened loop over (head g, q tile): one live address chain —
-  // a nested g/qt loop form kept per-head pointer sets alive across the
-  // whole accumulator section and spilled ~190 extra bytes/lane
+  // single flattened loop over (head g, q tile), advanced by INCREMENT
+  // (no it/ntiles division: the division result is non-scalar to the
+  // compiler, turning every per-head base pointer into VGPR pairs that
+  // spill; nested g/qt loops likewise kept per-head pointer sets alive
+  // across the whole accumulator section at ~190 extra bytes/lane)
   const int ntiles = qtn - qt0;
+  int64_t hoff = ((int64_t)b * Hq + (int64_t)hkv * rep) * S;  // head row base
+  int qt = qt0;
   for (int it = 0; it < rep * ntiles; ++it) {
-    const int g = it / ntiles;
-    const int qt = qt0 + (it - g * ntiles);
     {
-      const int hq = hkv * rep + g;
-      const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
-      const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
-      const float* delb = delta + ((int64_t)b * Hq + hq) * S;
-      const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
+      const bf16_t* Qb = Q + hoff * DH;
+      const bf16_t* dOb = dO + hoff * DH;
+      const float* delb = delta + hoff;
+      const float* lseb = lse2 + hoff;
       const int64_t q0t = (int64_t)qt * 64;
       const int64_t q0 = q0t + qsub * 32;   // this wave's 32-q subtile
-      // stage the Q/dO tile rows into the padded layout with plain
-      // vector copies (b128 in, b128 out — no transpose, no scatter)
+      // prefetch this wave's first PREF or2 A-operand chunks (q = q0+col)
+      // BEFORE the staging stores so their load latency hides under them
+      bf16frag qrow[PREF > 0 ? PREF : 1], dorow[PREF > 0 ? PREF : 1];
+#pragma unroll
+      for (int c = 0; c < PREF; ++c) {
+        qrow[c] = *reinterpret_cast<const bf16frag*>(
+            Qb + (q0 + col) * DH + c * 16 + half * 8);
+        dorow[c] = *reinterpret_cast<const bf16frag*>(
+            dOb + (q0 + col) * DH + c * 16 + half * 8);
+      }
+      // stage Q^T / dO^T [128][64] (1024 units of [1 q][8 d] / 256 thr = 4)
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        int o = u * 4096 + tid * 16;
-        int row = o >> 8;
-        int colb = o & 255;
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(
-            Qb + (q0t + row) * DH + (colb >> 1));
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(
-            dOb + (q0t + row) * DH + (colb >> 1));
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(qrows) + row * 272 + colb) = vq;
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(dorows) + row * 272 + colb) = vd;
+        int unit = tid + u * 256;
+        int q = unit & 63;
+        int d0 = (unit >> 6) * 8;
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = d0 + j;
+          qtr[(row * 128 + vswz(row, q * 2)) >> 1] = vq.v[j];
+          dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
+        }
       }
-      if (tid < 64) {
-        lsed[tid] = lseb[q0t + tid];
-        deld[tid] = delb[q0t + tid];
-        if (DOC) dsd[tid] = doc_start[q0t + tid];
+      {
+        // all waves write the same values (benign duplicate) — an
+        // `if (tid < 64)` guard here re-inflated the register pressure
+        int sl = lane;
+        lsed[sl] = lseb[q0t + sl];
+        deld[sl] = delb[q0t + sl];
+        if (DOC) dsd[sl] = doc_start[q0t + sl];
       }
-      // pin iteration boundaries: without this the scheduler software-
-      // pipelines the tile loop and balloons liveness past the 256-VGPR
-      // budget (~390 B/lane scratch; guide trap 18/G9 family)
-      __builtin_amdgcn_sched_barrier(0);
       __syncthreads();
 
       const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
@@ -1400,14 +1402,19 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
       const bool diag = (q0 < kv0 + 64);
       if (live) {
         f32x16 s2 = f32x16{}, dp2 = f32x16{};
-        const int arow = qsub * 32 + col;   // this lane's q row in the tile
 #pragma unroll
         for (int c = 0; c < 8; ++c) {
           int colb = (c * 16 + half * 8) * 2;
-          bf16frag qc = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(qrows) + arow * 272 + colb);
-          bf16frag dc = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(dorows) + arow * 272 + colb);
+          bf16frag qc, dc;
+          if (c < PREF) {
+            qc = qrow[c < PREF ? c : 0];
+            dc = dorow[c < PREF ? c : 0];
+          } else {
+            qc = *reinterpret_cast<const bf16frag*>(
+                Qb + (q0 + col) * DH + c * 16 + half * 8);
+            dc = *reinterpret_cast<const bf16frag*>(
+                dOb + (q0 + col) * DH + c * 16 + half * 8);
+          }
           bf16frag kf = *reinterpret_cast<const bf16frag*>(
               reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
           bf16frag vf = *reinterpret_cast<const bf16frag*>(
@@ -1426,8 +1433,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
             int qls = qsub * 32 + qrm;   // LDS slot within the 64-q tile
             bool masked = diag && ((int64_t)kv0 + kvslice * 32 + col > q0 + qrm);
             if (DOC)
-              masked = masked ||
-                       ((int64_t)kv0 + kvslice * 32 + col < dsd[qls]);
+              masked = masked || ((int64_t)kv0 + kvslice * 32 + col < dsd[qls]);
             float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lsed[qls]);
             pv[rr] = pp;
             gv[rr] = pp * (dp2[r2] - deld[qls]) * scale;
@@ -1455,36 +1461,27 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
                    half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
           da2[mch] = __builtin_bit_cast(bf16frag, u2);
         }
-        // B-frags (k = q, n = d) assembled from the row-major tiles as 8
-        // scalar column reads each (row-broadcast pattern, ~2-way bank
-        // conflicts) — the price of dropping the transposed staging pass
+        // B-frag q-chunks live in this wave's half of the 64-wide tiles
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
-          const int qb2 = qsub * 32 + mch * 16 + half * 8;
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
-            const int dcol = (dblk * 32 + col) * 2;
-            const char* dbase = reinterpret_cast<const char*>(dorows) +
-                                qb2 * 272 + dcol;
-            const char* qbase = reinterpret_cast<const char*>(qrows) +
-                                qb2 * 272 + dcol;
-            bf16frag dof, qf;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              dof[j] = *reinterpret_cast<const bf16_t*>(dbase + j * 272);
-              qf[j] = *reinterpret_cast<const bf16_t*>(qbase + j * 272);
-            }
+            int trow = dblk * 32 + col;
+            int colb = (qsub * 32 + mch * 16 + half * 8) * 2;
+            bf16frag dof = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(dotr) + trow * 128 + vswz(trow, colb));
+            bf16frag qf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(qtr) + trow * 128 + vswz(trow, colb));
             dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
             dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
-            // keep each (assemble 16 scalar reads -> 2 MFMAs) cluster
-            // intact: unfenced, the scheduler hoists all 128 scalar reads
-            // + packs above the first MFMA (~390 B/lane scratch)
-            __builtin_amdgcn_sched_barrier(0);
           }
         }
       }
-      __builtin_amdgcn_sched_barrier(0);
       __syncthreads();
+    }
+    if (++qt == qtn) {
+      qt = qt0;
+      hoff += S;
     }
   }
 
@@ -1724,7 +1721,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true>), grid_kv, dim3(256), 68608, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 66560, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1741,7 +1738,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false>), grid_kv, dim3(256), 68608, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 66560, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1762,26 +1759,30 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   return 0;
 }
 
-/* probe: the dispatched dkv v6 alone (dK/dV [B,Hkv,S,D], no varlen, no dq)
- * for isolating dkv time; `variant` is reserved (ignored). */
+/* probe: the dispatched dkv v6 at prefetch depths 0/4/8 for on-box A/B
+ * (dK/dV [B,Hkv,S,D], no varlen). */
 extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                                            const uint16_t* V, const uint16_t* dO,
                                            const float* delta, const float* lse2,
                                            uint16_t* dK, uint16_t* dV, int B,
                                            int Hq, int Hkv, int64_t S,
-                                           float scale, int variant, void* stream) {
-  (void)variant;
+                                           float scale, int pref, void* stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 64), (uint32_t)(B * Hkv));
-  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false>), grid, dim3(256), 68608, s,
-                     reinterpret_cast<const bf16_t*>(Q),
-                     reinterpret_cast<const bf16_t*>(K),
-                     reinterpret_cast<const bf16_t*>(V),
-                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,
-                     reinterpret_cast<bf16_t*>(dK),
-                     reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
-                     Hkv, S, scale);
+#define VH_DKV6(P_)                                                           \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 66560,   \
+                     s, reinterpret_cast<const bf16_t*>(Q),                   \
+                     reinterpret_cast<const bf16_t*>(K),                      \
+                     reinterpret_cast<const bf16_t*>(V),                      \
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,        \
+                     reinterpret_cast<bf16_t*>(dK),                           \
+                     reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
+                     Hkv, S, scale)
+  if (pref == 0) VH_DKV6(0);
+  else if (pref == 8) VH_DKV6(8);
+  else VH_DKV6(4);
+#undef VH_DKV6
   VH_HIP(hipGetLastError());
   return 0;
 }
