@@ -963,6 +963,9 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv(
     if (live) {
       // or2: C = [q regs][kv lanes]; A-operands are the lane's Q/dO rows,
       // loaded per chunk (k = q0+col) so only 8 regs are live per iteration.
+      // lse/delta: per-lane own-row loads + cross-lane shuffles (see dkv_g).
+      const float lse_own = lseb[q0 + col];
+      const float del_own = delb[q0 + col];
       f32x16 s2 = f32x16{}, dp2 = f32x16{};
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
@@ -987,9 +990,11 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv(
           int r2 = r + rr;
           int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
           bool masked = diag && ((int64_t)kv0 + wave * 32 + col > q0 + qrm);
-          float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lseb[q0 + qrm]);
+          float lse_q = __shfl(lse_own, qrm, 32);
+          float del_q = __shfl(del_own, qrm, 32);
+          float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lse_q);
           pv[rr] = pp;
-          gv[rr] = pp * (dp2[r2] - delb[q0 + qrm]) * scale;
+          gv[rr] = pp * (dp2[r2] - del_q) * scale;
         }
         pk2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[0]) |
                       ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[1]) << 16);
@@ -1285,13 +1290,6 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 16384);    // [64][128] 16 K
   bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 32768);     // [128][64] 16 K
   bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 49152);    // [128][64] 16 K
-  // per-q-tile softmax scalars: lseb[q0+qrm]/delb[q0+qrm] read straight
-  // from global are SCATTERED per-lane gathers (qrm depends on lane>>5)
-  // sitting on the critical path between the s2 MFMA chain and the pack;
-  // stage them once per tile and read as LDS broadcasts instead.
-  float* lsed = reinterpret_cast<float*>(smem + 65536);      // [64]  256 B
-  float* deld = reinterpret_cast<float*>(smem + 65792);      // [64]  256 B
-  int* dsd = reinterpret_cast<int*>(smem + 66048);           // [64]  256 B
   float* red = reinterpret_cast<float*>(smem);               // epilogue reuse
 
   const int kvb = blockIdx.x;          // 64-row kv strip
@@ -1346,20 +1344,19 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     de_wave = doc_end[kv0 + kvslice * 32 + 31];   // wave-live bound
   }
 
-  // single flattened loop over (head g, q tile), advanced by INCREMENT
-  // (no it/ntiles division: the division result is non-scalar to the
-  // compiler, turning every per-head base pointer into VGPR pairs that
-  // spill; nested g/qt loops likewise kept per-head pointer sets alive
-  // across the whole accumulator section at ~190 extra bytes/lane)
+  // single flattened loop over (head g, q tile): one live address chain —
+  // a nested g/qt loop form kept per-head pointer sets alive across the
+  // whole accumulator section and spilled ~190 extra bytes/lane
   const int ntiles = qtn - qt0;
-  int64_t hoff = ((int64_t)b * Hq + (int64_t)hkv * rep) * S;  // head row base
-  int qt = qt0;
   for (int it = 0; it < rep * ntiles; ++it) {
+    const int g = it / ntiles;
+    const int qt = qt0 + (it - g * ntiles);
     {
-      const bf16_t* Qb = Q + hoff * DH;
-      const bf16_t* dOb = dO + hoff * DH;
-      const float* delb = delta + hoff;
-      const float* lseb = lse2 + hoff;
+      const int hq = hkv * rep + g;
+      const bf16_t* Qb = Q + (((int64_t)b * Hq + hq) * S) * DH;
+      const bf16_t* dOb = dO + (((int64_t)b * Hq + hq) * S) * DH;
+      const float* delb = delta + ((int64_t)b * Hq + hq) * S;
+      const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
       const int64_t q0t = (int64_t)qt * 64;
       const int64_t q0 = q0t + qsub * 32;   // this wave's 32-q subtile
       // prefetch this wave's first PREF or2 A-operand chunks (q = q0+col)
@@ -1387,20 +1384,21 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
         }
       }
-      {
-        // all waves write the same values (benign duplicate) — an
-        // `if (tid < 64)` guard here re-inflated the register pressure
-        int sl = lane;
-        lsed[sl] = lseb[q0t + sl];
-        deld[sl] = delb[q0t + sl];
-        if (DOC) dsd[sl] = doc_start[q0t + sl];
-      }
       __syncthreads();
 
       const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
                         (!DOC || q0 < de_wave);
       const bool diag = (q0 < kv0 + 64);
       if (live) {
+        // per-lane softmax scalars for the lane's own q row; the per-element
+        // values come from cross-lane shuffles below. Reading
+        // lseb[q0+qrm]/delb[q0+qrm] straight from global per element was a
+        // SCATTERED gather (qrm depends on lane>>5) on the critical path
+        // between the s2 MFMA chain and the pack; these two coalesced loads
+        // issue before the MFMA chain and hide under it.
+        const float lse_own = lseb[q0 + col];
+        const float del_own = delb[q0 + col];
+        const int ds_own = DOC ? doc_start[q0 + col] : 0;
         f32x16 s2 = f32x16{}, dp2 = f32x16{};
 #pragma unroll
         for (int c = 0; c < 8; ++c) {
@@ -1430,13 +1428,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           for (int rr = 0; rr < 2; ++rr) {
             int r2 = r + rr;
             int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
-            int qls = qsub * 32 + qrm;   // LDS slot within the 64-q tile
+            float lse_q = __shfl(lse_own, qrm, 32);
+            float del_q = __shfl(del_own, qrm, 32);
             bool masked = diag && ((int64_t)kv0 + kvslice * 32 + col > q0 + qrm);
             if (DOC)
-              masked = masked || ((int64_t)kv0 + kvslice * 32 + col < dsd[qls]);
-            float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lsed[qls]);
+              masked = masked ||
+                       ((int64_t)kv0 + kvslice * 32 + col < __shfl(ds_own, qrm, 32));
+            float pp = masked ? 0.f : __builtin_exp2f(s2[r2] * scale2 - lse_q);
             pv[rr] = pp;
-            gv[rr] = pp * (dp2[r2] - deld[qls]) * scale;
+            gv[rr] = pp * (dp2[r2] - del_q) * scale;
           }
           pk2[r >> 1] = (uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[0]) |
                         ((uint32_t)__builtin_bit_cast(uint16_t, (__bf16)pv[1]) << 16);
@@ -1478,10 +1478,6 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         }
       }
       __syncthreads();
-    }
-    if (++qt == qtn) {
-      qt = qt0;
-      hoff += S;
     }
   }
 
@@ -1721,7 +1717,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 66560, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1738,7 +1734,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 66560, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1771,7 +1767,7 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 64), (uint32_t)(B * Hkv));
 #define VH_DKV6(P_)                                                           \
-  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 66560,   \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 65536,   \
                      s, reinterpret_cast<const bf16_t*>(Q),                   \
                      reinterpret_cast<const bf16_t*>(K),                      \
                      reinterpret_cast<const bf16_t*>(V),                      \
