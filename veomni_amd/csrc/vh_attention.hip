@@ -1723,7 +1723,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 65536, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 67584, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1740,7 +1740,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 65536, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 67584, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1773,7 +1773,7 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 64), (uint32_t)(B * Hkv));
 #define VH_DKV6(P_)                                                           \
-  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 65536,   \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 67584,   \
                      s, reinterpret_cast<const bf16_t*>(Q),                   \
                      reinterpret_cast<const bf16_t*>(K),                      \
                      reinterpret_cast<const bf16_t*>(V),                      \
