@@ -1381,10 +1381,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         *reinterpret_cast<bf16x8*>(
             reinterpret_cast<char*>(dorows) + row * 272 + colb) = vd;
       }
+      // IR-level memory clobber: TBAA treats the bf16x8 stores and the
+      // bf16frag/__bf16 tile reads as no-alias, and the loop pipeliner then
+      // hoists the NEXT tile's staging across the barrier ahead of this
+      // tile's compute (observed in ISA as stage(0);bar;stage(1);bar;
+      // compute(0) — a one-tile data shift). The clobber pins stores and
+      // loads to program order at the IR level; sched_barrier alone acts
+      // too late (backend) to stop it.
+      asm volatile("" ::: "memory");
       __syncthreads();
-      // pin: the padded variant's first bring-up showed the B-frag LDS
-      // reads scheduled ABOVE this barrier (cross-thread race, garbage
-      // dK/dV); forbid any motion across it
       __builtin_amdgcn_sched_barrier(0);
 
       const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
@@ -1482,6 +1487,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
       }
       // pin the tail too: reads of THIS tile must not sink below the next
       // iteration's staging barrier
+      asm volatile("" ::: "memory");
       __builtin_amdgcn_sched_barrier(0);
       __syncthreads();
     }
