@@ -1787,9 +1787,11 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  if (pref == 0) VH_DKV6(0);
-  else if (pref == 8) VH_DKV6(8);
-  else VH_DKV6(4);
+  // variant bisect: bit0 = A-frags from global, bit1 = B-frags from global
+  if (pref == 1) VH_DKV6(1);
+  else if (pref == 2) VH_DKV6(2);
+  else if (pref == 3) VH_DKV6(3);
+  else VH_DKV6(0);
 #undef VH_DKV6
   VH_HIP(hipGetLastError());
   return 0;
