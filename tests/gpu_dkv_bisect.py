@@ -56,11 +56,21 @@ def main():
                  B, Hq, Hkv, S, scale, var, L.cur_stream())
         assert rc == 0
         torch.cuda.synchronize()
-        ek = (dk.float() - kf.grad).abs().max().item()
-        ev = (dv.float() - vf.grad).abs().max().item()
+        ekr = (dk.float() - kf.grad).abs().amax(dim=(0, 1, 3))  # per kv row
+        evr = (dv.float() - vf.grad).abs().amax(dim=(0, 1, 3))
+        ek = ekr.max().item()
+        ev = evr.max().item()
         print(f"variant {var} (A={'glb' if var & 1 else 'lds'} "
               f"B={'glb' if var & 2 else 'lds'}): |dK err|={ek:.4g} |dV err|={ev:.4g}",
               flush=True)
+        if var == 0:
+            # per-32-row error profile: which kv slices are wrong?
+            prof = [round(evr[i:i + 32].max().item(), 3) for i in range(0, S, 32)]
+            print("  dV err per 32-kv-row block:", prof, flush=True)
+            print("  dv[0,0,0,:4]   ", dv.float()[0, 0, 0, :4].tolist(), flush=True)
+            print("  dvref[0,0,0,:4]", vf.grad[0, 0, 0, :4].tolist(), flush=True)
+            print("  dv[0,0,40,:4]  ", dv.float()[0, 0, 40, :4].tolist(), flush=True)
+            print("  dvref[0,0,40,:4]", vf.grad[0, 0, 40, :4].tolist(), flush=True)
 
 
 if __name__ == "__main__":
