@@ -1429,8 +1429,14 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           for (int rr = 0; rr < 2; ++rr) {
             int r2 = r + rr;
             int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
-            float lse_q = __shfl(lse_own, qrm, 32);
-            float del_q = __shfl(del_own, qrm, 32);
+            float lse_q, del_q;
+            if (PREF & 4) {  // probe: per-element global gathers (pre-shfl)
+              lse_q = lseb[q0 + qrm];
+              del_q = delb[q0 + qrm];
+            } else {
+              lse_q = __shfl(lse_own, qrm, 32);
+              del_q = __shfl(del_own, qrm, 32);
+            }
             bool masked = diag && ((int64_t)kv0 + kvslice * 32 + col > q0 + qrm);
             if (DOC)
               masked = masked ||
@@ -1791,6 +1797,7 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   if (pref == 1) VH_DKV6(1);
   else if (pref == 2) VH_DKV6(2);
   else if (pref == 3) VH_DKV6(3);
+  else if (pref == 7) VH_DKV6(7);
   else VH_DKV6(0);
 #undef VH_DKV6
   VH_HIP(hipGetLastError());
