@@ -1368,7 +1368,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
       // stage the Q/dO tile rows into the padded layout with plain
       // vector copies (b128 in, b128 out — no transpose, no scatter)
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
+      for (int u = 0; u < 4 && !(PREF & 8); ++u) {
         int o = u * 4096 + tid * 16;
         int row = o >> 8;
         int colb = o & 255;
@@ -1798,6 +1798,7 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   else if (pref == 2) VH_DKV6(2);
   else if (pref == 3) VH_DKV6(3);
   else if (pref == 7) VH_DKV6(7);
+  else if (pref == 15) VH_DKV6(15);
   else VH_DKV6(0);
 #undef VH_DKV6
   VH_HIP(hipGetLastError());
