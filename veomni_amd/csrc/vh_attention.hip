@@ -1288,14 +1288,8 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [64][128] 16 K
   bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 16384);    // [64][128] 16 K
-  // Q/dO tiles ROW-major with a 16-B row pad (stride 272 B): staging is 8
-  // plain b128 copies per thread (the transposed-vswz layout needed 32
-  // scalar ds_writes per thread per tile — PMC showed 2.9x the dq kernel's
-  // LDS instruction count and 66% WAIT_ANY), A-side chunks read back as
-  // b128 rows, B-frags as 8 scalar column reads at constant immediate
-  // offsets (base + j*272).
-  bf16_t* qrows = reinterpret_cast<bf16_t*>(smem + 32768);   // [64][272B] 17 K
-  bf16_t* dorows = reinterpret_cast<bf16_t*>(smem + 50176);  // [64][272B] 17 K
+  bf16_t* qtr = reinterpret_cast<bf16_t*>(smem + 32768);     // [128][64] 16 K
+  bf16_t* dotr = reinterpret_cast<bf16_t*>(smem + 49152);    // [128][64] 16 K
   float* red = reinterpret_cast<float*>(smem);               // epilogue reuse
 
   const int kvb = blockIdx.x;          // 64-row kv strip
@@ -1365,32 +1359,32 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
       const float* lseb = lse2 + ((int64_t)b * Hq + hq) * S;
       const int64_t q0t = (int64_t)qt * 64;
       const int64_t q0 = q0t + qsub * 32;   // this wave's 32-q subtile
-      // stage the Q/dO tile rows into the padded layout with plain
-      // vector copies (b128 in, b128 out — no transpose, no scatter)
+      // prefetch this wave's first PREF or2 A-operand chunks (q = q0+col)
+      // BEFORE the staging stores so their load latency hides under them
+      bf16frag qrow[PREF > 0 ? PREF : 1], dorow[PREF > 0 ? PREF : 1];
 #pragma unroll
-      for (int u = 0; u < 4 && !(PREF & 8); ++u) {
-        int o = u * 4096 + tid * 16;
-        int row = o >> 8;
-        int colb = o & 255;
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(
-            Qb + (q0t + row) * DH + (colb >> 1));
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(
-            dOb + (q0t + row) * DH + (colb >> 1));
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(qrows) + row * 272 + colb) = vq;
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(dorows) + row * 272 + colb) = vd;
+      for (int c = 0; c < PREF; ++c) {
+        qrow[c] = *reinterpret_cast<const bf16frag*>(
+            Qb + (q0 + col) * DH + c * 16 + half * 8);
+        dorow[c] = *reinterpret_cast<const bf16frag*>(
+            dOb + (q0 + col) * DH + c * 16 + half * 8);
       }
-      // IR-level memory clobber: TBAA treats the bf16x8 stores and the
-      // bf16frag/__bf16 tile reads as no-alias, and the loop pipeliner then
-      // hoists the NEXT tile's staging across the barrier ahead of this
-      // tile's compute (observed in ISA as stage(0);bar;stage(1);bar;
-      // compute(0) — a one-tile data shift). The clobber pins stores and
-      // loads to program order at the IR level; sched_barrier alone acts
-      // too late (backend) to stop it.
-      asm volatile("" ::: "memory");
+      // stage Q^T / dO^T [128][64] (1024 units of [1 q][8 d] / 256 thr = 4)
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        int unit = tid + u * 256;
+        int q = unit & 63;
+        int d0 = (unit >> 6) * 8;
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = d0 + j;
+          qtr[(row * 128 + vswz(row, q * 2)) >> 1] = vq.v[j];
+          dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
+        }
+      }
       __syncthreads();
-      __builtin_amdgcn_sched_barrier(0);
 
       const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
                         (!DOC || q0 < de_wave);
@@ -1405,15 +1399,20 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         const float lse_own = lseb[q0 + col];
         const float del_own = delb[q0 + col];
         const int ds_own = DOC ? doc_start[q0 + col] : 0;
-        const int arow = qsub * 32 + col;   // this lane's q row in the tile
         f32x16 s2 = f32x16{}, dp2 = f32x16{};
 #pragma unroll
         for (int c = 0; c < 8; ++c) {
           int colb = (c * 16 + half * 8) * 2;
-          bf16frag qc = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(qrows) + arow * 272 + colb);
-          bf16frag dc = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(dorows) + arow * 272 + colb);
+          bf16frag qc, dc;
+          if (c < PREF) {
+            qc = qrow[c < PREF ? c : 0];
+            dc = dorow[c < PREF ? c : 0];
+          } else {
+            qc = *reinterpret_cast<const bf16frag*>(
+                Qb + (q0 + col) * DH + c * 16 + half * 8);
+            dc = *reinterpret_cast<const bf16frag*>(
+                dOb + (q0 + col) * DH + c * 16 + half * 8);
+          }
           bf16frag kf = *reinterpret_cast<const bf16frag*>(
               reinterpret_cast<const char*>(krow) + kvrow_l * 256 + kswz(kvrow_l, colb));
           bf16frag vf = *reinterpret_cast<const bf16frag*>(
@@ -1429,14 +1428,8 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           for (int rr = 0; rr < 2; ++rr) {
             int r2 = r + rr;
             int qrm = (r2 & 3) + 8 * (r2 >> 2) + 4 * half;
-            float lse_q, del_q;
-            if (PREF & 4) {  // probe: per-element global gathers (pre-shfl)
-              lse_q = lseb[q0 + qrm];
-              del_q = delb[q0 + qrm];
-            } else {
-              lse_q = __shfl(lse_own, qrm, 32);
-              del_q = __shfl(del_own, qrm, 32);
-            }
+            float lse_q = __shfl(lse_own, qrm, 32);
+            float del_q = __shfl(del_own, qrm, 32);
             bool masked = diag && ((int64_t)kv0 + kvslice * 32 + col > q0 + qrm);
             if (DOC)
               masked = masked ||
@@ -1468,33 +1461,22 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
                    half ? dg2[4 * mch + 2] : e0, half ? dg2[4 * mch + 3] : e1};
           da2[mch] = __builtin_bit_cast(bf16frag, u2);
         }
-        // B-frags (k = q, n = d): 8 scalar column reads each from the
-        // padded tiles, address = one base + constant immediates (j*272)
+        // B-frag q-chunks live in this wave's half of the 64-wide tiles
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
-          const int qb2 = qsub * 32 + mch * 16 + half * 8;
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
-            const int dcol = (dblk * 32 + col) * 2;
-            const char* dbase = reinterpret_cast<const char*>(dorows) +
-                                qb2 * 272 + dcol;
-            const char* qbase = reinterpret_cast<const char*>(qrows) +
-                                qb2 * 272 + dcol;
-            bf16frag dof, qf;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              dof[j] = *reinterpret_cast<const bf16_t*>(dbase + j * 272);
-              qf[j] = *reinterpret_cast<const bf16_t*>(qbase + j * 272);
-            }
+            int trow = dblk * 32 + col;
+            int colb = (qsub * 32 + mch * 16 + half * 8) * 2;
+            bf16frag dof = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(dotr) + trow * 128 + vswz(trow, colb));
+            bf16frag qf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(qtr) + trow * 128 + vswz(trow, colb));
             dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
             dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
           }
         }
       }
-      // pin the tail too: reads of THIS tile must not sink below the next
-      // iteration's staging barrier
-      asm volatile("" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
       __syncthreads();
     }
   }
@@ -1735,7 +1717,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 67584, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1752,7 +1734,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 67584, s,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0>), grid_kv, dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1785,7 +1767,7 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 64), (uint32_t)(B * Hkv));
 #define VH_DKV6(P_)                                                           \
-  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 67584,   \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, P_>), grid, dim3(256), 65536,   \
                      s, reinterpret_cast<const bf16_t*>(Q),                   \
                      reinterpret_cast<const bf16_t*>(K),                      \
                      reinterpret_cast<const bf16_t*>(V),                      \
@@ -1793,13 +1775,9 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  // variant bisect: bit0 = A-frags from global, bit1 = B-frags from global
-  if (pref == 1) VH_DKV6(1);
-  else if (pref == 2) VH_DKV6(2);
-  else if (pref == 3) VH_DKV6(3);
-  else if (pref == 7) VH_DKV6(7);
-  else if (pref == 15) VH_DKV6(15);
-  else VH_DKV6(0);
+  if (pref == 0) VH_DKV6(0);
+  else if (pref == 8) VH_DKV6(8);
+  else VH_DKV6(4);
 #undef VH_DKV6
   VH_HIP(hipGetLastError());
   return 0;
