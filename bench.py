@@ -278,6 +278,9 @@ def main():
                 "global_batch": mbs * n_gpus,
                 "seq_len": seq,
                 "parallelism": f"dp{world}" + (f"_ep{ep_size}" if ep_size > 1 else ""),
+                # the aux load-balancing loss IS part of every timed step
+                # (reference Qwen3-MoE default; ref load_balancing_loss/eager.py)
+                "router_aux_loss_coef": cfg.router_aux_loss_coef,
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
