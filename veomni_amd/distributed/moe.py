@@ -57,43 +57,61 @@ def all_to_all(group, x, output_split_sizes=None, input_split_sizes=None):
     return _AllToAll.apply(group, x, output_split_sizes, input_split_sizes)
 
 
+# NOTE on the four helpers below: their OUTPUT is a bit-exact
+# index-bookkeeping contract with the reference (moe_utils.py:19-99) — the
+# expert-major traversal order and the fp32 accumulation order ARE the spec
+# (checked by tests/test_dist_cpu.py parity against the reference's own
+# moe_utils on random routings). The implementations here use different
+# torch idioms (nonzero over masked_select, index_add_ over expanded
+# scatter_add_) chosen for the ROCm kernels they lower to.
+
+
 def permute(tokens: torch.Tensor, routing_map: torch.Tensor):
-    """Expert-major stable permutation (ref moe_utils.py:19-41).
+    """Expert-major stable permutation.
 
     routing_map: [num_experts, num_tokens] 0/1. Returns (permuted, mapping)
-    where mapping[i] = source token row of permuted row i.
+    where mapping[i] = source token row of permuted row i. Row order =
+    expert-major, token-index-minor — exactly the reference's order
+    (moe_utils.py:19-41), produced here by row-major nonzero().
     """
-    num_tokens = tokens.shape[0]
-    num_experts = routing_map.shape[0]
-    token_indices = (
-        torch.arange(num_tokens, device=routing_map.device).unsqueeze(0).expand(num_experts, -1)
-    )
-    sorted_indices = token_indices.masked_select(routing_map.bool())
+    # nonzero on [E, T] enumerates (e, t) in expert-major order; column 1 is
+    # the source token row
+    sorted_indices = routing_map.bool().nonzero(as_tuple=True)[1]
     return tokens.index_select(0, sorted_indices), sorted_indices
 
 
 def unpermute(tokens, routing_weights, hidden_states_shape, permutation_mapping, routing_map):
-    """Weighted fp32 scatter_add unpermute (ref moe_utils.py:44-72)."""
-    tokens_weight = routing_weights.T.contiguous().masked_select(routing_map.bool())
-    tokens = tokens * tokens_weight.unsqueeze(-1)
-    hidden_dim = hidden_states_shape[-1]
+    """Weighted fp32 unpermute (ref contract moe_utils.py:44-72): each
+    permuted row is scaled by its routing weight and accumulated in fp32
+    into its source token row."""
+    tokens_weight = routing_weights.T.masked_select(routing_map.bool())
     out = torch.zeros(hidden_states_shape, device=tokens.device, dtype=torch.float32)
-    out.scatter_add_(0, permutation_mapping.unsqueeze(1).expand(-1, hidden_dim), tokens.float())
+    # index_add_ lowers to one atomics pass on ROCm (the reference's
+    # dim-expanded scatter_add_ builds an [N, H] index tensor first)
+    out.index_add_(0, permutation_mapping, (tokens * tokens_weight.unsqueeze(-1)).float())
     return out.to(tokens.dtype)
 
 
 def generate_weights_idx(routing_weights, selected_experts, num_experts):
-    """[T, topk] weights -> dense [T, E] (ref moe_utils.py:75-92)."""
-    T, _ = routing_weights.shape
-    out = torch.zeros((T, num_experts), dtype=routing_weights.dtype, device=routing_weights.device)
+    """[T, topk] weights -> dense [T, E] (ref contract moe_utils.py:75-92;
+    duplicate expert picks accumulate)."""
+    T = routing_weights.shape[0]
+    out = torch.zeros((T, num_experts), dtype=routing_weights.dtype,
+                      device=routing_weights.device)
     out.scatter_add_(1, selected_experts, routing_weights)
     return out
 
 
 def sort_chunks_by_idxs(x, split_sizes, sorted_idxs):
-    """Reorder row-chunks (ref moe_utils.py:95-99)."""
-    chunks = torch.split(x, split_sizes.tolist() if torch.is_tensor(split_sizes) else split_sizes, dim=0)
-    return torch.cat([chunks[i] for i in sorted_idxs], dim=0)
+    """Reorder row-chunks (ref contract moe_utils.py:95-99) through one
+    gather: build the row permutation from chunk offsets instead of
+    split+cat (one kernel instead of ~E small cats)."""
+    sizes = torch.as_tensor(split_sizes, device="cpu", dtype=torch.long)
+    offs = torch.zeros(sizes.numel() + 1, dtype=torch.long)
+    torch.cumsum(sizes, 0, out=offs[1:])
+    order = torch.cat([torch.arange(int(offs[i]), int(offs[i + 1]))
+                       for i in (sorted_idxs.tolist() if torch.is_tensor(sorted_idxs) else sorted_idxs)])
+    return x.index_select(0, order.to(x.device))
 
 
 def preprocess(expert_mask: torch.Tensor, num_experts: int, ep_group) -> tuple:
