@@ -174,6 +174,101 @@ def test_ep_dispatch_combine():
     spawn(_ep_dispatch)
 
 
+def _torch_mlp_fwd(tokens, cumsum, gate_up, down):
+    out = torch.empty(tokens.shape[0], down.shape[1], dtype=tokens.dtype)
+    fc1s = torch.empty(tokens.shape[0], gate_up.shape[1], dtype=tokens.dtype)
+    start = 0
+    for g in range(gate_up.shape[0]):
+        end = int(cumsum[g])
+        x = tokens[start:end]
+        gu = x @ gate_up[g].t()
+        fc1s[start:end] = gu
+        gate, up = gu.chunk(2, -1)
+        out[start:end] = (torch.nn.functional.silu(gate) * up) @ down[g].t()
+        start = end
+    return out, (tokens, gate_up, down, fc1s)
+
+
+def _torch_mlp_bwd_dgrad(dY, cumsum, saved):
+    tokens, gate_up, down, fc1s = saved
+    d_tokens = torch.empty_like(tokens)
+    d_fc1 = torch.empty_like(fc1s)
+    start = 0
+    for g in range(gate_up.shape[0]):
+        end = int(cumsum[g])
+        gu = fc1s[start:end].detach().requires_grad_(True)
+        with torch.enable_grad():
+            gate, up = gu.chunk(2, -1)
+            act = torch.nn.functional.silu(gate) * up
+            act.backward(dY[start:end] @ down[g])
+        d_fc1[start:end] = gu.grad
+        d_tokens[start:end] = gu.grad @ gate_up[g]
+        start = end
+    return d_tokens, (dY, d_fc1)
+
+
+def _torch_mlp_bwd_wgrad(cumsum, saved, stash):
+    tokens, gate_up, down, fc1s = saved
+    dY, d_fc1 = stash
+    d_gu = torch.zeros_like(gate_up)
+    d_down = torch.zeros_like(down)
+    start = 0
+    for g in range(gate_up.shape[0]):
+        end = int(cumsum[g])
+        gate, up = fc1s[start:end].chunk(2, -1)
+        act = torch.nn.functional.silu(gate) * up
+        d_down[g] = dY[start:end].t() @ act
+        d_gu[g] = d_fc1[start:end].t() @ tokens[start:end]
+        start = end
+    return (d_gu, d_down)
+
+
+def _ep_dispatch_a2a_overlap(rank, ws):
+    """The overlapped EP class (dispatch a2a owned by the autograd node,
+    return a2a launched before the wgrads) must match the plain
+    dispatch_to_ep_class path bit-for-bit on outputs AND all grads."""
+    from veomni_amd.distributed.moe import (dispatch_to_ep_a2a_class,
+                                            dispatch_to_ep_class,
+                                            make_ep_a2a_class)
+    from veomni_amd.distributed.parallel_state import init_parallel_state
+
+    init_parallel_state(ep_size=ws)
+    E, topk, T, H, I = 8, 2, 16, 32, 24
+    torch.manual_seed(5)
+    gate_up = torch.randn(E, 2 * I, H) * 0.1
+    down = torch.randn(E, H, I) * 0.1
+    torch.manual_seed(100 + rank)
+    hidden = torch.randn(T, H)
+    sel = torch.randint(0, E, (T, topk))
+    sel[:, 1] = (sel[:, 0] + 1) % E
+    rw = torch.softmax(torch.randn(T, topk), -1)
+
+    local_e = E // ws
+    my_gu = gate_up[rank * local_e:(rank + 1) * local_e].clone().requires_grad_(True)
+    my_down = down[rank * local_e:(rank + 1) * local_e].clone().requires_grad_(True)
+    h_ref = hidden.clone().requires_grad_(True)
+    out_ref = dispatch_to_ep_class(_TorchEpClass, E, rw, sel, h_ref,
+                                   my_gu, my_down)
+    out_ref.sum().backward()
+
+    cls = make_ep_a2a_class(_torch_mlp_fwd, _torch_mlp_bwd_dgrad,
+                            _torch_mlp_bwd_wgrad)
+    my_gu2 = gate_up[rank * local_e:(rank + 1) * local_e].clone().requires_grad_(True)
+    my_down2 = down[rank * local_e:(rank + 1) * local_e].clone().requires_grad_(True)
+    h2 = hidden.clone().requires_grad_(True)
+    out2 = dispatch_to_ep_a2a_class(cls, E, rw, sel, h2, my_gu2, my_down2)
+    out2.sum().backward()
+
+    torch.testing.assert_close(out2, out_ref, rtol=0, atol=0)
+    torch.testing.assert_close(h2.grad, h_ref.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(my_gu2.grad, my_gu.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(my_down2.grad, my_down.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_ep_dispatch_a2a_overlap_equivalence():
+    spawn(_ep_dispatch_a2a_overlap)
+
+
 # ------------------------------------------------- FSDP2 grad-norm equivalence
 def _fsdp_equivalence(rank, ws, preset="tiny-dense"):
     from veomni_amd.distributed.fsdp2 import build_parallelize_model

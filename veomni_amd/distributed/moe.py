@@ -174,6 +174,121 @@ def tokens_post_all2all(expert_outputs, routing_weights, selected_experts, num_e
                      local_input_permutation_mapping, routing_map)
 
 
+def make_ep_a2a_class(mlp_fwd: Callable, mlp_bwd_dgrad: Callable,
+                      mlp_bwd_wgrad: Callable):
+    """Build an EP expert autograd class that owns the DISPATCH a2a + chunk
+    resort (instead of leaving them as separate autograd nodes, the
+    reference's structure — moe_layer.py:120-186 + comm.py:20-54).
+
+    Why: in backward, the return all-to-all of d_tokens does not depend on
+    the wgrad GEMMs, but as a separate autograd node it can only start
+    after the expert backward RETURNS — i.e. after the wgrads. Owning the
+    a2a lets backward launch it (async, on the c10d comm stream) right
+    after the dgrad chain and run the wgrad GEMMs underneath it — on 8
+    GPUs the per-layer exchange (~134 MB/dir/rank) rides xGMI while ~2/3
+    of the backward FLOPs execute.
+
+    mlp_fwd(tokens, cumsum, *weights) -> (out, saved_tuple)
+    mlp_bwd_dgrad(dY, cumsum, saved) -> (d_tokens, stash)
+    mlp_bwd_wgrad(cumsum, saved, stash) -> tuple of weight grads
+    """
+
+    class _EPA2AClass(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx, local_permuted, cumsum, group, input_splits,
+                    output_splits, sort_sizes, sort_order, *weights):
+            ws = dist.get_world_size(group=group) if group is not None else 1
+            if ws == 1:
+                global_permuted = local_permuted
+            else:
+                lp = local_permuted.contiguous()
+                out = torch.empty((sum(output_splits), lp.size(1)),
+                                  dtype=lp.dtype, device=lp.device)
+                dist.all_to_all_single(out, lp,
+                                       output_split_sizes=list(output_splits),
+                                       input_split_sizes=list(input_splits),
+                                       group=group)
+                global_permuted = out
+            sorted_tokens = sort_chunks_by_idxs(global_permuted, sort_sizes,
+                                                sort_order)
+            y, saved = mlp_fwd(sorted_tokens, cumsum, *weights)
+            ctx.save_for_backward(cumsum, *saved)
+            ctx.group = group
+            ctx.splits = (input_splits, output_splits)
+            ctx.sort = (sort_sizes, sort_order)
+            ctx.n_weights = len(weights)
+            return y
+
+        @staticmethod
+        def backward(ctx, dY):
+            cumsum, *saved = ctx.saved_tensors
+            group = ctx.group
+            input_splits, output_splits = ctx.splits
+            sort_sizes, sort_order = ctx.sort
+            ws = dist.get_world_size(group=group) if group is not None else 1
+
+            d_sorted, stash = mlp_bwd_dgrad(dY.contiguous(), cumsum, saved)
+
+            # inverse chunk resort: chunk i of d_sorted has size
+            # sort_sizes[sort_order[i]]; send it back to slot sort_order[i]
+            order = sort_order.tolist() if torch.is_tensor(sort_order) else list(sort_order)
+            sizes = sort_sizes.tolist() if torch.is_tensor(sort_sizes) else list(sort_sizes)
+            inv = [0] * len(order)
+            for i, o in enumerate(order):
+                inv[o] = i
+            d_global = sort_chunks_by_idxs(d_sorted, [sizes[o] for o in order], inv)
+
+            # launch the return a2a NOW (comm stream) ...
+            work = None
+            if ws == 1:
+                d_local = d_global
+            else:
+                d_global = d_global.contiguous()
+                d_local = torch.empty((sum(input_splits), d_global.size(1)),
+                                      dtype=d_global.dtype, device=d_global.device)
+                work = dist.all_to_all_single(
+                    d_local, d_global, output_split_sizes=list(input_splits),
+                    input_split_sizes=list(output_splits), group=group,
+                    async_op=True)
+            # ... and run the wgrad GEMMs underneath it
+            dws = mlp_bwd_wgrad(cumsum, saved, stash)
+            if work is not None:
+                work.wait()
+            return (d_local, None, None, None, None, None, None, *dws)
+
+    return _EPA2AClass
+
+
+def dispatch_to_ep_a2a_class(ep_a2a_class, num_experts: int, routing_weights,
+                             selected_experts, hidden_states, *weights):
+    """dispatch_to_ep_class variant for make_ep_a2a_class classes: the
+    dispatch a2a + resort live inside the autograd class (overlapped
+    backward); the combine side is unchanged (tokens_post_all2all)."""
+    ep_state = get_parallel_state()
+    ep_group = ep_state.ep_group
+    expert_mask = torch.nn.functional.one_hot(
+        selected_experts, num_classes=num_experts
+    ).permute(2, 1, 0)
+    input_splits, output_splits, num_global_per_local, num_global_sum_per_local = preprocess(
+        expert_mask, num_experts, ep_group
+    )
+    hidden_dim = hidden_states.size(-1)
+    hs = hidden_states.reshape(-1, hidden_dim)
+    routing_map = expert_mask.sum(dim=1)
+    local_permuted, local_mapping = permute(hs, routing_map)
+    cumsum = torch.cumsum(num_global_sum_per_local, dim=0).to(hs.device)
+    num_local_experts = num_experts // ep_group.size()
+    sort_order = torch.arange(num_experts).reshape(-1, num_local_experts).T.ravel().tolist()
+    final = ep_a2a_class.apply(
+        local_permuted, cumsum, ep_group, tuple(input_splits),
+        tuple(output_splits), num_global_per_local.ravel(), sort_order, *weights)
+    return tokens_post_all2all(
+        final, routing_weights, selected_experts, num_experts,
+        input_splits, output_splits, num_global_per_local, routing_map,
+        local_mapping, hs.shape, ep_group,
+    )
+
+
 def dispatch_to_ep_class(ep_class: Callable, num_experts: int, routing_weights,
                          selected_experts, hidden_states, *ep_class_args: Any):
     """Shared EP plumbing (ref moe_layer.py:120-186): preprocess ->

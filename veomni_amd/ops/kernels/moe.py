@@ -97,7 +97,10 @@ class HipFusedMoeFunction(torch.autograd.Function):
 
 class EPMergedFc1HipGroupGemm(torch.autograd.Function):
     """EP expert-MLP (parity: EPMergedFc1GroupGemm, moe_layer.py:414-567);
-    routing weights applied later by `unpermute` in tokens_post_all2all."""
+    routing weights applied later by `unpermute` in tokens_post_all2all.
+    Kept for the plain dispatch path; the DISPATCHED EP path is
+    EPMergedFc1HipGroupGemmA2A below (owns the dispatch a2a for the
+    wgrad/a2a backward overlap)."""
 
     @staticmethod
     def forward(ctx, permute_tokens, cumsum, fc1_1_2_weight, fc2_weight):
@@ -122,6 +125,44 @@ class EPMergedFc1HipGroupGemm(torch.autograd.Function):
         return d_tokens, None, d_fc1_w, d_fc2_w
 
 
+# ---- overlapped EP class: dispatch a2a owned by the autograd node so its
+# backward launches the return exchange before the wgrad GEMMs ------------
+
+def _ep_mlp_fwd(tokens, cumsum, fc1_1_2_weight, fc2_weight):
+    fc1 = hip_lib.group_gemm_nk(tokens, fc1_1_2_weight, cumsum, trans_b=True)
+    act = hip_lib.silu_mul_weighted(fc1, None)
+    fc2 = hip_lib.group_gemm_nk(act, fc2_weight, cumsum, trans_b=True)
+    return fc2, (tokens, fc1_1_2_weight, fc2_weight, fc1, act)
+
+
+def _ep_mlp_bwd_dgrad(dY, cumsum, saved):
+    _, w1, w2, fc1, _ = saved
+    d_act = hip_lib.group_gemm_nk(dY, hip_lib.weight_transpose(w2), cumsum,
+                                  trans_b=True)
+    d_fc1, _ = hip_lib.silu_mul_weighted_bwd(d_act, fc1, None)
+    d_tokens = hip_lib.group_gemm_nk(d_fc1, hip_lib.weight_transpose(w1),
+                                     cumsum, trans_b=True)
+    return d_tokens, (dY, d_fc1)
+
+
+def _ep_mlp_bwd_wgrad(cumsum, saved, stash):
+    tokens, _, w2, _, act = saved
+    dY, d_fc1 = stash
+    G = w2.shape[0]
+    d_fc2_w = hip_lib.group_gemm_mn(dY, act, cumsum, G)
+    d_fc1_w = hip_lib.group_gemm_mn(d_fc1, tokens, cumsum, G)
+    return (d_fc1_w, d_fc2_w)
+
+
+def _make_ep_a2a():
+    from ...distributed.moe import make_ep_a2a_class
+
+    return make_ep_a2a_class(_ep_mlp_fwd, _ep_mlp_bwd_dgrad, _ep_mlp_bwd_wgrad)
+
+
+EPMergedFc1HipGroupGemmA2A = _make_ep_a2a()
+
+
 def hip_fused_moe_forward(num_experts, routing_weights, selected_experts,
                           hidden_states, fc1_1_weight, fc1_2_weight, fc2_weight,
                           fc1_1_2_weight=None, swiglu_limit=None):
@@ -137,8 +178,10 @@ def hip_fused_moe_forward(num_experts, routing_weights, selected_experts,
         fc1_1_2_weight = torch.cat([fc1_1_weight, fc1_2_weight], dim=1).contiguous()
     hs = hidden_states.reshape(-1, hidden_states.shape[-1])
     if get_parallel_state().ep_enabled:
-        out = dispatch_to_ep_class(
-            EPMergedFc1HipGroupGemm, num_experts, routing_weights,
+        from ...distributed.moe import dispatch_to_ep_a2a_class
+
+        out = dispatch_to_ep_a2a_class(
+            EPMergedFc1HipGroupGemmA2A, num_experts, routing_weights,
             selected_experts, hs, fc1_1_2_weight, fc2_weight,
         )
     else:
