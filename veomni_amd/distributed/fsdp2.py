@@ -45,6 +45,7 @@ def build_parallelize_model(
     param_dtype: torch.dtype = torch.bfloat16,
     reduce_dtype: torch.dtype = torch.float32,
     enable_forward_prefetch: bool = True,
+    force_wrap: bool = False,
 ) -> nn.Module:
     """Apply EP slicing + per-layer FSDP2 wrap. Entry point parity:
     ref torch_parallelize.py:590 (`build_parallelize_model`)."""
@@ -55,9 +56,11 @@ def build_parallelize_model(
     if plan is not None and ps.ep_enabled:
         plan.apply(model)
 
-    if ps.fsdp_size == 1 and not ps.ep_enabled:
+    if ps.fsdp_size == 1 and not ps.ep_enabled and not force_wrap:
         # degenerate world: fully_shard would only add allgather/copy churn
         # (no actual sharding); keep the plain module + the same clip entry.
+        # force_wrap=True runs the full FSDP2 mechanics anyway (1-GPU RCCL
+        # smoke of the wrap/prefetch/reduce machinery).
         model.clip_grad_norm_ = functools.partial(clip_grad_norm, model)
         return model
 
