@@ -70,11 +70,22 @@ int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                            const int64_t* cumsum, int G, int64_t N, int64_t K,
                            int64_t total_rows, int trans_b, void* stream);
 
-/* 256-square double-buffered glds variant (trans_b semantics; K % 64 == 0):
- * the fastest forward kernel; auto-dispatched for large trans_b shapes. */
+/* 256-square double-buffered glds variant (trans_b semantics; K % 64 == 0);
+ * kept for A/B benchmarking — the dispatched fast path is nk256s below. */
 int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
                              const int64_t* cumsum, int G, int64_t N,
                              int64_t K, int64_t total_rows, void* stream);
+
+/* nk256 inner loop under a device-built tile schedule with XCD-clustered
+ * persistent blocks (L2 reuse of the per-group A/B tiles; no skew-sized
+ * null-block grid). The dispatched trans_b fast path. G <= 4096. NOTE: on
+ * first use this entry hipMallocs one small (~16 KB) schedule workspace
+ * that lives for the process (the single exception to the no-allocation
+ * convention; calls on one stream serialize on it). */
+int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
+                              uint16_t* C, const int64_t* cumsum, int G,
+                              int64_t N, int64_t K, int64_t total_rows,
+                              void* stream);
 
 /* Register-staged 256-square ring variants (auto-dispatched):
  * dgrad8 = !trans_b semantics; mn8 = wgrad (A^T B) semantics. */

@@ -96,6 +96,45 @@ def main():
     t = timeit(run256b)
     print(f"nk256 fc2 fwd (M{rows} N{H} K{I} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
 
+    # nk256s (XCD-clustered persistent schedule) A/B at the same shapes
+    def run256s():
+        rc = lib2.vh_group_gemm_nk256s_bf16(a.data_ptr(), w1.data_ptr(), c256.data_ptr(),
+                                            cumsum.data_ptr(), E, 2 * I, H, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    fl = 2.0 * rows * 2 * I * H
+    t = timeit(run256s)
+    ok = torch.allclose(c256.float(), ref.float(), rtol=2e-2, atol=2e-2)
+    print(f"nk256s fc1 fwd (M{rows} N{2*I} K{H} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s  correct={ok}")
+    def run256sb():
+        rc = lib2.vh_group_gemm_nk256s_bf16(act.data_ptr(), w2.data_ptr(), c256b.data_ptr(),
+                                            cumsum.data_ptr(), E, H, I, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    fl = 2.0 * rows * H * I
+    t = timeit(run256sb)
+    print(f"nk256s fc2 fwd (M{rows} N{H} K{I} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
+
+    # non-EP bench shape: G=128 experts (the N=1 30B case), 64k rows sample
+    E2, rows2 = 128, 65536
+    cs2 = torch.full((E2,), rows2 // E2).cumsum(0).to(dev)
+    a2 = (torch.randn(rows2, H, device=dev) * 0.3).to(torch.bfloat16)
+    w1b = (torch.randn(E2, 2 * I, H, device=dev) * 0.3).to(torch.bfloat16)
+    c2 = torch.empty(rows2, 2 * I, dtype=torch.bfloat16, device=dev)
+    fl = 2.0 * rows2 * 2 * I * H
+    def run256_g128():
+        rc = lib2.vh_group_gemm_nk256_bf16(a2.data_ptr(), w1b.data_ptr(), c2.data_ptr(),
+                                           cs2.data_ptr(), E2, 2 * I, H, rows2, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    t = timeit(run256_g128)
+    print(f"nk256  fc1 fwd (M{rows2} N{2*I} K{H} G{E2}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
+    def run256s_g128():
+        rc = lib2.vh_group_gemm_nk256s_bf16(a2.data_ptr(), w1b.data_ptr(), c2.data_ptr(),
+                                            cs2.data_ptr(), E2, 2 * I, H, rows2, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    t = timeit(run256s_g128)
+    ref2 = L.group_gemm_nk(a2[:4096], w1b[:2], torch.tensor([2048, 4096], device=dev), trans_b=True)
+    ok2 = torch.allclose(c2[:4096].float(), ref2.float(), rtol=2e-2, atol=2e-2)
+    print(f"nk256s fc1 fwd (M{rows2} N{2*I} K{H} G{E2}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s  correct={ok2}")
+
     # hipBLASLt comparison: one dense bf16 GEMM of the fc1-fwd size
     bd = w1.reshape(E * 2 * I, H)
     t = timeit(lambda: torch.matmul(a, bd.t()))

@@ -356,6 +356,10 @@ extern "C" int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B,
                                         uint16_t* C, const int64_t* cumsum,
                                         int G, int64_t N, int64_t K,
                                         int64_t total_rows, void* stream);
+extern "C" int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
+                                         uint16_t* C, const int64_t* cumsum,
+                                         int G, int64_t N, int64_t K,
+                                         int64_t total_rows, void* stream);
 extern "C" int vh_group_gemm_nk8_bf16(const uint16_t* A, const uint16_t* B,
                                       uint16_t* C, const int64_t* cumsum,
                                       int G, int64_t N, int64_t K,
@@ -381,9 +385,11 @@ extern "C" int vh_group_gemm_nk_bf16(const uint16_t* A, const uint16_t* B,
   //   !trans_b wide   -> nk8   (glds A + transposed B ring)
   //   everything else -> the 128x128 2-phase kernel
   if (!accumulate && activation == 0 && trans_b && K % 64 == 0 && N >= 256 &&
-      total_rows >= (int64_t)G * 256) {
-    return vh_group_gemm_nk256_bf16(A, B, C, cumsum, G, N, K, total_rows,
-                                    stream);
+      total_rows >= (int64_t)G * 256 && G <= 4096) {
+    // nk256s: device-built tile schedule + XCD-clustered persistent blocks
+    // (L2 reuse; no skew-sized null-block grid) — see vh_group_gemm8.hip
+    return vh_group_gemm_nk256s_bf16(A, B, C, cumsum, G, N, K, total_rows,
+                                     stream);
   }
   if (!accumulate && activation == 0 && !trans_b && K % 32 == 0 && K >= 1024 &&
       N >= 1024 && total_rows >= 16 * G * 16) {

@@ -744,7 +744,189 @@ __global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256(
       }
 }
 
+// ============================================================================
+// nk256s: the nk256 inner loop under a DEVICE-BUILT tile schedule with
+// XCD-clustered persistent blocks.
+//
+// Why (PMC, profiles/r02_groupgemm_pmc.txt): nk256's grid (tiles x G)
+// dispatches consecutive blocks round-robin across the 8 XCDs, so blocks
+// sharing a group's A/B tiles land on DIFFERENT L2s — zero reuse. At the
+// bench shapes every 256x256 tile streams ~2 MB from HBM (~6.5 TB/s:
+// HBM-bound at 20% MFMA-busy, vs the Tensile reference's 82%). The grid is
+// also sized for worst-case skew (ceil(total/256) m-tiles for EVERY group):
+// at G=128 that is ~786k blocks, 99% of which exit immediately.
+//
+// Structure:
+//   - k_gg_sched (1 tiny launch) turns cumsum into a per-group tile prefix
+//     sum in a device workspace: group g owns tiles_m_g x tiles_n tiles,
+//     enumerated bm-inner (consecutive tiles share the B n-slab).
+//   - the GEMM launches a FIXED grid of 256 persistent blocks (1/CU at
+//     128 KiB LDS). Block b runs on XCD b%8 (dispatch affinity);
+//     it walks the CONTIGUOUS tile range [X*L, (X+1)*L) of the global list
+//     with stride 32, so the ~32 concurrently-resident blocks of an XCD
+//     process consecutive tiles of the SAME group — the group's B n-slab
+//     (~1 MB) and the shared A m-tiles stay hot in that XCD's 4 MB L2.
+//   - zero-count groups occupy no schedule slots; per-tile group lookup is
+//     a binary search over the prefix (G<=4096).
+// ============================================================================
+
+__global__ void k_gg_sched(const int64_t* __restrict__ cumsum, int G,
+                           int tiles_n, int* __restrict__ ws) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    int total = 0;
+    int64_t prev = 0;
+    for (int g = 0; g < G; ++g) {
+      int64_t m = cumsum[g] - prev;
+      prev = cumsum[g];
+      ws[1 + g] = total;
+      total += (int)((m + BM8 - 1) / BM8) * tiles_n;
+    }
+    ws[1 + G] = total;
+    ws[0] = total;
+  }
+}
+
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256s(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_n, const int* __restrict__ ws) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 32768; };          // 2 x 32 KiB
+  auto tb = [&](int buf) { return sm + 16384 + buf * 32768; };  // 2 x 32 KiB
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const int total_tiles = ws[0];
+  const int X = blockIdx.x & 7;        // this block's XCD (dispatch affinity)
+  const int slot = blockIdx.x >> 3;    // 0..31 within the XCD
+  const int L = (total_tiles + 7) / 8; // contiguous tile range per XCD
+  const int t_end = min((X + 1) * L, total_tiles);
+  const int nk = (int)(K / BK64);
+
+  for (int t = X * L + slot; t < t_end; t += 32) {
+    // group lookup: largest g with ws[1+g] <= t
+    int lo = 0, hi = G - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (ws[1 + mid] <= t) lo = mid;
+      else hi = mid - 1;
+    }
+    const int gid = lo;
+    const int local = t - ws[1 + gid];
+    const int tiles_m_g = (ws[2 + gid] - ws[1 + gid]) / tiles_n;
+    const int bm = local % tiles_m_g;   // bm-inner: consecutive t share bn
+    const int bn = local / tiles_m_g;
+
+    const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+    const int64_t m_size = cumsum[gid] - row_start;
+    const bf16_t* Ag = A + row_start * K;
+    const bf16_t* Bg = B + (int64_t)gid * N * K;
+    bf16_t* Cg = C + row_start * N;
+
+    KStage64 sa, sb;
+    sa.init(Ag, K, [&](int r) -> int64_t {
+      int64_t gm = (int64_t)bm * BM8 + r;
+      return gm % m_size;
+    }, tid);
+    sb.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    sa.stage(ta(0), 0);
+    sb.stage(tb(0), 0);
+    __syncthreads();
+    int cur = 0;
+    for (int kt = 0; kt < nk; ++kt) {
+      if (kt + 1 < nk) {
+        sa.stage(ta(cur ^ 1), (int64_t)(kt + 1) * BK64);
+        sb.stage(tb(cur ^ 1), (int64_t)(kt + 1) * BK64);
+      }
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16frag af[4], bfr[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) bfr[j] = frag_read64(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + i * 16, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + 64 + i * 16, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+
+    const int col_in = lane & 15;
+    const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                      row_base_in + rr;
+          int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+          if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+        }
+  }
+}
+
 }  // namespace
+
+// lazily-allocated device workspace for the tile schedule (ws[0] = total,
+// ws[1..G+1] = per-group tile prefix). Single compute stream per process;
+// calls on one stream serialize, so one buffer suffices.
+static int* vh_gg_sched_ws = nullptr;
+static constexpr int VH_GG_SCHED_MAX_G = 4096;
+
+extern "C" int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
+                                         uint16_t* C, const int64_t* cumsum,
+                                         int G, int64_t N, int64_t K,
+                                         int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % BK64 == 0, "K %% 64 != 0");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  VH_CHECK(G <= VH_GG_SCHED_MAX_G, "G > %d", VH_GG_SCHED_MAX_G);
+  if (vh_gg_sched_ws == nullptr) {
+    VH_HIP(hipMalloc(&vh_gg_sched_ws, (VH_GG_SCHED_MAX_G + 2) * sizeof(int)));
+  }
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  hipLaunchKernelGGL(k_gg_sched, dim3(1), dim3(64), 0, s, cumsum, G, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  hipLaunchKernelGGL(k_group_gemm_nk256s, dim3(256), dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
 
 extern "C" int vh_group_gemm_nk256_bf16(const uint16_t* A, const uint16_t* B,
                                         uint16_t* C, const int64_t* cumsum,
