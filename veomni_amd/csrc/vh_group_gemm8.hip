@@ -1109,6 +1109,119 @@ __global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_wg256(
       }
 }
 
+
+// wg256 under the XCD-clustered persistent schedule (see nk256s): every
+// group owns the same tiles_m x tiles_n output tiles (weight-shaped C), so
+// the schedule needs no device prefix — block b (XCD b%8) walks the
+// contiguous tile range [X*L, (X+1)*L) with stride 32, bm-inner, keeping a
+// group's B' k-slab hot in its XCD's L2.
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_wg256s(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ padded_cumsum, int G,
+    int64_t M, int64_t N, int64_t PR, int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 32768; };
+  auto tb = [&](int buf) { return sm + 16384 + buf * 32768; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const int tpg = tiles_m * tiles_n;
+  const int total = G * tpg;
+  const int X = blockIdx.x & 7;
+  const int slot = blockIdx.x >> 3;
+  const int L = (total + 7) / 8;
+  const int t_end = min((X + 1) * L, total);
+
+  for (int t = X * L + slot; t < t_end; t += 32) {
+    const int gid = t / tpg;
+    const int local = t - gid * tpg;
+    const int bm = local % tiles_m;   // bm-inner: consecutive t share bn
+    const int bn = local / tiles_m;
+
+    const int64_t p_start = (gid > 0) ? padded_cumsum[gid - 1] : 0;
+    const int64_t klen = padded_cumsum[gid] - p_start;
+    bf16_t* Cg = C + (int64_t)gid * M * N;
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    if (klen > 0) {
+      const bf16_t* Ag = A + p_start;
+      const bf16_t* Bg = B + p_start;
+      KStage64 sa, sb;
+      sa.init(Ag, PR, [&](int r) -> int64_t {
+        int64_t gm = (int64_t)bm * BM8 + r;
+        return gm % M;
+      }, tid);
+      sb.init(Bg, PR, [&](int r) -> int64_t {
+        int64_t gn = (int64_t)bn * BN8 + r;
+        return gn % N;
+      }, tid);
+
+      const int nk = (int)(klen / BK64);
+      sa.stage(ta(0), 0);
+      sb.stage(tb(0), 0);
+      __syncthreads();
+      int cur = 0;
+      for (int kt = 0; kt < nk; ++kt) {
+        if (kt + 1 < nk) {
+          sa.stage(ta(cur ^ 1), (int64_t)(kt + 1) * BK64);
+          sb.stage(tb(cur ^ 1), (int64_t)(kt + 1) * BK64);
+        }
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          bf16frag af[4], bfr[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j) bfr[j] = frag_read64(tb(cur), wc * 64 + j * 16, ks, lane);
+#pragma unroll
+          for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + i * 16, ks, lane);
+          __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+          for (int i = 0; i < 4; ++i) af[i] = frag_read64(ta(cur), wr * 128 + 64 + i * 16, ks, lane);
+          __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+        }
+        __syncthreads();
+        cur ^= 1;
+      }
+    } else {
+      // keep barrier parity with peer waves (none needed: no LDS touched)
+    }
+
+    const int col_in = lane & 15;
+    const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                      row_base_in + rr;
+          int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+          if (m < M && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+        }
+  }
+}
+
 }  // namespace
 
 extern "C" int vh_transpose_pad_bf16(const uint16_t* src, uint16_t* dst,
@@ -1137,8 +1250,7 @@ extern "C" int vh_group_gemm_wg256_bf16(const uint16_t* A, const uint16_t* B,
   VH_CHECK(PR % 64 == 0, "PR %% 64 != 0");
   int tiles_m = (int)((M + BM8 - 1) / BM8);
   int tiles_n = (int)((N + BN8 - 1) / BN8);
-  dim3 grid(tiles_m * tiles_n, G);
-  hipLaunchKernelGGL(k_group_gemm_wg256, grid, dim3(THREADS8), 131072, s,
+  hipLaunchKernelGGL(k_group_gemm_wg256s, dim3(256), dim3(THREADS8), 131072, s,
                      reinterpret_cast<const bf16_t*>(A),
                      reinterpret_cast<const bf16_t*>(B),
                      reinterpret_cast<bf16_t*>(C), padded_cumsum, G, M, N, PR,
