@@ -113,6 +113,23 @@ def main():
     t = timeit(run256sb)
     print(f"nk256s fc2 fwd (M{rows} N{H} K{I} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
 
+    # nk8s (counted-vmcnt ring + XCD schedule) A/B
+    def run8s():
+        rc = lib2.vh_group_gemm_nk8s_bf16(a.data_ptr(), w1.data_ptr(), c256.data_ptr(),
+                                          cumsum.data_ptr(), E, 2 * I, H, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    fl = 2.0 * rows * 2 * I * H
+    t = timeit(run8s)
+    ok = torch.allclose(c256.float(), ref.float(), rtol=2e-2, atol=2e-2)
+    print(f"nk8s  fc1 fwd (M{rows} N{2*I} K{H} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s  correct={ok}")
+    def run8sb():
+        rc = lib2.vh_group_gemm_nk8s_bf16(act.data_ptr(), w2.data_ptr(), c256b.data_ptr(),
+                                          cumsum.data_ptr(), E, H, I, rows, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    fl = 2.0 * rows * H * I
+    t = timeit(run8sb)
+    print(f"nk8s  fc2 fwd (M{rows} N{H} K{I} G{E}): {t*1e3:.2f} ms  {fl/t/1e12:.0f} TF/s")
+
     # non-EP bench shape: G=128 experts (the N=1 30B case), 64k rows sample
     E2, rows2 = 128, 65536
     cs2 = torch.full((E2,), rows2 // E2).cumsum(0).to(dev)
@@ -130,6 +147,12 @@ def main():
         rc = lib2.vh_group_gemm_nk256s_bf16(a2.data_ptr(), w1b.data_ptr(), c2.data_ptr(),
                                             cs2.data_ptr(), E2, 2 * I, H, rows2, L.cur_stream())
         assert rc == 0, lib2.vh_last_error()
+    def run8s_g128():
+        rc = lib2.vh_group_gemm_nk8s_bf16(a2.data_ptr(), w1b.data_ptr(), c2.data_ptr(),
+                                          cs2.data_ptr(), E2, 2 * I, H, rows2, L.cur_stream())
+        assert rc == 0, lib2.vh_last_error()
+    t8 = timeit(run8s_g128)
+    print(f"nk8s   fc1 fwd (M{rows2} N{2*I} K{H} G{E2}): {t8*1e3:.2f} ms  {fl/t8/1e12:.0f} TF/s")
     t = timeit(run256s_g128)
     gsz = rows2 // E2
     ok2 = all(

@@ -896,6 +896,137 @@ __global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256s(
   }
 }
 
+
+// nk8's 4-deep KSUB=32 counted-vmcnt ring under the same XCD-clustered
+// persistent schedule (trans_b path only — its transposed-write staging for
+// !trans_b has a 8-16-way LDS write conflict, and dgrad routes through the
+// weight transpose + trans_b anyway). With the schedule removing the HBM
+// wall, the deeper pipeline's shorter barrier drains can show.
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk8s(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_n, const int* __restrict__ ws) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto slotA = [&](int s) { return sm + (s & 3) * 8192; };          // 4 x 16 KiB
+  auto slotB = [&](int s) { return sm + 32768 + (s & 3) * 8192; };  // 4 x 16 KiB
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const int total_tiles = ws[0];
+  const int X = blockIdx.x & 7;
+  const int slot_id = blockIdx.x >> 3;
+  const int L = (total_tiles + 7) / 8;
+  const int t_end = min((X + 1) * L, total_tiles);
+  const int nsub = (int)(K / KSUB);
+
+  for (int t = X * L + slot_id; t < t_end; t += 32) {
+    int lo = 0, hi = G - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (ws[1 + mid] <= t) lo = mid;
+      else hi = mid - 1;
+    }
+    const int gid = lo;
+    const int local = t - ws[1 + gid];
+    const int tiles_m_g = (ws[2 + gid] - ws[1 + gid]) / tiles_n;
+    const int bm = local % tiles_m_g;
+    const int bn = local / tiles_m_g;
+
+    const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+    const int64_t m_size = cumsum[gid] - row_start;
+    const bf16_t* Ag = A + row_start * K;
+    const bf16_t* Bg = B + (int64_t)gid * N * K;
+    bf16_t* Cg = C + row_start * N;
+
+    KStage8 sa, sb;
+    sa.init(Ag, K, [&](int r) -> int64_t {
+      int64_t gm = (int64_t)bm * BM8 + r;
+      return gm % m_size;
+    }, tid);
+    sb.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+
+    auto stage = [&](int s, int64_t k0) {
+      sa.stage(slotA(s), k0);
+      sb.stage(slotB(s), k0);
+    };
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    stage(0, 0);
+    stage(1, KSUB);
+    if (nsub > 2) stage(2, 2 * KSUB);
+    for (int s = 0; s < nsub; ++s) {
+      if (s + 3 < nsub) stage(s + 3, (int64_t)(s + 3) * KSUB);
+      {
+        const int rem = nsub - s;
+        if (rem >= 4) VMCNT(12);
+        else if (rem == 3) VMCNT(8);
+        else if (rem == 2) VMCNT(4);
+        else VMCNT(0);
+      }
+      raw_barrier();
+      {
+        const bf16_t* TA = slotA(s);
+        const bf16_t* TB = slotB(s);
+        bf16frag af[4], bfr[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read8(TA, wr * 128 + i * 16, lane);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(TB, wc * 64 + j * 16, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        raw_barrier();
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = frag_read8(TA, wr * 128 + 64 + i * 16, lane);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(TB, wc * 64 + j * 16, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      raw_barrier();
+    }
+
+    const int col_in = lane & 15;
+    const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                      row_base_in + rr;
+          int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+          if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+        }
+    // drain this tile's in-flight glds before the ring is re-staged for the
+    // next tile (slots would otherwise be overwritten while still pending)
+    VMCNT(0);
+    raw_barrier();
+  }
+}
+
 }  // namespace
 
 // lazily-allocated device workspace for the tile schedule (ws[0] = total,
@@ -920,6 +1051,31 @@ extern "C" int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
                      vh_gg_sched_ws);
   VH_HIP(hipGetLastError());
   hipLaunchKernelGGL(k_group_gemm_nk256s, dim3(256), dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* probe: nk8's counted-vmcnt ring + the XCD schedule (trans_b only). */
+extern "C" int vh_group_gemm_nk8s_bf16(const uint16_t* A, const uint16_t* B,
+                                       uint16_t* C, const int64_t* cumsum,
+                                       int G, int64_t N, int64_t K,
+                                       int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % KSUB == 0 && K / KSUB >= 2, "K must be a multiple of 32, >= 64");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  VH_CHECK(G <= VH_GG_SCHED_MAX_G, "G > max");
+  if (vh_gg_sched_ws == nullptr) {
+    VH_HIP(hipMalloc(&vh_gg_sched_ws, (VH_GG_SCHED_MAX_G + 2) * sizeof(int)));
+  }
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  hipLaunchKernelGGL(k_gg_sched, dim3(1), dim3(64), 0, s, cumsum, G, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  hipLaunchKernelGGL(k_group_gemm_nk8s, dim3(256), dim3(THREADS8), 131072, s,
                      reinterpret_cast<const bf16_t*>(A),
                      reinterpret_cast<const bf16_t*>(B),
                      reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
