@@ -1027,6 +1027,183 @@ __global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk8s(
   }
 }
 
+
+// ============================================================================
+// nkp: the guide's 256-square 8-phase pipeline (cdna_hip_programming.md §5
+// "The 256² 8-phase template"), reconstructed for the GROUPED case and run
+// under the XCD-clustered persistent schedule.
+//
+// Reading of the template made self-consistent here (the example file is
+// not shipped): a K-tile (BK=64) is staged as FOUR K-SPLIT half-tiles
+// (A-k0, B-k0, A-k1, B-k1; each [256 rows][32 k] = 16 KiB, the proven
+// conflict-free KStage8/swz32 layout), double-buffered = 8 ring slots =
+// 128 KiB. Four phases per K-tile, each = one k-chunk x one m-half of the
+// wave's 128x64 panel (16 MFMA); B-frags of a k-chunk are read once (ph0/
+// ph2) and reused by the mh=1 phase. Each phase: [ds_reads; stage ONE
+// half-tile of tile kt+1; (vmcnt clearing at ph1/ph3); raw barrier;
+// lgkmcnt(0) implicit via frag use; setprio around the MFMA; raw barrier].
+// Clearing schedule (per-wave glds = 2 per half): ph1 clears k1(kt)
+// [vmcnt(4): outstanding = k1(kt) + k0(kt+1) = 8 loads], ph3 clears
+// k0(kt+1) [vmcnt(4)]; last tile uses vmcnt(0) at ph1. ds_reads lead each
+// phase: they target slots cleared >= 1 phase earlier, so they overlap the
+// barrier wait.
+// ============================================================================
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nkp(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_n, const int* __restrict__ ws) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  // slot(t, h): h = 0 A-k0, 1 B-k0, 2 A-k1, 3 B-k1
+  auto slot = [&](int t, int h) { return sm + (((t & 1) << 2) + h) * 8192; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const int total_tiles = ws[0];
+  const int X = blockIdx.x & 7;
+  const int slot_id = blockIdx.x >> 3;
+  const int L = (total_tiles + 7) / 8;
+  const int t_end = min((X + 1) * L, total_tiles);
+  const int ntk = (int)(K / 64);
+
+  for (int t = X * L + slot_id; t < t_end; t += 32) {
+    int lo = 0, hi = G - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (ws[1 + mid] <= t) lo = mid;
+      else hi = mid - 1;
+    }
+    const int gid = lo;
+    const int local = t - ws[1 + gid];
+    const int tiles_m_g = (ws[2 + gid] - ws[1 + gid]) / tiles_n;
+    const int bm = local % tiles_m_g;
+    const int bn = local / tiles_m_g;
+
+    const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+    const int64_t m_size = cumsum[gid] - row_start;
+    const bf16_t* Ag = A + row_start * K;
+    const bf16_t* Bg = B + (int64_t)gid * N * K;
+    bf16_t* Cg = C + row_start * N;
+
+    KStage8 sa, sb;
+    sa.init(Ag, K, [&](int r) -> int64_t {
+      int64_t gm = (int64_t)bm * BM8 + r;
+      return gm % m_size;
+    }, tid);
+    sb.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    // prologue: tile 0's four halves, then clear its k0 pair
+    sa.stage(slot(0, 0), 0);
+    sb.stage(slot(0, 1), 0);
+    sa.stage(slot(0, 2), KSUB);
+    sb.stage(slot(0, 3), KSUB);
+    VMCNT(4);
+    raw_barrier();
+
+    for (int kt = 0; kt < ntk; ++kt) {
+      const bool more = kt + 1 < ntk;
+      const int64_t nk0 = (int64_t)(kt + 1) * 64;
+      bf16frag af[4], bfr[4];
+
+      // ---- ph0: (k0, mh0); stage A-k0(kt+1)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(slot(kt, 1), wc * 64 + j * 16, lane);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(slot(kt, 0), wr * 128 + i * 16, lane);
+      if (more) sa.stage(slot(kt + 1, 0), nk0);
+      raw_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      raw_barrier();
+
+      // ---- ph1: (k0, mh1); stage B-k0(kt+1); clear k1(kt)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(slot(kt, 0), wr * 128 + 64 + i * 16, lane);
+      if (more) {
+        sb.stage(slot(kt + 1, 1), nk0);
+        VMCNT(4);
+      } else {
+        VMCNT(0);
+      }
+      raw_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      raw_barrier();
+
+      // ---- ph2: (k1, mh0); stage A-k1(kt+1)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bfr[j] = frag_read8(slot(kt, 3), wc * 64 + j * 16, lane);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(slot(kt, 2), wr * 128 + i * 16, lane);
+      if (more) sa.stage(slot(kt + 1, 2), nk0 + KSUB);
+      raw_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      raw_barrier();
+
+      // ---- ph3: (k1, mh1); stage B-k1(kt+1); clear k0(kt+1)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) af[i] = frag_read8(slot(kt, 2), wr * 128 + 64 + i * 16, lane);
+      if (more) {
+        sb.stage(slot(kt + 1, 3), nk0 + KSUB);
+        VMCNT(4);
+      }
+      raw_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bfr[j], acc[i + 4][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      raw_barrier();
+    }
+
+    const int col_in = lane & 15;
+    const int row_base_in = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          int64_t m = (int64_t)bm * BM8 + wr * 128 + (i & 3) * 16 + (i >> 2) * 64 +
+                      row_base_in + rr;
+          int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 16 + col_in;
+          if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][rr]);
+        }
+    VMCNT(0);
+    raw_barrier();
+  }
+}
+
 }  // namespace
 
 // lazily-allocated device workspace for the tile schedule (ws[0] = total,
@@ -1051,6 +1228,31 @@ extern "C" int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
                      vh_gg_sched_ws);
   VH_HIP(hipGetLastError());
   hipLaunchKernelGGL(k_group_gemm_nk256s, dim3(256), dim3(THREADS8), 131072, s,
+                     reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* probe: the 8-phase K-split pipeline + XCD schedule (trans_b only). */
+extern "C" int vh_group_gemm_nkp_bf16(const uint16_t* A, const uint16_t* B,
+                                      uint16_t* C, const int64_t* cumsum,
+                                      int G, int64_t N, int64_t K,
+                                      int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % 64 == 0, "K %% 64 != 0");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  VH_CHECK(G <= VH_GG_SCHED_MAX_G, "G > max");
+  if (vh_gg_sched_ws == nullptr) {
+    VH_HIP(hipMalloc(&vh_gg_sched_ws, (VH_GG_SCHED_MAX_G + 2) * sizeof(int)));
+  }
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  hipLaunchKernelGGL(k_gg_sched, dim3(1), dim3(64), 0, s, cumsum, G, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  hipLaunchKernelGGL(k_group_gemm_nkp, dim3(256), dim3(THREADS8), 131072, s,
                      reinterpret_cast<const bf16_t*>(A),
                      reinterpret_cast<const bf16_t*>(B),
                      reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,

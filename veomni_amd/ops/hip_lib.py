@@ -56,6 +56,8 @@ def get_lib() -> ctypes.CDLL:
          c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_nk8s_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
          c_i64, c_i64, c_p)
+    _sig(lib, "vh_group_gemm_nkp_bf16", c_p, c_p, c_p, c_p, c_int, c_i64,
+         c_i64, c_i64, c_p)
     _sig(lib, "vh_group_gemm_mn_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
     _sig(lib, "vh_transpose_pad_bf16", c_p, c_p, c_p, c_p, c_int, c_i64, c_i64, c_p)
     _sig(lib, "vh_wtranspose_bf16", c_p, c_p, c_int, c_i64, c_i64, c_p)
