@@ -235,8 +235,19 @@ def main():
         rec = prof["group_gemm_nk"]
         fl = sum(rec["work"]) / rec["count"]
         ach = fl / (rec["ms_avg"] / 1000.0) / 1e12
+        # HBM bytes/launch for the dominant (fc1 mbs8) launch shape, from the
+        # committed rocprofv3 TCC passes at that exact shape
+        # (profiles/r02_gg_traffic.txt: FETCH_SIZE x2 gfx950 correction +
+        # WRITE_SIZE; separate --pmc runs per the slot limits). Only reported
+        # for the default 30B workload the passes were collected on.
+        traffic = 3.761e9 if (preset == "qwen3-moe-30b" and mbs == 8
+                              and seq == 4096) else None
         roofline = {"bound": "mfma", "achieved": round(ach, 1), "peak": 2500.0,
-                    "unit": "TFLOP/s", "frac": round(ach / 2500.0, 4), "traffic": None,
+                    "unit": "TFLOP/s", "frac": round(ach / 2500.0, 4),
+                    "traffic": traffic,
+                    "traffic_source": ("rocprofv3 TCC pass at the fc1 mbs8 "
+                                       "launch shape; profiles/r02_gg_traffic.txt"
+                                       if traffic else None),
                     "kernel": "vh_group_gemm_nk*_bf16 (shape-dispatched)"}
     elif "ce_fwd" in prof:
         rec = prof["ce_fwd"]
