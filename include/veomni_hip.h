@@ -87,6 +87,17 @@ int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
                               int64_t N, int64_t K, int64_t total_rows,
                               void* stream);
 
+/* Probe kernels kept for on-box A/B (trans_b semantics, XCD schedule):
+ * nk8s = KSUB=32 counted-vmcnt 4-deep ring; nkp = 8-phase K-split pipeline
+ * reconstruction. Both race-screened correct; neither dispatched (measured
+ * slower than nk256s — see profiles/r02_groupgemm_pmc.txt / DESIGN.md). */
+int vh_group_gemm_nk8s_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                            const int64_t* cumsum, int G, int64_t N,
+                            int64_t K, int64_t total_rows, void* stream);
+int vh_group_gemm_nkp_bf16(const uint16_t* A, const uint16_t* B, uint16_t* C,
+                           const int64_t* cumsum, int G, int64_t N, int64_t K,
+                           int64_t total_rows, void* stream);
+
 /* Register-staged 256-square ring variants (auto-dispatched):
  * dgrad8 = !trans_b semantics; mn8 = wgrad (A^T B) semantics. */
 int vh_group_gemm_dgrad8_bf16(const uint16_t* A, const uint16_t* B,
