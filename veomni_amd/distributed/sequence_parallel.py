@@ -190,6 +190,16 @@ def gather_outputs(x: Tensor, gather_dim: int, padding_dim: Optional[int] = None
     return x
 
 
+def slice_input_tensor(x: Tensor, dim: int, scale_grad: bool = True,
+                       group=None) -> Tensor:
+    """Slice this rank's seq chunk (ref sequence_parallel/data.py
+    slice_input_tensor); backward all-gathers."""
+    group = get_ulysses_sequence_parallel_group() if group is None else group
+    if not group:
+        return x
+    return _Slice.apply(group, x, dim, scale_grad)
+
+
 class ReduceLoss(torch.autograd.Function):
     """Token-weighted SP loss mean with zero-valid guard (ref loss.py:24-65)."""
 

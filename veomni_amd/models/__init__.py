@@ -59,6 +59,61 @@ PRESETS = {
     ),
 }
 
+# VLM presets (VLConfig: text + vision towers), built via build_vl_model
+VL_PRESETS = {}
+
+
+def _init_vl_presets():
+    from .modeling import ModelConfig
+    from .vlm import VisionConfig, VLConfig
+
+    if VL_PRESETS:
+        return
+    # BASELINE config 4: Qwen2.5-VL-7B (text = Qwen2.5-7B w/ attention bias
+    # + mrope sections [16,24,24]; vision: 32-block windowed ViT)
+    VL_PRESETS["qwen25-vl-7b"] = VLConfig(
+        text=ModelConfig(
+            name="qwen25-vl-7b-text", vocab_size=152064, hidden_size=3584,
+            intermediate_size=18944, num_hidden_layers=28,
+            num_attention_heads=28, num_key_value_heads=4, head_dim=128,
+            rope_theta=1000000.0, rms_norm_eps=1e-6, attention_bias=True,
+            qk_norm=False, mrope_section=(16, 24, 24),
+        ),
+        vision=VisionConfig(),
+        image_token_id=151655,
+        name="qwen25-vl-7b",
+    )
+    VL_PRESETS["tiny-vl"] = VLConfig(
+        text=ModelConfig(
+            name="tiny-vl-text", vocab_size=512, hidden_size=128,
+            intermediate_size=256, num_hidden_layers=2,
+            num_attention_heads=4, num_key_value_heads=2, head_dim=32,
+            attention_bias=True, qk_norm=False, mrope_section=(4, 6, 6),
+        ),
+        vision=VisionConfig(
+            depth=2, hidden_size=64, num_heads=4, intermediate_size=128,
+            out_hidden_size=128, patch_size=2, temporal_patch_size=1,
+            in_channels=3, spatial_merge_size=2, window_size=8,
+            fullatt_block_indexes=(1,),
+        ),
+        image_token_id=511,
+        name="tiny-vl",
+    )
+
+
+def build_vl_model(preset: str, dtype=None, device=None):
+    import torch
+
+    from .vlm import VLForCausalLM
+
+    _init_vl_presets()
+    cfg = VL_PRESETS[preset]
+    with torch.device(device or "cpu"):
+        model = VLForCausalLM(cfg)
+    if dtype is not None:
+        model = model.to(dtype)
+    return model
+
 
 def build_model(preset: str, dtype=None, device=None, empty_init=False) -> ForCausalLM:
     import torch

@@ -61,6 +61,8 @@ class ModelConfig:
     moe_intermediate_size: int = 768
     norm_topk_prob: bool = True
     router_aux_loss_coef: float = 0.0  # >0 enables the aux loss path
+    # multimodal (Qwen2.5-VL) 3D-RoPE channel sections (t, h, w); None = 1D
+    mrope_section: Optional[tuple] = None
     name: str = "model"
 
     def __post_init__(self):
@@ -111,10 +113,28 @@ class RotaryEmbedding(nn.Module):
             ** (torch.arange(0, config.head_dim, 2, dtype=torch.float32) / config.head_dim)
         )
         self.register_buffer("inv_freq", inv_freq, persistent=False)
+        self.mrope_section = tuple(config.mrope_section) if config.mrope_section else None
 
     @torch.no_grad()
     def forward(self, x: torch.Tensor, position_ids: torch.Tensor):
-        # position_ids [B, S]; fp32 angles, output in x.dtype (HF default rope)
+        # position_ids [B, S] (1D rope) or [3, B, S] (Qwen2.5-VL mrope);
+        # fp32 angles, output in x.dtype (HF default rope). For mrope the
+        # per-axis cos/sin are SECTION-MERGED here once
+        # (ref apply_multimodal_rotary_pos_emb, patched_modeling_qwen2_5_vl
+        # _gpu.py:1027-1068: channel sections t/h/w, duplicated for the
+        # rotate-half second half) — downstream attention then applies
+        # STANDARD rope, so the HIP rope kernel serves the VLM unchanged.
+        if position_ids.dim() == 3:
+            inv = self.inv_freq[None, None, :, None].float().expand(
+                3, position_ids.shape[1], -1, 1)
+            pos = position_ids[:, :, None, :].float()
+            freqs = (inv @ pos).transpose(2, 3)       # [3, B, S, D/2]
+            emb = torch.cat((freqs, freqs), dim=-1)   # [3, B, S, D]
+            cos, sin = emb.cos(), emb.sin()
+            sec = list(self.mrope_section) * 2
+            cos = torch.cat([m[i % 3] for i, m in enumerate(cos.split(sec, dim=-1))], dim=-1)
+            sin = torch.cat([m[i % 3] for i, m in enumerate(sin.split(sec, dim=-1))], dim=-1)
+            return cos.to(x.dtype), sin.to(x.dtype)
         inv = self.inv_freq[None, :, None].float().expand(position_ids.shape[0], -1, 1)
         pos = position_ids[:, None, :].float()
         freqs = (inv @ pos).transpose(1, 2)
@@ -438,10 +458,10 @@ class Model(nn.Module):
         self.rotary_emb = RotaryEmbedding(config)
 
     def forward(self, input_ids, position_ids=None, use_checkpoint=False,
-                attn_kwargs=None):
-        hidden = self.embed_tokens(input_ids)
+                attn_kwargs=None, inputs_embeds=None):
+        hidden = inputs_embeds if inputs_embeds is not None else self.embed_tokens(input_ids)
         if position_ids is None:
-            position_ids = torch.arange(input_ids.shape[1], device=input_ids.device)[None]
+            position_ids = torch.arange(hidden.shape[1], device=hidden.device)[None]
         pos_emb = self.rotary_emb(hidden, position_ids)
         router_logits = []
         for layer in self.layers:
@@ -527,7 +547,8 @@ class ForCausalLM(nn.Module):
                 attn_kwargs = {"doc_start": ds, "doc_end": de}
         hidden, router_logits = self.model(
             input_ids, position_ids=position_ids,
-            use_checkpoint=self.use_checkpoint, attn_kwargs=attn_kwargs
+            use_checkpoint=self.use_checkpoint, attn_kwargs=attn_kwargs,
+            inputs_embeds=kwargs.get("inputs_embeds"),
         )
         if labels is None:
             logits = self.lm_head(hidden)
