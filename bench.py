@@ -128,23 +128,35 @@ def main():
     from veomni_amd.data import synthetic_batch
 
     preset = args.model or "qwen3-moe-30b"
-    cfg = PRESETS[preset]
+    from veomni_amd.models import VL_PRESETS, _init_vl_presets
+    _init_vl_presets()
+    is_vl = preset in VL_PRESETS
+    vl_cfg = VL_PRESETS.get(preset)
+    cfg = vl_cfg.text if is_vl else PRESETS[preset]
     # auto micro-batch, sized for 288 GB HBM at N=1 (measured peaks):
     # llama-8b 4x4096 tokens without checkpointing; the 30B MoE 8x4096 with
     # checkpointing (236 GiB peak incl. bf16 params/grads/optimizer states —
     # larger expert GEMM groups lift the grouped-GEMM rate ~35%).
-    mbs = args.batch or (8 if cfg.is_moe else 4)
+    mbs = args.batch or (1 if is_vl else (8 if cfg.is_moe else 4))
+    if is_vl and args.seq_len == 4096 and args.model:
+        # BASELINE config 4 default: image-text seq 8192
+        args.seq_len = 8192
     ep_size = n_gpus if (cfg.is_moe and n_gpus > 1) else 1
     init_parallel_state(ep_size=ep_size, device_type="cuda")
     bind_ops(HIP_OPS_CONFIG)
 
     log(f"building {preset} on cuda:{local_rank} (ep={ep_size}, ws={world})")
     t_build = time.time()
-    model = build_model(preset, dtype=torch.bfloat16, device="cuda")
+    if is_vl:
+        from veomni_amd.models import build_vl_model
+
+        model = build_vl_model(preset, dtype=torch.bfloat16, device="cuda")
+    else:
+        model = build_model(preset, dtype=torch.bfloat16, device="cuda")
     # 288 GB HBM3E: dense llama-8b at N=1 holds full activations comfortably
     # (no forward recompute). The 30B MoE keeps checkpointing (scattered
     # expert activations are ~1 GB/layer/rank).
-    model.use_checkpoint = cfg.is_moe
+    model.use_checkpoint = cfg.is_moe or is_vl
     model = build_parallelize_model(model)
     def make_opt(kind):
         if kind == "ve":
@@ -178,7 +190,13 @@ def main():
     torch.cuda.reset_peak_memory_stats()
 
     seq = args.seq_len
-    batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs, seed=42 + rank, device="cuda")
+    if is_vl:
+        from veomni_amd.data import synthetic_vlm_batch
+
+        batch = synthetic_vlm_batch(vl_cfg, seq, batch=mbs, seed=42 + rank,
+                                    device="cuda")
+    else:
+        batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs, seed=42 + rank, device="cuda")
 
     n_valid = (batch["labels"] != -100).sum()
 
@@ -227,6 +245,12 @@ def main():
     ms_per_step = dt / args.steps * 1000.0
 
     flops = step_flops(cfg, seq, seq) * mbs * n_gpus  # per step whole job
+    if is_vl:
+        # vision tower contribution: 6 * P_vis * n_patches (dense-equivalent;
+        # window-attention quadratic terms are small at these grids)
+        p_vis = sum(p.numel() for p in model.visual.parameters())
+        n_patches = int(batch["pixel_values"].shape[0])
+        flops += 6.0 * p_vis * n_patches * n_gpus
     mfu = flops * args.steps / dt / (n_gpus * 2.5e15)
 
     # roofline: dominant hand-written kernel
@@ -258,7 +282,7 @@ def main():
                     "kernel": "vh_ce_fwd_bf16"}
 
     cpu_baseline = None
-    if rank == 0 and n_gpus == 1 and not args.no_cpu_baseline:
+    if rank == 0 and n_gpus == 1 and not args.no_cpu_baseline and not is_vl:
         log("timing CPU baseline (host cores, bounded sample)")
         try:
             cpu_baseline = run_cpu_baseline(preset, seq)
