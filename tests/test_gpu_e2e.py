@@ -158,3 +158,27 @@ def test_rccl_loopback_fsdp2_smoke():
         bind_ops("eager")
         if created:
             dist.destroy_process_group()
+
+
+def test_vlm_hip_vs_eager_gpu():
+    """tiny-vl image+text step: HIP op stack vs eager on GPU (the vision
+    tower's own SDPA path + the text stack's HIP kernels)."""
+    import torch
+
+    from veomni_amd.data import synthetic_vlm_batch
+    from veomni_amd.models import VL_PRESETS, _init_vl_presets, build_vl_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.ops import HIP_OPS_CONFIG
+
+    _init_vl_presets()
+    torch.manual_seed(0)
+    batch = synthetic_vlm_batch(VL_PRESETS["tiny-vl"], 128, seed=5, device="cuda")
+    batch["image_grid_thw"] = batch["image_grid_thw"]
+    bind_ops("eager")
+    model = build_vl_model("tiny-vl", dtype=torch.bfloat16, device="cuda")
+    l_eager, g_eager = _loss_and_gradnorm(model, batch)
+    bind_ops(HIP_OPS_CONFIG)
+    l_hip, g_hip = _loss_and_gradnorm(model, batch)
+    bind_ops("eager")
+    assert abs(l_hip - l_eager) < 2e-2 * max(abs(l_eager), 1.0), (l_hip, l_eager)
+    assert abs(g_hip - g_eager) < 5e-2 * max(g_eager, 1.0), (g_hip, g_eager)
