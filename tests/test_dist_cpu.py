@@ -713,3 +713,56 @@ def test_checkpoint_ep_reshard(tmp_path_factory):
     d1 = tempfile.mkdtemp(prefix="vh_ckpt_re1_")
     d2 = tempfile.mkdtemp(prefix="vh_ckpt_re2_")
     spawn(_ckpt_ep_reshard, d1, d2)
+
+
+def _hsdp_ep_moe_equivalence(rank, ws):
+    """HSDP + EP composed: world 4 = dp_replicate 2 x
+    dp_shard_sp 2, ep 2 — expert mesh (ep_replicate 2, ep_fsdp 1, ep 2),
+    dense mesh (2, 2). Loss and grad norm must match a single-process
+    reference on a replicated batch (ref parallel_state.py:598-627,
+    torch_parallelize.py:345-384)."""
+    from veomni_amd.distributed.fsdp2 import build_parallelize_model
+    from veomni_amd.distributed.parallel_state import (init_parallel_state,
+                                                       set_parallel_state)
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+    from veomni_amd.data import synthetic_batch
+
+    ps_ep = init_parallel_state(dp_replicate_size=2, ep_size=2)
+    assert ps_ep.ep_fsdp_size == 1 and ps_ep.dp_replicate_enabled
+    set_parallel_state(None)
+    ps_plain = init_parallel_state()
+    set_parallel_state(ps_ep)
+    bind_ops("eager")
+    model = build_model("tiny-moe")
+    ref = build_model("tiny-moe")
+
+    model = build_parallelize_model(model, param_dtype=torch.float32,
+                                    reduce_dtype=torch.float32)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-2, betas=(0.9, 0.95))
+    ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2, betas=(0.9, 0.95))
+
+    batch = synthetic_batch(512, 64, seed=13)
+    for step in range(2):
+        loss, _ = model(**batch)
+        loss.backward()
+        gn = model.clip_grad_norm_(1e9)
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+
+        set_parallel_state(ps_plain)
+        rloss, _ = ref(**batch)
+        rloss.backward()
+        rgn = torch.nn.utils.get_total_norm(
+            [p.grad for p in ref.parameters() if p.grad is not None])
+        ropt.step()
+        ropt.zero_grad(set_to_none=True)
+        set_parallel_state(ps_ep)
+
+        torch.testing.assert_close(loss.detach().float(), rloss.detach().float(),
+                                   rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(gn.float(), rgn.float(), rtol=1e-3, atol=1e-4)
+
+
+def test_hsdp_ep_moe_equivalence():
+    spawn(_hsdp_ep_moe_equivalence, world_size=4)

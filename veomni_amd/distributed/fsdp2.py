@@ -159,7 +159,9 @@ def clip_grad_norm(model: nn.Module, max_norm: float, norm_type: float = 2.0,
     if ep:
         ep_sum = _local_pth_sum(ep, norm_type)
         if ps.ep_fsdp_size > 1:
-            dist.all_reduce(ep_sum, group=ps.ep_fsdp_mesh.get_group())
+            # distinct shards only: under HSDP the replicas hold identical
+            # grads, so the reduce runs over the 1-D ep_fsdp group
+            dist.all_reduce(ep_sum, group=ps.ep_shard_group)
         dist.all_reduce(ep_sum, group=ps.ep_group)
         total = total + ep_sum
 

@@ -86,13 +86,26 @@ class ParallelState:
 
     @property
     def ep_fsdp_mesh(self) -> Optional[DeviceMesh]:
+        """Mesh FSDP wraps expert modules over: 2-D (ep_replicate, ep_fsdp)
+        under HSDP (HSDP-style fully_shard), 1-D ep_fsdp otherwise."""
         if self.ep_device_mesh is None:
             return None
+        if self.dp_replicate_enabled:
+            return self.ep_device_mesh["ep_replicate", "ep_fsdp"]
         return self.ep_device_mesh["ep_fsdp"]
 
     @property
+    def ep_shard_group(self):
+        """The 1-D ep_fsdp group (distinct expert shards once — the
+        grad-norm reduce group; replicas hold identical grads)."""
+        if self.ep_device_mesh is None:
+            return None
+        return self.ep_device_mesh["ep_fsdp"].get_group()
+
+    @property
     def ep_fsdp_size(self) -> int:
-        return self.fsdp_size // self.ep_size
+        # distinct-shard extent (excludes HSDP replication)
+        return self.dp_size * self.ulysses_size // self.ep_size
 
     # ---------------------------------------------------------------- groups
     @property
@@ -190,8 +203,6 @@ def init_parallel_state(
         dp_replicate_size, dp_size, ulysses_size, world_size)
     dp_shard_sp = dp_size * ulysses_size
     assert dp_shard_sp % ep_size == 0, f"ep_size {ep_size} must divide dp_shard*sp {dp_shard_sp}"
-    assert dp_replicate_size == 1 or ep_size == 1, \
-        "HSDP + EP needs a replicate-aware expert mesh (round-2 scope)"
 
     if dp_replicate_size > 1:
         # HSDP: outer replicate dim; FSDP consumes the 2-D
@@ -210,9 +221,16 @@ def init_parallel_state(
 
     ep_mesh = None
     if ep_size > 1:
-        ep_mesh = init_device_mesh(
-            device_type, (dp_shard_sp // ep_size, ep_size), mesh_dim_names=("ep_fsdp", "ep")
-        )
+        if dp_replicate_size > 1:
+            # HSDP + EP (ref :605-627): experts replicate with the dense
+            # outer dim -> mesh (ep_replicate, ep_fsdp, ep)
+            ep_mesh = init_device_mesh(
+                device_type, (dp_replicate_size, dp_shard_sp // ep_size, ep_size),
+                mesh_dim_names=("ep_replicate", "ep_fsdp", "ep"))
+        else:
+            ep_mesh = init_device_mesh(
+                device_type, (dp_shard_sp // ep_size, ep_size), mesh_dim_names=("ep_fsdp", "ep")
+            )
 
     _PARALLEL_STATE = ParallelState(
         world_size=world_size,
