@@ -228,10 +228,13 @@ def reduce_sequence_parallel_loss(loss: Tensor, num_valid_tokens: Tensor, group=
 
 
 # ---------------------------------------------------------------- async path
-# Async Ulysses (parity target: async_ulysses.py:48-212 fwd): the three QKV
-# all-to-alls are launched per-tensor with async_op=True so the exchange of q
-# overlaps the k/v projections (and all three overlap on the comm stream).
-# Backward stays the synchronous mirrored a2a this round.
+# Async Ulysses (parity target: async_ulysses.py:48-419): the three QKV
+# all-to-alls are launched per-tensor with async_op=True so the exchange of
+# q overlaps the k/v projections (and all three overlap on the comm
+# stream). BACKWARD MIRRORS the overlap: _A2AWait.backward STARTS the
+# reverse exchange async and _A2AStartSeqHeads.backward finishes it — in
+# autograd's reverse order the three reverse a2a all start back-to-back and
+# fly while the v/k projection weight-grad GEMMs execute.
 _ASYNC_WORK: dict = {}
 
 
@@ -261,12 +264,22 @@ class _A2AStartSeqHeads(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output: Tensor):
+        work = _ASYNC_WORK.pop(id(grad_output), None)
+        ws = dist.get_world_size(ctx.group)
+        if work is not None:
+            # mirrored async: _A2AWait.backward started the reverse exchange;
+            # finish it and apply the post-concat (scatter seq, gather heads)
+            work.wait()
+            out = torch.cat(grad_output.split(grad_output.size(0) // ws),
+                            dim=ctx.head_dim)
+            return (None, out, None, None)
         return (None, all_to_all_tensor(grad_output, ctx.seq_dim, ctx.head_dim, ctx.group), None, None)
 
 
 class _A2AWait(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: Tensor) -> Tensor:
+    def forward(ctx, x: Tensor, group) -> Tensor:
+        ctx.group = group
         work = _ASYNC_WORK.pop(id(x), None)
         if work is not None:
             work.wait()
@@ -274,7 +287,15 @@ class _A2AWait(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output: Tensor):
-        return grad_output
+        if ctx.group is not None:
+            # start the reverse exchange NOW; the paired _A2AStartSeqHeads
+            # backward (several autograd nodes later) finishes it
+            x = grad_output.contiguous()
+            out = torch.empty_like(x)
+            work = dist.all_to_all_single(out, x, group=ctx.group, async_op=True)
+            _ASYNC_WORK[id(out)] = work
+            return out, None
+        return grad_output, None
 
 
 def gather_seq_scatter_heads_async(x: Tensor, group=None) -> Tensor:
@@ -289,4 +310,4 @@ def wait_gathered(x: Tensor, group=None) -> Tensor:
     group = get_ulysses_sequence_parallel_group() if group is None else group
     if not group:
         return x
-    return _A2AWait.apply(x)
+    return _A2AWait.apply(x, group)

@@ -205,6 +205,15 @@ class Attention(nn.Module):
 
     def forward(self, hidden_states, position_embeddings, **kwargs):
         input_shape = hidden_states.shape[:-1]
+        ps = get_parallel_state()
+        if ps.ulysses_enabled and ps.async_ulysses:
+            # deepened async Ulysses (ref async_ulysses.py:48-419): the qkv
+            # GEMM is split back into per-tensor launches on the comm path
+            # so each all-to-all overlaps the next projection
+            out = self._async_ulysses_attention(hidden_states,
+                                                position_embeddings, **kwargs)
+            out = out.reshape(*input_shape, -1).contiguous()
+            return self.o_proj(out)
         # one fused qkv GEMM (weights concatenated at call time, so the
         # per-projection parameters/state-dict stay reference-shaped); the
         # per-element math is identical to three separate linears.
@@ -224,14 +233,7 @@ class Attention(nn.Module):
         q, k, v = q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2)
         cos, sin = position_embeddings
         q, k = apply_rotary_pos_emb(q, k, cos, sin)
-        ps = get_parallel_state()
-        if ps.ulysses_enabled and ps.async_ulysses:
-            # async Ulysses (ref async_ulysses.py:48-212): per-tensor async
-            # a2a launched as each projection finishes; RoPE commutes with
-            # the seq gather (elementwise per position), so it runs on the
-            # local slice. KV heads repeat first when sp > kv heads.
-            out = self._async_ulysses_attention(q, k, v, **kwargs)
-        elif veomni_attention.use_non_eager_impl:
+        if veomni_attention.use_non_eager_impl:
             # the bound attention kernel is SP-aware (does the Ulysses
             # exchange itself, ref attention/flash.py:236-299); varlen
             # doc bounds ride the kwargs (ref flash.py:61-91)
@@ -273,7 +275,18 @@ class Attention(nn.Module):
         out = gather_heads_scatter_seq(out.squeeze(0), head_dim=1, seq_dim=0, group=group)
         return out[None]
 
-    def _async_ulysses_attention(self, q, k, v, **kwargs):
+    def _async_ulysses_attention(self, hidden_states, position_embeddings,
+                                 **kwargs):
+        """Async Ulysses comm path (ref async_ulysses.py:48-419): per-tensor
+        projection -> qk-norm -> local-slice RoPE -> ASYNC seq->head a2a, in
+        q/k/v order, so q's exchange flies during the k projection GEMM and
+        k's during v's. The three waits then land just before core
+        attention. Backward mirrors the overlap: _A2AWait.backward starts the
+        reverse exchanges (all three back-to-back in autograd order) and
+        _A2AStartSeqHeads.backward finishes each, so the v/k projection
+        weight-grad GEMMs run under q/k's reverse a2a. RoPE commutes with the
+        seq gather (elementwise per position), so it runs on the local slice;
+        KV heads repeat before the a2a when sp > kv heads."""
         from ..distributed.sequence_parallel import (
             gather_heads_scatter_seq,
             gather_seq_scatter_heads_async,
@@ -283,24 +296,49 @@ class Attention(nn.Module):
         ps = get_parallel_state()
         sp = ps.ulysses_size
         group = ps.ulysses_group
-        # [B, h, S_loc, D] -> [S_loc, h, D] (B == 1 packed path)
-        assert q.shape[0] == 1, "async ulysses expects packed batch (B == 1)"
-        kv = k.shape[1]
-        if sp > kv:
-            rep = sp // kv
+        assert hidden_states.shape[0] == 1, "async ulysses expects packed batch (B == 1)"
+        S_loc = hidden_states.shape[1]
+        nh, nkv = self.num_heads, self.num_key_value_heads
+        rep = sp // nkv if sp > nkv else 1
+        cos, sin = position_embeddings
+        cos_u, sin_u = cos.unsqueeze(1), sin.unsqueeze(1)
+
+        q = self.q_proj(hidden_states).view(1, S_loc, nh, self.head_dim)
+        if self.q_norm is not None:
+            q = self.q_norm(q)
+        q = q.transpose(1, 2)
+        q = (q * cos_u) + (rotate_half(q) * sin_u)
+        qb = gather_seq_scatter_heads_async(
+            q.squeeze(0).transpose(0, 1).contiguous(), group=group)
+
+        k = self.k_proj(hidden_states).view(1, S_loc, nkv, self.head_dim)
+        if self.k_norm is not None:
+            k = self.k_norm(k)
+        k = k.transpose(1, 2)
+        k = (k * cos_u) + (rotate_half(k) * sin_u)
+        if rep > 1:
             k = torch.repeat_interleave(k, dim=1, repeats=rep)
+        kb = gather_seq_scatter_heads_async(
+            k.squeeze(0).transpose(0, 1).contiguous(), group=group)
+
+        v = self.v_proj(hidden_states).view(1, S_loc, nkv, self.head_dim).transpose(1, 2)
+        if rep > 1:
             v = torch.repeat_interleave(v, dim=1, repeats=rep)
-        qs = q.squeeze(0).transpose(0, 1)  # [S_loc, h, D]
-        ks = k.squeeze(0).transpose(0, 1)
-        vs = v.squeeze(0).transpose(0, 1)
-        qb = gather_seq_scatter_heads_async(qs.contiguous(), group=group)
-        kb = gather_seq_scatter_heads_async(ks.contiguous(), group=group)
-        vb = gather_seq_scatter_heads_async(vs.contiguous(), group=group)
+        vb = gather_seq_scatter_heads_async(
+            v.squeeze(0).transpose(0, 1).contiguous(), group=group)
+
         qg = wait_gathered(qb, group=group).transpose(0, 1)[None]  # [1, h/sp, S, D]
         kg = wait_gathered(kb, group=group).transpose(0, 1)[None]
         vg = wait_gathered(vb, group=group).transpose(0, 1)[None]
-        out, _ = sdpa_attention(self, qg, kg, vg, None, dropout=0.0,
-                                scaling=self.scaling, **kwargs)  # [1, S, h/sp, D]
+        if veomni_attention.use_non_eager_impl:
+            # gathered-sequence core through the bound HIP pair (exchange
+            # already done here, hence skip_ulysses)
+            out, _ = veomni_attention(self, qg, kg, vg, None, dropout=0.0,
+                                      scaling=self.scaling, skip_ulysses=True,
+                                      **kwargs)  # [1, S, h/sp, D]
+        else:
+            out, _ = sdpa_attention(self, qg, kg, vg, None, dropout=0.0,
+                                    scaling=self.scaling, **kwargs)
         out = out.squeeze(0)
         out = gather_heads_scatter_seq(out, head_dim=1, seq_dim=0, group=group)
         return out[None]
