@@ -103,6 +103,10 @@ def main():
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--batch", type=int, default=0, help="micro-batch per rank (0 = auto)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--sp", type=int, default=1,
+                    help="Ulysses sequence-parallel degree (divides --gpus)")
+    ap.add_argument("--sync-ulysses", action="store_true",
+                    help="use the synchronous a2a path (A/B for the async overlap)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -138,11 +142,17 @@ def main():
     # checkpointing (236 GiB peak incl. bf16 params/grads/optimizer states —
     # larger expert GEMM groups lift the grouped-GEMM rate ~35%).
     mbs = args.batch or (1 if is_vl else (8 if cfg.is_moe else 4))
+    if args.sp > 1:
+        mbs = 1  # SP rides the packed B==1 path
     if is_vl and args.seq_len == 4096 and args.model:
         # BASELINE config 4 default: image-text seq 8192
         args.seq_len = 8192
-    ep_size = n_gpus if (cfg.is_moe and n_gpus > 1) else 1
-    init_parallel_state(ep_size=ep_size, device_type="cuda")
+    sp = max(args.sp, 1)
+    assert n_gpus % sp == 0, "--sp must divide --gpus"
+    ep_size = n_gpus if (cfg.is_moe and n_gpus > 1 and sp == 1) else 1
+    init_parallel_state(ep_size=ep_size, ulysses_size=sp,
+                        async_ulysses=sp > 1 and not args.sync_ulysses,
+                        device_type="cuda")
     bind_ops(HIP_OPS_CONFIG)
 
     log(f"building {preset} on cuda:{local_rank} (ep={ep_size}, ws={world})")
@@ -196,7 +206,14 @@ def main():
         batch = synthetic_vlm_batch(vl_cfg, seq, batch=mbs, seed=42 + rank,
                                     device="cuda")
     else:
-        batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs, seed=42 + rank, device="cuda")
+        # sp-group peers must hold the SAME sequence (each rank keeps a
+        # 1/sp slice), so seed by dp-rank
+        batch = synthetic_batch(cfg.vocab_size, seq, batch=mbs,
+                                seed=42 + rank // sp, device="cuda")
+        if sp > 1:
+            from veomni_amd.data import sp_collate
+            assert mbs == 1, "SP bench path is the packed B==1 path"
+            batch = sp_collate(batch)
 
     n_valid = (batch["labels"] != -100).sum()
 
@@ -240,11 +257,11 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dt = float(t)
 
-    tokens_per_step = seq * mbs * n_gpus
+    tokens_per_step = seq * mbs * n_gpus // sp
     toks_per_s = tokens_per_step * args.steps / dt
     ms_per_step = dt / args.steps * 1000.0
 
-    flops = step_flops(cfg, seq, seq) * mbs * n_gpus  # per step whole job
+    flops = step_flops(cfg, seq, seq) * mbs * n_gpus // sp  # per step whole job
     if is_vl:
         # vision tower contribution: 6 * P_vis * n_patches (dense-equivalent;
         # window-attention quadratic terms are small at these grids)
@@ -312,7 +329,9 @@ def main():
                 "model": preset,
                 "global_batch": mbs * n_gpus,
                 "seq_len": seq,
-                "parallelism": f"dp{world}" + (f"_ep{ep_size}" if ep_size > 1 else ""),
+                "parallelism": f"dp{world // sp}"
+                + (f"_sp{sp}" + ("sync" if args.sync_ulysses else "") if sp > 1 else "")
+                + (f"_ep{ep_size}" if ep_size > 1 else ""),
                 # the aux load-balancing loss IS part of every timed step
                 # (reference Qwen3-MoE default; ref load_balancing_loss/eager.py)
                 "router_aux_loss_coef": cfg.router_aux_loss_coef,
