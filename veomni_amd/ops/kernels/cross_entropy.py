@@ -22,6 +22,8 @@ from .. import hip_lib
 from ..kernel_registry import KERNEL_REGISTRY, HardwareRequirement, KernelSpec
 
 IGNORE_INDEX = -100
+# stash-vs-stream switch for the dlogits buffer (test-patchable)
+STASH_LIMIT_BYTES = 16 << 30
 
 
 class HipChunkCE(torch.autograd.Function):
@@ -35,20 +37,42 @@ class HipChunkCE(torch.autograd.Function):
         inv = 1.0 / max(int(num_items), 1)
 
         total = torch.zeros((), dtype=torch.float32, device=hidden_states.device)
-        # stash per-chunk dlogits (bf16, [T, V] — 288 GB HBM affords it) so
-        # dgrad/wgrad run as single full-T GEMMs with hipBLASLt's internal
-        # fp32 accumulation; a per-chunk fp32 `grad_w +=` RMW measured
-        # ~1 ms/chunk of pure elementwise traffic on the 8B vocab grad.
-        dlog_all = flat_h.new_empty((flat_h.shape[0], weight.shape[0]))
-        for s in range(0, flat_h.shape[0], chunk_size):
-            e = min(s + chunk_size, flat_h.shape[0])
-            h = flat_h[s:e]
-            logits = torch.matmul(h, weight.t())          # bf16 (hipBLASLt)
-            loss_rows, _ = hip_lib.ce_fwd(logits, flat_l[s:e], inv, IGNORE_INDEX,
-                                          dlogits_out=dlog_all[s:e])
-            total += loss_rows.sum() * inv
-        grad_h = torch.matmul(dlog_all, weight)
-        grad_w = torch.matmul(dlog_all.t(), flat_h)
+        # Fast path: stash per-chunk dlogits (bf16, [T, V]) so dgrad/wgrad run
+        # as single full-T GEMMs with hipBLASLt's internal fp32 accumulation;
+        # a per-chunk fp32 `grad_w +=` RMW measured ~1 ms/chunk of pure
+        # elementwise traffic on the 8B vocab grad. Above 16 GiB of stash
+        # (seq-16384 dynamic batches, config 5) switch to the reference
+        # chunk_loss's no-[T,V] property: per-chunk dgrad/wgrad with an fp32
+        # grad_w accumulator, peak extra memory = one [chunk, V] dlogits +
+        # the fp32 [V, H] accumulator.
+        T_flat, V = flat_h.shape[0], weight.shape[0]
+        lowmem = T_flat * V * flat_h.element_size() > STASH_LIMIT_BYTES
+        if lowmem:
+            grad_h = torch.empty_like(flat_h)
+            grad_w32 = torch.zeros(weight.shape, dtype=torch.float32,
+                                   device=weight.device)
+            for s in range(0, T_flat, chunk_size):
+                e = min(s + chunk_size, T_flat)
+                h = flat_h[s:e]
+                logits = torch.matmul(h, weight.t())
+                loss_rows, dlog = hip_lib.ce_fwd(logits, flat_l[s:e], inv,
+                                                 IGNORE_INDEX)
+                total += loss_rows.sum() * inv
+                torch.matmul(dlog, weight, out=grad_h[s:e])
+                grad_w32 += torch.matmul(dlog.t(), h).float()
+            grad_w = grad_w32.to(weight.dtype)
+        else:
+            dlog_all = flat_h.new_empty((T_flat, V))
+            for s in range(0, T_flat, chunk_size):
+                e = min(s + chunk_size, T_flat)
+                h = flat_h[s:e]
+                logits = torch.matmul(h, weight.t())      # bf16 (hipBLASLt)
+                loss_rows, _ = hip_lib.ce_fwd(logits, flat_l[s:e], inv,
+                                              IGNORE_INDEX,
+                                              dlogits_out=dlog_all[s:e])
+                total += loss_rows.sum() * inv
+            grad_h = torch.matmul(dlog_all, weight)
+            grad_w = torch.matmul(dlog_all.t(), flat_h)
         ctx.save_for_backward(grad_h, grad_w)
         ctx.hshape = hidden_states.shape
         return total
