@@ -766,3 +766,47 @@ def _hsdp_ep_moe_equivalence(rank, ws):
 
 def test_hsdp_ep_moe_equivalence():
     spawn(_hsdp_ep_moe_equivalence, world_size=4)
+
+
+def _vl_moe_ep_async_sp(rank, ws):
+    """BASELINE config-5 composition at ws 2: VL + MoE text, EP 2 borrowing
+    the SP ranks, async Ulysses SP 2. Loss must match a single-process
+    no-SP run on the full batch (ViT replicated, ids sp-sliced, pixel
+    values full on every rank)."""
+    from veomni_amd.data import sp_collate, synthetic_vlm_batch
+    from veomni_amd.distributed.fsdp2 import build_parallelize_model
+    from veomni_amd.distributed.parallel_state import (init_parallel_state,
+                                                       set_parallel_state)
+    from veomni_amd.models import VL_PRESETS, _init_vl_presets, build_vl_model
+    from veomni_amd.models.modeling import bind_ops
+
+    bind_ops("eager")
+    _init_vl_presets()
+    torch.manual_seed(0)
+
+    ps_sp = init_parallel_state(ulysses_size=ws, ep_size=ws,
+                                async_ulysses=True)
+    set_parallel_state(None)
+    ps_plain = init_parallel_state()
+
+    set_parallel_state(ps_plain)
+    full = synthetic_vlm_batch(VL_PRESETS["tiny-vl-moe"], 128, batch=1, seed=5)
+    ref = build_vl_model("tiny-vl-moe", dtype=torch.float32)
+    rloss, raux = ref(**full)
+    (rloss + 0.001 * raux).backward()
+
+    set_parallel_state(ps_sp)
+    model = build_vl_model("tiny-vl-moe", dtype=torch.float32)
+    model = build_parallelize_model(model, param_dtype=torch.float32,
+                                    reduce_dtype=torch.float32)
+    batch = sp_collate(full)
+    loss, aux = model(**batch)
+    (loss + 0.001 * aux).backward()
+
+    torch.testing.assert_close(loss.detach().float(), rloss.detach().float(),
+                               rtol=5e-3, atol=5e-4)
+    assert float(aux) > 0
+
+
+def test_vl_moe_ep_async_sp():
+    spawn(_vl_moe_ep_async_sp)

@@ -94,3 +94,28 @@ def test_vlm_train_step_runs(gold, model):
         if p.grad is not None:
             assert torch.isfinite(p.grad).all(), n
     model.zero_grad(set_to_none=True)
+
+
+def test_vl_moe_train_step_runs():
+    """BASELINE config 5 model shape (VL + MoE text): fwd+bwd, aux loss on,
+    grads finite (tiny-vl-moe)."""
+    import torch as t
+
+    from veomni_amd.data import synthetic_vlm_batch
+    from veomni_amd.distributed.parallel_state import (init_parallel_state,
+                                                       set_parallel_state)
+    from veomni_amd.models import VL_PRESETS, _init_vl_presets, build_vl_model
+    from veomni_amd.models.modeling import bind_ops
+
+    set_parallel_state(None)
+    init_parallel_state(device_type="cpu")
+    bind_ops("eager")
+    _init_vl_presets()
+    m = build_vl_model("tiny-vl-moe", dtype=t.float32)
+    b = synthetic_vlm_batch(VL_PRESETS["tiny-vl-moe"], 128, batch=2, seed=1)
+    loss, aux = m(**b)
+    assert aux is not None and float(aux) > 0
+    (loss + 0.001 * aux).backward()
+    for n, p_ in m.named_parameters():
+        if p_.grad is not None:
+            assert t.isfinite(p_.grad).all(), n

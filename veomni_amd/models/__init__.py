@@ -83,6 +83,43 @@ def _init_vl_presets():
         image_token_id=151655,
         name="qwen25-vl-7b",
     )
+    # BASELINE config 5: Qwen3-VL-MoE-30B-A3B — the 30B MoE text stack
+    # (ref qwen3_vl_moe family: same decoder as qwen3_moe + mrope) under the
+    # windowed ViT. Vision tower kept at the Qwen2.5-VL shape (the ref's
+    # Qwen3-VL ViT differs in deepstack taps; the compute shape is the same
+    # class) — documented deviation, DESIGN.md VLM section.
+    VL_PRESETS["qwen3-vl-moe-30b"] = VLConfig(
+        text=ModelConfig(
+            name="qwen3-vl-moe-30b-text", vocab_size=151936, hidden_size=2048,
+            intermediate_size=6144, num_hidden_layers=48,
+            num_attention_heads=32, num_key_value_heads=4, head_dim=128,
+            rope_theta=1000000.0, qk_norm=True, num_experts=128,
+            num_experts_per_tok=8, moe_intermediate_size=768,
+            norm_topk_prob=True, router_aux_loss_coef=0.001,
+            initializer_range=0.02, mrope_section=(24, 20, 20),
+        ),
+        vision=VisionConfig(out_hidden_size=2048),
+        image_token_id=151655,
+        name="qwen3-vl-moe-30b",
+    )
+    VL_PRESETS["tiny-vl-moe"] = VLConfig(
+        text=ModelConfig(
+            name="tiny-vl-moe-text", vocab_size=512, hidden_size=128,
+            intermediate_size=256, num_hidden_layers=2,
+            num_attention_heads=4, num_key_value_heads=2, head_dim=32,
+            qk_norm=True, mrope_section=(4, 6, 6), num_experts=8,
+            num_experts_per_tok=2, moe_intermediate_size=64,
+            norm_topk_prob=True, router_aux_loss_coef=0.001,
+        ),
+        vision=VisionConfig(
+            depth=2, hidden_size=64, num_heads=4, intermediate_size=128,
+            out_hidden_size=128, patch_size=2, temporal_patch_size=1,
+            in_channels=3, spatial_merge_size=2, window_size=8,
+            fullatt_block_indexes=(1,),
+        ),
+        image_token_id=511,
+        name="tiny-vl-moe",
+    )
     VL_PRESETS["tiny-vl"] = VLConfig(
         text=ModelConfig(
             name="tiny-vl-text", vocab_size=512, hidden_size=128,
