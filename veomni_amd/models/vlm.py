@@ -113,9 +113,14 @@ def _apply_vision_rope(q, k, cos, sin):
 
 
 class VisionAttention(nn.Module):
-    """NON-causal varlen attention over [S, hidden] (ref :446-535): one SDPA
-    over the block-diagonal segment mask (full-attention blocks use the
-    per-image cu_seqlens, window blocks the window cu_seqlens)."""
+    """NON-causal varlen attention over [S, hidden] (ref :446-535):
+    block-diagonal segments (full-attention blocks use the per-image
+    cu_seqlens, window blocks the window cu_seqlens), executed as MASKLESS
+    batched SDPA over runs of equal-length segments — numerically identical
+    to the bool-masked single call, but it keeps torch on the flash/
+    mem-efficient backend (a bool attn_mask forces the math path, which
+    materializes [h, S, S] scores: 5+ GiB and an OOM at the seq-8192 mbs-4
+    VLM bench shape)."""
 
     def __init__(self, cfg: VisionConfig):
         super().__init__()
@@ -125,18 +130,28 @@ class VisionAttention(nn.Module):
         self.proj = nn.Linear(cfg.hidden_size, cfg.hidden_size)
         self.scaling = self.head_dim ** -0.5
 
-    def forward(self, x, seg_mask, cos, sin):
+    def forward(self, x, cu, cos, sin):
         S = x.shape[0]
         q, k, v = (self.qkv(x).reshape(S, 3, self.num_heads, -1)
                    .permute(1, 0, 2, 3).unbind(0))
         q, k = _apply_vision_rope(q, k, cos, sin)
-        q = q.transpose(0, 1).unsqueeze(0)   # [1, h, S, D]
-        k = k.transpose(0, 1).unsqueeze(0)
-        v = v.transpose(0, 1).unsqueeze(0)
-        out = F.scaled_dot_product_attention(q, k, v, attn_mask=seg_mask,
-                                             scale=self.scaling, is_causal=False)
-        out = out.squeeze(0).transpose(0, 1).reshape(S, -1)
-        return self.proj(out)
+        out = torch.empty_like(v)            # [S, h, D]
+        i = 0
+        n_seg = len(cu) - 1
+        while i < n_seg:
+            j = i
+            L = cu[i + 1] - cu[i]
+            while j < n_seg and cu[j + 1] - cu[j] == L:
+                j += 1
+            s0, s1, n = cu[i], cu[j], j - i
+            qs = q[s0:s1].reshape(n, L, self.num_heads, -1).transpose(1, 2)
+            ks = k[s0:s1].reshape(n, L, self.num_heads, -1).transpose(1, 2)
+            vs = v[s0:s1].reshape(n, L, self.num_heads, -1).transpose(1, 2)
+            o = F.scaled_dot_product_attention(qs, ks, vs, scale=self.scaling,
+                                               is_causal=False)
+            out[s0:s1] = o.transpose(1, 2).reshape(n * L, self.num_heads, -1)
+            i = j
+        return self.proj(out.reshape(S, -1))
 
 
 class VisionBlock(nn.Module):
@@ -147,8 +162,8 @@ class VisionBlock(nn.Module):
         self.attn = VisionAttention(cfg)
         self.mlp = VisionMLP(cfg)
 
-    def forward(self, x, seg_mask, cos, sin):
-        x = x + self.attn(self.norm1(x), seg_mask, cos, sin)
+    def forward(self, x, cu, cos, sin):
+        x = x + self.attn(self.norm1(x), cu, cos, sin)
         x = x + self.mlp(self.norm2(x))
         return x
 
@@ -215,13 +230,6 @@ def vision_window_index(grid_list: List, cfg: VisionConfig):
     return torch.cat(parts), cu, cu_win
 
 
-def _segment_mask(cu: List[int], device) -> torch.Tensor:
-    """Block-diagonal bool mask [S, S] from a cu_seqlens list."""
-    lens = torch.tensor([b - a for a, b in zip(cu[:-1], cu[1:])])
-    seg = torch.repeat_interleave(torch.arange(lens.numel()), lens).to(device)
-    return seg[None, :] == seg[:, None]
-
-
 class VisionTower(nn.Module):
     """Windowed ViT (ref :616-905). Forward takes the packed patch tensor
     and the host-side grid list; returns merged features [n_units, out_h]."""
@@ -253,11 +261,9 @@ class VisionTower(nn.Module):
         emb = torch.cat((freqs, freqs), dim=-1)
         cos, sin = emb.cos(), emb.sin()
 
-        full_mask = _segment_mask(cu, x.device)
-        win_mask = _segment_mask(cu_win, x.device)
         for i, blk in enumerate(self.blocks):
-            mask = full_mask if i in cfg.fullatt_block_indexes else win_mask
-            x = blk(x, mask, cos, sin)
+            seg = cu if i in cfg.fullatt_block_indexes else cu_win
+            x = blk(x, seg, cos, sin)
 
         merged = self.merger(x)                      # [S/unit, out_hidden]
         reverse = torch.argsort(window_index)
