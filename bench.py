@@ -141,7 +141,10 @@ def main():
     # llama-8b 4x4096 tokens without checkpointing; the 30B MoE 8x4096 with
     # checkpointing (236 GiB peak incl. bf16 params/grads/optimizer states —
     # larger expert GEMM groups lift the grouped-GEMM rate ~35%).
-    mbs = args.batch or (1 if is_vl else (8 if cfg.is_moe else 4))
+    # VL dense: mbs 4 measured 9,797 vs 7,826 tok/s at mbs 1 (155 GiB peak);
+    # VL-MoE keeps mbs 1 (30B text states + ViT activations at seq 8192)
+    mbs = args.batch or ((1 if cfg.is_moe else 4) if is_vl
+                         else (8 if cfg.is_moe else 4))
     if args.sp > 1:
         mbs = 1  # SP rides the packed B==1 path
     if is_vl and args.seq_len == 4096 and args.model:
