@@ -1276,8 +1276,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
 //     via vh_attn_bwd2_dkv6probe_bf16.
 //   - DOC: packed-varlen block-diagonal causal via doc_start/doc_end
 //     (per-token document bounds; see k_attn_fwd).
+//   - DB (probe g2): register double-buffer of the Q^T/dO^T staging. PMC on
+//     v6 (profiles/r02_dkv_pmc.txt): SQ_WAIT_ANY = 66% of wave cycles —
+//     waves parked at the per-tile barrier pair draining the staging
+//     GLOBAL loads. DB issues the NEXT tile's staging loads right after
+//     the stage barrier (they drain under the MFMA chain) so only the
+//     ds_writes + lgkm sit between barriers. 32 VGPRs (4x bf16x8 x 2
+//     tensors), paid for by PREF=0 (the dispatched depth).
 // LDS 64 KB: K/V strips 2x16 K + Q^T/dO^T tiles 2x16 K -> 2 blocks/CU.
-template <bool DOC, int PREF>
+template <bool DOC, int PREF, int DBN = 0, bool DBLATE = false>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1348,7 +1355,26 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
   // a nested g/qt loop form kept per-head pointer sets alive across the
   // whole accumulator section and spilled ~190 extra bytes/lane
   const int ntiles = qtn - qt0;
-  for (int it = 0; it < rep * ntiles; ++it) {
+  const int itn = rep * ntiles;
+  bf16x8 svq[DBN > 0 ? DBN : 1], svd[DBN > 0 ? DBN : 1];  // DB staging regs
+#define VH_DKV_LDNEXT(IT2)                                                     \
+  {                                                                            \
+    const int g2_ = (IT2) / ntiles;                                            \
+    const bf16_t* Qb2_ = Q + (((int64_t)b * Hq + hkv * rep + g2_) * S) * DH;   \
+    const bf16_t* dOb2_ = dO + (((int64_t)b * Hq + hkv * rep + g2_) * S) * DH; \
+    const int64_t q0t2_ = (int64_t)(qt0 + ((IT2) - g2_ * ntiles)) * 64;        \
+    _Pragma("unroll") for (int u = 0; u < DBN; ++u) {                          \
+      int unit_ = tid + u * 256;                                               \
+      int qq_ = unit_ & 63;                                                    \
+      int d0_ = (unit_ >> 6) * 8;                                              \
+      svq[u] = *reinterpret_cast<const bf16x8*>(Qb2_ + (q0t2_ + qq_) * DH + d0_); \
+      svd[u] = *reinterpret_cast<const bf16x8*>(dOb2_ + (q0t2_ + qq_) * DH + d0_); \
+    }                                                                          \
+  }
+  if constexpr (DBN > 0) {
+    if (itn > 0) VH_DKV_LDNEXT(0);
+  }
+  for (int it = 0; it < itn; ++it) {
     const int g = it / ntiles;
     const int qt = qt0 + (it - g * ntiles);
     {
@@ -1375,8 +1401,14 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         int unit = tid + u * 256;
         int q = unit & 63;
         int d0 = (unit >> 6) * 8;
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
+        bf16x8 vq, vd;
+        if (DBN > 0 && u < DBN) {
+          vq = svq[u < DBN ? u : 0];
+          vd = svd[u < DBN ? u : 0];
+        } else {
+          vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
+          vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int row = d0 + j;
@@ -1385,6 +1417,10 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         }
       }
       __syncthreads();
+      if constexpr (DBN > 0 && !DBLATE) {
+        // next tile's staging loads drain under the MFMA chain below
+        if (it + 1 < itn) VH_DKV_LDNEXT(it + 1);
+      }
 
       const bool live = ((q0 + 31) >= (kv0 + kvslice * 32)) &&
                         (!DOC || q0 < de_wave);
@@ -1477,9 +1513,14 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           }
         }
       }
+      if constexpr (DBN > 0 && DBLATE) {
+        // late placement: loads drain under the final barrier + next store
+        if (it + 1 < itn) VH_DKV_LDNEXT(it + 1);
+      }
       __syncthreads();
     }
   }
+#undef VH_DKV_LDNEXT
 
   // combine the two q-subtile partials per kv slice through LDS, then store.
   // red layout: [slice 2][lane 64][64 floats] = 32 KB (reuses K/V space —
@@ -1775,10 +1816,24 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  if (pref == 0) VH_DKV6(0);
+#define VH_DKVDB(N_, L_)                                                      \
+  hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, N_, L_>), grid, dim3(256),   \
+                     65536, s, reinterpret_cast<const bf16_t*>(Q),            \
+                     reinterpret_cast<const bf16_t*>(K),                      \
+                     reinterpret_cast<const bf16_t*>(V),                      \
+                     reinterpret_cast<const bf16_t*>(dO), delta, lse2,        \
+                     reinterpret_cast<bf16_t*>(dK),                           \
+                     reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
+                     Hkv, S, scale)
+  if (pref == 16) VH_DKVDB(2, false);
+  else if (pref == 17) VH_DKVDB(4, false);
+  else if (pref == 18) VH_DKVDB(4, true);
+  else if (pref == 19) VH_DKVDB(2, true);
+  else if (pref == 0) VH_DKV6(0);
   else if (pref == 8) VH_DKV6(8);
   else VH_DKV6(4);
 #undef VH_DKV6
+#undef VH_DKVDB
   VH_HIP(hipGetLastError());
   return 0;
 }
