@@ -19,8 +19,9 @@ extern "C" __global__ void k_trprobe(short* out, int mode) {
     case 2: addr = (tid & 15) * 256 + (tid >> 4) * 8; break; // 16 q rows x 4 lanes
     default: addr = (tid >> 4) * 256 + (tid & 15) * 8; break; // 4 lanes/q ordered
   }
-  s16x4 r = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-      (__attribute__((address_space(3))) void*)(reinterpret_cast<char*>(s) + addr));
+  typedef __attribute__((address_space(3))) s16x4 as3_s16x4;
+  auto* base = (__attribute__((address_space(3))) char*)s;
+  s16x4 r = __builtin_amdgcn_ds_read_tr16_b64_v4i16((as3_s16x4*)(base + addr));
   for (int j = 0; j < 4; ++j) out[tid * 4 + j] = r[j];
 }
 
