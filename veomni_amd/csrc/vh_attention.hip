@@ -1284,7 +1284,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
 //     ds_writes + lgkm sit between barriers. 32 VGPRs (4x bf16x8 x 2
 //     tensors), paid for by PREF=0 (the dispatched depth).
 // LDS 64 KB: K/V strips 2x16 K + Q^T/dO^T tiles 2x16 K -> 2 blocks/CU.
-template <bool DOC, int PREF, int DBN = 0, bool DBLATE = false>
+//   - TR (probe g3): natural-layout Q/dO staging image + CDNA4
+//     ds_read_b64_tr_b16 hardware transpose reads for the dv/dk B-frags
+//     (guide T10). Replaces the 64 scalar b16 transpose stores per thread
+//     per tile with 8 b128 stores into a bank-permuted natural image
+//     (off(q,d) = q*256 + g*8 + (d&3)*2, g = (c&7)|(((q&3)^(c>>3))&3)<<3,
+//     c = d>>2: reads 32 distinct 8-B slots/phase, stores 16 distinct
+//     16-B slots/phase -> conflict-free both ways; mapping derived from
+//     the on-box dump tests/gpu_tr16_probe.hip -> gpurun_out/tr16map.txt).
+template <bool DOC, int PREF, int DBN = 0, bool DBLATE = false, bool TR = false>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1399,8 +1407,15 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         int unit = tid + u * 256;
-        int q = unit & 63;
-        int d0 = (unit >> 6) * 8;
+        int q, d0;
+        if constexpr (TR) {
+          // phase-conflict-free assignment for the permuted natural image
+          q = ((unit >> 2) & 15) | (((unit >> 6) & 3) << 4);
+          d0 = (unit & 3) * 8 + ((unit >> 8) & 3) * 32;
+        } else {
+          q = unit & 63;
+          d0 = (unit >> 6) * 8;
+        }
         bf16x8 vq, vd;
         if (DBN > 0 && u < DBN) {
           vq = svq[u < DBN ? u : 0];
@@ -1409,11 +1424,20 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
           vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
         }
+        if constexpr (TR) {
+          int c = d0 >> 2;
+          int g = (c & 7) | ((((q & 3) ^ (c >> 3)) & 3) << 3);
+          *reinterpret_cast<bf16x8*>(
+              reinterpret_cast<char*>(qtr) + q * 256 + g * 8) = vq;
+          *reinterpret_cast<bf16x8*>(
+              reinterpret_cast<char*>(dotr) + q * 256 + g * 8) = vd;
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int row = d0 + j;
-          qtr[(row * 128 + vswz(row, q * 2)) >> 1] = vq.v[j];
-          dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
+          for (int j = 0; j < 8; ++j) {
+            int row = d0 + j;
+            qtr[(row * 128 + vswz(row, q * 2)) >> 1] = vq.v[j];
+            dotr[(row * 128 + vswz(row, q * 2)) >> 1] = vd.v[j];
+          }
         }
       }
       __syncthreads();
@@ -1502,12 +1526,40 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
         for (int mch = 0; mch < 2; ++mch) {
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
-            int trow = dblk * 32 + col;
-            int colb = (qsub * 32 + mch * 16 + half * 8) * 2;
-            bf16frag dof = *reinterpret_cast<const bf16frag*>(
-                reinterpret_cast<const char*>(dotr) + trow * 128 + vswz(trow, colb));
-            bf16frag qf = *reinterpret_cast<const bf16frag*>(
-                reinterpret_cast<const char*>(qtr) + trow * 128 + vswz(trow, colb));
+            bf16frag dof, qf;
+            if constexpr (TR) {
+              // two 4-q hardware-transpose reads per frag; lane m of each
+              // 16-lane group supplies (q = qb+(m>>2), chunk (m&3)) and
+              // receives (q = qb+0..3, d = db + lane%16)
+              const int m_ = lane & 15;
+              const int colhi_ = (lane >> 4) & 1;
+              const int qb0 = qsub * 32 + mch * 16 + half * 8 + (m_ >> 2);
+              const int c_r = dblk * 8 + colhi_ * 4 + (m_ & 3);
+              const int g_r =
+                  (c_r & 7) | ((((m_ >> 2) ^ (c_r >> 3)) & 3) << 3);
+              const int a0 = qb0 * 256 + g_r * 8;
+              auto* dob3 = (__attribute__((address_space(3))) char*)dotr;
+              auto* qb3 = (__attribute__((address_space(3))) char*)qtr;
+              typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+              typedef __attribute__((address_space(3))) bf16x4t as3b4;
+              bf16x4t d0v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(dob3 + a0));
+              bf16x4t d1v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(dob3 + a0 + 1024));
+              bf16x4t q0v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(qb3 + a0));
+              bf16x4t q1v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(qb3 + a0 + 1024));
+              dof = __builtin_shufflevector(d0v, d1v, 0, 1, 2, 3, 4, 5, 6, 7);
+              qf = __builtin_shufflevector(q0v, q1v, 0, 1, 2, 3, 4, 5, 6, 7);
+            } else {
+              int trow = dblk * 32 + col;
+              int colb = (qsub * 32 + mch * 16 + half * 8) * 2;
+              dof = *reinterpret_cast<const bf16frag*>(
+                  reinterpret_cast<const char*>(dotr) + trow * 128 + vswz(trow, colb));
+              qf = *reinterpret_cast<const bf16frag*>(
+                  reinterpret_cast<const char*>(qtr) + trow * 128 + vswz(trow, colb));
+            }
             dv_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa2[mch], dof, dv_acc[dblk], 0, 0, 0);
             dk_acc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da2[mch], qf, dk_acc[dblk], 0, 0, 0);
           }
@@ -1825,7 +1877,16 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  if (pref == 16) VH_DKVDB(2, false);
+  if (pref == 20)
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, true>), grid,
+                       dim3(256), 65536, s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dK),
+                       reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
+                       Hkv, S, scale);
+  else if (pref == 16) VH_DKVDB(2, false);
   else if (pref == 17) VH_DKVDB(4, false);
   else if (pref == 18) VH_DKVDB(4, true);
   else if (pref == 19) VH_DKVDB(2, true);
