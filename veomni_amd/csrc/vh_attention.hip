@@ -63,7 +63,10 @@ __device__ __forceinline__ float xor32h(float v, int half) {
 // start indices (doc_start[t] = cu_seqlens[i] for t in document i; B == 1).
 // Replaces the reference's flash-attn varlen cu_seqlens path
 // (ops/kernels/attention/flash.py:61-91, kwargs from data_collator.py:50).
-template <int MODE, bool DOC>
+// TRF: natural-layout V image + tr16 hardware-transpose PV reads (same
+// scheme as the dkv TR variant; image off(kv,d) = kv*256 + g*8 + (d&3)*2,
+// g = (c&7)|(((kv&3)^(c>>3))&3)<<3, c = d>>2).
+template <int MODE, bool DOC, bool TRF = false>
 __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
@@ -128,8 +131,12 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     ksrc[i] = Kb + (int64_t)row * DH + (kswz(row, colb) >> 1);
     klds[i] = base >> 1;
   }
-  const int v_kv = tid & 63;
-  const int v_d0 = (tid >> 6) * 8;   // 2 units: rows v_d0 and v_d0+64
+  // TRF store assignment keeps every 16-lane phase on 16 distinct 16-B
+  // slots of the permuted image (conflict-free)
+  const int v_kv = TRF ? (((tid >> 2) & 15) | (((tid >> 6) & 3) << 4))
+                       : (tid & 63);
+  const int v_d0 = TRF ? ((tid & 3) * 8 + ((tid >> 8) & 1) * 32)
+                       : ((tid >> 6) * 8);  // 2 units: rows v_d0 and v_d0+64
   const bf16_t* vsrc = Vb + (int64_t)v_kv * DH + v_d0;
 
   f32x16 oacc[4];
@@ -147,11 +154,21 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
       glds16a(ksrc[i] + (int64_t)t0 * KB * DH, kt(0) + klds[i]);
     bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH);
     bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH + 64);
+    if constexpr (TRF) {
+      int c0 = v_d0 >> 2, c1 = (v_d0 + 64) >> 2;
+      int g0 = (c0 & 7) | ((((v_kv & 3) ^ (c0 >> 3)) & 3) << 3);
+      int g1 = (c1 & 7) | ((((v_kv & 3) ^ (c1 >> 3)) & 3) << 3);
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(vt(0)) + v_kv * 256 + g0 * 8) = v0;
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(vt(0)) + v_kv * 256 + g1 * 8) = v1;
+    } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int r0 = v_d0 + j, r1 = v_d0 + 64 + j;
-      vt(0)[(r0 * 128 + vswz(r0, v_kv * 2)) >> 1] = v0.v[j];
-      vt(0)[(r1 * 128 + vswz(r1, v_kv * 2)) >> 1] = v1.v[j];
+      for (int j = 0; j < 8; ++j) {
+        int r0 = v_d0 + j, r1 = v_d0 + 64 + j;
+        vt(0)[(r0 * 128 + vswz(r0, v_kv * 2)) >> 1] = v0.v[j];
+        vt(0)[(r1 * 128 + vswz(r1, v_kv * 2)) >> 1] = v1.v[j];
+      }
     }
   }
   __syncthreads();
@@ -231,6 +248,8 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
         for (int mch = 0; mch < 2; ++mch)
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
+            // probe path: non-TR image only (probe launcher instantiates
+            // TRF=false)
             int vrow = d * 32 + col;
             int colb2 = (sub * 32 + mch * 16 + half * 8) * 2;
             bf16frag vf = *reinterpret_cast<const bf16frag*>(
@@ -298,10 +317,28 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
         for (int mch = 0; mch < 2; ++mch) {
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
-            int vrow = d * 32 + col;
-            int colb = (sub * 32 + mch * 16 + half * 8) * 2;
-            bf16frag vf = *reinterpret_cast<const bf16frag*>(
-                reinterpret_cast<const char*>(vtc) + vrow * 128 + vswz(vrow, colb));
+            bf16frag vf;
+            if constexpr (TRF) {
+              const int m_ = lane & 15;
+              const int colhi_ = (lane >> 4) & 1;
+              const int kvb0 = sub * 32 + mch * 16 + half * 8 + (m_ >> 2);
+              const int c_r = d * 8 + colhi_ * 4 + (m_ & 3);
+              const int g_r =
+                  (c_r & 7) | ((((m_ >> 2) ^ (c_r >> 3)) & 3) << 3);
+              auto* vb3 = (__attribute__((address_space(3))) char*)vtc;
+              typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+              typedef __attribute__((address_space(3))) bf16x4t as3b4;
+              bf16x4t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(vb3 + kvb0 * 256 + g_r * 8));
+              bf16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(vb3 + kvb0 * 256 + g_r * 8 + 1024));
+              vf = __builtin_shufflevector(r0, r1, 0, 1, 2, 3, 4, 5, 6, 7);
+            } else {
+              int vrow = d * 32 + col;
+              int colb = (sub * 32 + mch * 16 + half * 8) * 2;
+              vf = *reinterpret_cast<const bf16frag*>(
+                  reinterpret_cast<const char*>(vtc) + vrow * 128 + vswz(vrow, colb));
+            }
             oacc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[mch], vf, oacc[d], 0, 0, 0);
           }
         }
@@ -312,11 +349,21 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 
     // ---- T14 write-late: flush the next V tile, then the tile barrier
     if (more) {
+      if constexpr (TRF) {
+        int c0 = v_d0 >> 2, c1 = (v_d0 + 64) >> 2;
+        int g0 = (c0 & 7) | ((((v_kv & 3) ^ (c0 >> 3)) & 3) << 3);
+        int g1 = (c1 & 7) | ((((v_kv & 3) ^ (c1 >> 3)) & 3) << 3);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(vt(cur ^ 1)) + v_kv * 256 + g0 * 8) = vn0;
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(vt(cur ^ 1)) + v_kv * 256 + g1 * 8) = vn1;
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int r0 = v_d0 + j, r1 = v_d0 + 64 + j;
-        vt(cur ^ 1)[(r0 * 128 + vswz(r0, v_kv * 2)) >> 1] = vn0.v[j];
-        vt(cur ^ 1)[(r1 * 128 + vswz(r1, v_kv * 2)) >> 1] = vn1.v[j];
+        for (int j = 0; j < 8; ++j) {
+          int r0 = v_d0 + j, r1 = v_d0 + 64 + j;
+          vt(cur ^ 1)[(r0 * 128 + vswz(r0, v_kv * 2)) >> 1] = vn0.v[j];
+          vt(cur ^ 1)[(r1 * 128 + vswz(r1, v_kv * 2)) >> 1] = vn1.v[j];
+        }
       }
     }
     __syncthreads();
@@ -383,6 +430,13 @@ extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
   if (mode == 1) VH_AM(1);
   else if (mode == 2) VH_AM(2);
   else if (mode == 3) VH_AM(3);
+  else if (mode == 20)  // TRF: tr16 V image A/B
+    hipLaunchKernelGGL((k_attn_fwd<0, false, true>), grid, dim3(512), 65536,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<bf16_t*>(O), LSE, nullptr, B, Hq, Hkv,
+                       S, scale);
   else VH_AM(0);
 #undef VH_AM
   VH_HIP(hipGetLastError());
