@@ -53,8 +53,28 @@ def prep(B, Hq, Hkv, S):
     return q, k, v, do, delta, lse2, scale
 
 
+def bind_dq(lib):
+    fn = lib.vh_attn_bwd2_dqprobe_bf16
+    fn.restype = ctypes.c_int
+    fn.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 3 + [
+        ctypes.c_int64, ctypes.c_float, ctypes.c_int, ctypes.c_void_p]
+    return fn
+
+
+def run_dq(fn, q, k, v, do, delta, lse2, scale, mode):
+    B, Hq, S = q.shape[0], q.shape[1], q.shape[2]
+    dq = torch.zeros(B, Hq, S, 128, dtype=torch.bfloat16, device="cuda")
+    rc = fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), do.data_ptr(),
+            delta.data_ptr(), lse2.data_ptr(), dq.data_ptr(),
+            B, Hq, k.shape[1], S, scale, mode, L.cur_stream())
+    assert rc == 0
+    torch.cuda.synchronize()
+    return dq
+
+
 def main():
     fn = bind(L.get_lib())
+    fnq = bind_dq(L.get_lib())
 
     # parity: TR vs the dispatched v6 (itself fp32-ref-verified) at two shapes
     for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
@@ -66,6 +86,14 @@ def main():
         print(f"shape {shp}: TR-vs-v6 |dK|={ek:.4g} |dV|={ev:.4g}", flush=True)
         assert ek == 0.0 and ev == 0.0, "TR must be bit-identical to v6"
 
+    for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
+        q, k, v, do, delta, lse2, scale = prep(*shp)
+        dq0 = run_dq(fnq, q, k, v, do, delta, lse2, scale, 0)
+        dq1 = run_dq(fnq, q, k, v, do, delta, lse2, scale, 20)
+        eq = (dq1.float() - dq0.float()).abs().max().item()
+        print(f"shape {shp}: dq TRQ-vs-v0 |dQ|={eq:.4g}", flush=True)
+        assert eq == 0.0
+
     # timing at the microbench shape
     q, k, v, do, delta, lse2, scale = prep(1, 32, 8, 8192)
     for pref, name in ((0, "v6"), (20, "TR"), (0, "v6b"), (20, "TRb")):
@@ -76,6 +104,17 @@ def main():
         t0.record()
         for _ in range(10):
             run(fn, q, k, v, do, delta, lse2, scale, pref)
+        t1.record()
+        torch.cuda.synchronize()
+        print(f"{name}: {t0.elapsed_time(t1) / 10 * 1000:.0f} us", flush=True)
+    for mode, name in ((0, "dq-v0"), (20, "dq-TRQ"), (0, "dq-v0b"), (20, "dq-TRQb")):
+        for _ in range(3):
+            run_dq(fnq, q, k, v, do, delta, lse2, scale, mode)
+        t0 = torch.cuda.Event(enable_timing=True)
+        t1 = torch.cuda.Event(enable_timing=True)
+        t0.record()
+        for _ in range(10):
+            run_dq(fnq, q, k, v, do, delta, lse2, scale, mode)
         t1.record()
         torch.cuda.synchronize()
         print(f"{name}: {t0.elapsed_time(t1) / 10 * 1000:.0f} us", flush=True)

@@ -1663,7 +1663,9 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 #undef VH_DKVG_COMBINE
 }
 
-template <bool DOC>
+// TRQ: tr16 hardware-transpose K^T reads over the bank-permuted natural
+// [32 kv][128 d] image (same scheme as dkv TR; see that comment).
+template <bool DOC, bool TRQ = false>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1739,13 +1741,26 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
       int unit = tid + u * 256;
-      int kv = unit & 31;
-      int d0 = (unit >> 5) * 8;
+      int kv, d0;
+      if constexpr (TRQ) {
+        kv = ((unit >> 2) & 7) | (((unit >> 5) & 3) << 3);
+        d0 = (unit & 3) * 8 + ((unit >> 7) & 3) * 32;
+      } else {
+        kv = unit & 31;
+        d0 = (unit >> 5) * 8;
+      }
       bf16x8 v = *reinterpret_cast<const bf16x8*>(Kb + (kvt0 + kv) * DH + d0);
+      if constexpr (TRQ) {
+        int c = d0 >> 2;
+        int g = (c & 7) | ((((kv & 3) ^ (c >> 3)) & 3) << 3);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(ktr) + kv * 256 + g * 8) = v;
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int row = d0 + j;
-        ktr[(row * 64 + qswz(row, kv * 2)) >> 1] = v.v[j];
+        for (int j = 0; j < 8; ++j) {
+          int row = d0 + j;
+          ktr[(row * 64 + qswz(row, kv * 2)) >> 1] = v.v[j];
+        }
       }
     }
     __syncthreads();
@@ -1799,10 +1814,27 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
       for (int dblk = 0; dblk < 4; ++dblk) {
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
-          int trow = dblk * 32 + col;
-          int colb = (mch * 16 + half * 8) * 2;
-          bf16frag ktf = *reinterpret_cast<const bf16frag*>(
-              reinterpret_cast<const char*>(ktr) + trow * 64 + qswz(trow, colb));
+          bf16frag ktf;
+          if constexpr (TRQ) {
+            const int m_ = lane & 15;
+            const int colhi_ = (lane >> 4) & 1;
+            const int kvb0 = mch * 16 + half * 8 + (m_ >> 2);
+            const int c_r = dblk * 8 + colhi_ * 4 + (m_ & 3);
+            const int g_r = (c_r & 7) | ((((m_ >> 2) ^ (c_r >> 3)) & 3) << 3);
+            auto* kb3 = (__attribute__((address_space(3))) char*)ktr;
+            typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+            typedef __attribute__((address_space(3))) bf16x4t as3b4;
+            bf16x4t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (as3b4*)(kb3 + kvb0 * 256 + g_r * 8));
+            bf16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (as3b4*)(kb3 + kvb0 * 256 + g_r * 8 + 1024));
+            ktf = __builtin_shufflevector(r0, r1, 0, 1, 2, 3, 4, 5, 6, 7);
+          } else {
+            int trow = dblk * 32 + col;
+            int colb = (mch * 16 + half * 8) * 2;
+            ktf = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(ktr) + trow * 64 + qswz(trow, colb));
+          }
           dq4[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1[mch], ktf, dq4[dblk], 0, 0, 0);
         }
       }
@@ -1953,6 +1985,36 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
   else VH_DKV6(4);
 #undef VH_DKV6
 #undef VH_DKVDB
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* probe: dq with the TRQ (tr16) K^T image vs the dispatched layout. */
+extern "C" int vh_attn_bwd2_dqprobe_bf16(const uint16_t* Q, const uint16_t* K,
+                                         const uint16_t* V, const uint16_t* dO,
+                                         const float* delta, const float* lse2,
+                                         uint16_t* dQ, int B, int Hq, int Hkv,
+                                         int64_t S, float scale, int mode,
+                                         void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(S % 128 == 0, "S %% 128 != 0");
+  dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
+  if (mode == 20)
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, true>), grid, dim3(256), 24576,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
+                       scale);
+  else
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, false>), grid, dim3(256), 24576,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
+                       scale);
   VH_HIP(hipGetLastError());
   return 0;
 }
