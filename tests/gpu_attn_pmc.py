@@ -23,6 +23,7 @@ def main():
     do = torch.randn_like(q)
     o, lse = L.attn_fwd(q, k, v, scale)
     for _ in range(6):
+        o, lse = L.attn_fwd(q, k, v, scale)
         L.attn_bwd(q, k, v, o, lse, do, scale)
     torch.cuda.synchronize()
     print("pmc run done")
