@@ -105,6 +105,8 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--sp", type=int, default=1,
                     help="Ulysses sequence-parallel degree (divides --gpus)")
+    ap.add_argument("--checkpoint", choices=["auto", "on", "off"],
+                    default="auto", help="activation checkpointing override")
     ap.add_argument("--sync-ulysses", action="store_true",
                     help="use the synchronous a2a path (A/B for the async overlap)")
     args = ap.parse_args()
@@ -169,7 +171,8 @@ def main():
     # 288 GB HBM3E: dense llama-8b at N=1 holds full activations comfortably
     # (no forward recompute). The 30B MoE keeps checkpointing (scattered
     # expert activations are ~1 GB/layer/rank).
-    model.use_checkpoint = cfg.is_moe or is_vl
+    model.use_checkpoint = (cfg.is_moe or is_vl) if args.checkpoint == "auto" \
+        else args.checkpoint == "on"
     model = build_parallelize_model(model)
     def make_opt(kind):
         if kind == "ve":
