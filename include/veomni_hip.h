@@ -87,6 +87,12 @@ int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
                               int64_t N, int64_t K, int64_t total_rows,
                               void* stream);
 
+/* probe: nk256s with the 32x32x16 MFMA inner loop (issue-stall attack). */
+int vh_group_gemm_nk256s32_bf16(const uint16_t* A, const uint16_t* B,
+                                uint16_t* C, const int64_t* cumsum, int G,
+                                int64_t N, int64_t K, int64_t total_rows,
+                                void* stream);
+
 /* Probe kernels kept for on-box A/B (trans_b semantics, XCD schedule):
  * nk8s = KSUB=32 counted-vmcnt 4-deep ring; nkp = 8-phase K-split pipeline
  * reconstruction. Both race-screened correct; neither dispatched (measured

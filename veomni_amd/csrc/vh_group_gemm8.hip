@@ -897,6 +897,142 @@ __global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256s(
 }
 
 
+// nk256s32: the nk256s persistent/XCD-clustered wrapper with the inner loop
+// on v_mfma_f32_32x32x16_bf16 instead of 16x16x32. PMC on nk256s
+// (profiles/r02_gg_pmc_sq.csv) shows the kernel ISSUE-bound
+// (SQ_WAIT_INST_ANY 42%, MFMA pipe 47% busy, zero LDS conflicts); the
+// 32x32 shape halves MFMA instruction count per FLOP (and is 1024 vs 964
+// FLOP/cyc-SIMD dense), attacking the issue stall directly. Fragment
+// mappings: A/B frag = 8 contiguous k at (row0 + lane%32,
+// ks*16 + (lane>>5)*8); C element (m = (r&3)+8*(r>>2)+4*(lane>>5),
+// n = lane&31) — the attention kernels' 32x32 convention.
+typedef __attribute__((ext_vector_type(16))) float f32x16gg;
+
+__device__ __forceinline__ bf16frag frag_read64_32(const bf16_t* tile,
+                                                   int row0, int ks,
+                                                   int lane) {
+  int row = row0 + (lane & 31);
+  int colb = (ks * 16 + ((lane >> 5) << 3)) * 2;
+  int off_b = row * 128 + swz64(row, colb);
+  return *reinterpret_cast<const bf16frag*>(
+      reinterpret_cast<const char*>(tile) + off_b);
+}
+
+__global__ __launch_bounds__(THREADS8, 2) void k_group_gemm_nk256s32(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const int64_t* __restrict__ cumsum, int G,
+    int64_t N, int64_t K, int tiles_n, const int* __restrict__ ws) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* sm = reinterpret_cast<bf16_t*>(smem);
+  auto ta = [&](int buf) { return sm + buf * 32768; };          // 2 x 32 KiB
+  auto tb = [&](int buf) { return sm + 16384 + buf * 32768; };  // 2 x 32 KiB
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+
+  const int total_tiles = ws[0];
+  const int X = blockIdx.x & 7;
+  const int slot = blockIdx.x >> 3;
+  const int L = (total_tiles + 7) / 8;
+  const int t_end = min((X + 1) * L, total_tiles);
+  const int nk = (int)(K / BK64);
+
+  for (int t = X * L + slot; t < t_end; t += 32) {
+    int lo = 0, hi = G - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (ws[1 + mid] <= t) lo = mid;
+      else hi = mid - 1;
+    }
+    const int gid = lo;
+    const int local = t - ws[1 + gid];
+    const int tiles_m_g = (ws[2 + gid] - ws[1 + gid]) / tiles_n;
+    const int bm = local % tiles_m_g;
+    const int bn = local / tiles_m_g;
+
+    const int64_t row_start = (gid > 0) ? cumsum[gid - 1] : 0;
+    const int64_t m_size = cumsum[gid] - row_start;
+    const bf16_t* Ag = A + row_start * K;
+    const bf16_t* Bg = B + (int64_t)gid * N * K;
+    bf16_t* Cg = C + row_start * N;
+
+    KStage64 sa, sb;
+    sa.init(Ag, K, [&](int r) -> int64_t {
+      int64_t gm = (int64_t)bm * BM8 + r;
+      return gm % m_size;
+    }, tid);
+    sb.init(Bg, K, [&](int r) -> int64_t {
+      int64_t gn = (int64_t)bn * BN8 + r;
+      return gn % N;
+    }, tid);
+
+    f32x16gg acc[4][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[i][j] = f32x16gg{};
+
+    sa.stage(ta(0), 0);
+    sb.stage(tb(0), 0);
+    __syncthreads();
+    int cur = 0;
+    for (int kt = 0; kt < nk; ++kt) {
+      if (kt + 1 < nk) {
+        sa.stage(ta(cur ^ 1), (int64_t)(kt + 1) * BK64);
+        sb.stage(tb(cur ^ 1), (int64_t)(kt + 1) * BK64);
+      }
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16frag af[2], bfr[2];
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          bfr[j] = frag_read64_32(tb(cur), wc * 64 + j * 32, ks, lane);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[i] = frag_read64_32(ta(cur), wr * 128 + i * 32, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af[i], bfr[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[i] = frag_read64_32(ta(cur), wr * 128 + 64 + i * 32, ks, lane);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i + 2][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af[i], bfr[j], acc[i + 2][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+
+    const int col32 = lane & 31;
+    const int half32 = lane >> 5;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int mrow = (r & 3) + 8 * (r >> 2) + 4 * half32;
+          int64_t m = (int64_t)bm * BM8 + wr * 128 + i * 32 + mrow;
+          int64_t n = (int64_t)bn * BN8 + wc * 64 + j * 32 + col32;
+          if (m < m_size && n < N) Cg[m * N + n] = f2bf(acc[i][j][r]);
+        }
+  }
+}
+
+
 // nk8's 4-deep KSUB=32 counted-vmcnt ring under the same XCD-clustered
 // persistent schedule (trans_b path only — its transposed-write staging for
 // !trans_b has a 8-16-way LDS write conflict, and dgrad routes through the
@@ -1229,6 +1365,31 @@ extern "C" int vh_group_gemm_nk256s_bf16(const uint16_t* A, const uint16_t* B,
   VH_HIP(hipGetLastError());
   hipLaunchKernelGGL(k_group_gemm_nk256s, dim3(256), dim3(THREADS8), 131072, s,
                      reinterpret_cast<const bf16_t*>(A),
+                     reinterpret_cast<const bf16_t*>(B),
+                     reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  return 0;
+}
+
+/* probe: nk256s with the 32x32x16 MFMA inner loop. */
+extern "C" int vh_group_gemm_nk256s32_bf16(const uint16_t* A, const uint16_t* B,
+                                           uint16_t* C, const int64_t* cumsum,
+                                           int G, int64_t N, int64_t K,
+                                           int64_t total_rows, void* stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  VH_CHECK(K % BK64 == 0, "K %% 64 != 0");
+  VH_CHECK(N % 16 == 0, "N %% 16 != 0");
+  VH_CHECK(G <= VH_GG_SCHED_MAX_G, "G > %d", VH_GG_SCHED_MAX_G);
+  if (vh_gg_sched_ws == nullptr) {
+    VH_HIP(hipMalloc(&vh_gg_sched_ws, (VH_GG_SCHED_MAX_G + 2) * sizeof(int)));
+  }
+  int tiles_n = (int)((N + BN8 - 1) / BN8);
+  hipLaunchKernelGGL(k_gg_sched, dim3(1), dim3(64), 0, s, cumsum, G, tiles_n,
+                     vh_gg_sched_ws);
+  VH_HIP(hipGetLastError());
+  hipLaunchKernelGGL(k_group_gemm_nk256s32, dim3(256), dim3(THREADS8), 131072,
+                     s, reinterpret_cast<const bf16_t*>(A),
                      reinterpret_cast<const bf16_t*>(B),
                      reinterpret_cast<bf16_t*>(C), cumsum, G, N, K, tiles_n,
                      vh_gg_sched_ws);
