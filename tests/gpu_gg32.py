@@ -5,9 +5,15 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+import ctypes
+
 import torch
 
 from veomni_amd.ops import hip_lib as L
+
+GG_ARGS = [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_int64,
+                                   ctypes.c_int64, ctypes.c_int64,
+                                   ctypes.c_void_p]
 
 
 def gg(fn, a, w, cumsum, G, N, K):
@@ -21,6 +27,10 @@ def gg(fn, a, w, cumsum, G, N, K):
 
 def main():
     lib = L.get_lib()
+    for name in ("vh_group_gemm_nk256s_bf16", "vh_group_gemm_nk256s32_bf16"):
+        fn = getattr(lib, name)
+        fn.restype = ctypes.c_int
+        fn.argtypes = GG_ARGS
     torch.manual_seed(0)
     dev = "cuda"
     # parity at a small ragged shape
