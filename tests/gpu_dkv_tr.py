@@ -80,11 +80,12 @@ def main():
     for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
         q, k, v, do, delta, lse2, scale = prep(*shp)
         dk0, dv0 = run(fn, q, k, v, do, delta, lse2, scale, 0)
-        dk1, dv1 = run(fn, q, k, v, do, delta, lse2, scale, 20)
-        ek = (dk1.float() - dk0.float()).abs().max().item()
-        ev = (dv1.float() - dv0.float()).abs().max().item()
-        print(f"shape {shp}: TR-vs-v6 |dK|={ek:.4g} |dV|={ev:.4g}", flush=True)
-        assert ek == 0.0 and ev == 0.0, "TR must be bit-identical to v6"
+        for mode, nm in ((20, "TR"), (21, "TR2")):
+            dk1, dv1 = run(fn, q, k, v, do, delta, lse2, scale, mode)
+            ek = (dk1.float() - dk0.float()).abs().max().item()
+            ev = (dv1.float() - dv0.float()).abs().max().item()
+            print(f"shape {shp}: {nm}-vs-v6 |dK|={ek:.4g} |dV|={ev:.4g}", flush=True)
+            assert ek == 0.0 and ev == 0.0, f"{nm} must be bit-identical"
 
     for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
         q, k, v, do, delta, lse2, scale = prep(*shp)
@@ -96,7 +97,7 @@ def main():
 
     # timing at the microbench shape
     q, k, v, do, delta, lse2, scale = prep(1, 32, 8, 8192)
-    for pref, name in ((0, "v6"), (20, "TR"), (0, "v6b"), (20, "TRb")):
+    for pref, name in ((0, "v6"), (20, "TR"), (21, "TR2"), (20, "TRb"), (21, "TR2b")):
         for _ in range(3):
             run(fn, q, k, v, do, delta, lse2, scale, pref)
         t0 = torch.cuda.Event(enable_timing=True)

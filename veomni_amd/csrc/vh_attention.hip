@@ -1346,7 +1346,12 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv4(
 //     c = d>>2: reads 32 distinct 8-B slots/phase, stores 16 distinct
 //     16-B slots/phase -> conflict-free both ways; mapping derived from
 //     the on-box dump tests/gpu_tr16_probe.hip -> gpurun_out/tr16map.txt).
-template <bool DOC, int PREF, int DBN = 0, bool DBLATE = false, bool TR = false>
+//   - TR2 (probe g4): like TR but over the L16 latin-square image
+//     off(q,d) = q*256 + L16*16 + (d&7)*2, L16 = (q&15) ^ (((p&3)<<2)|(p>>2)),
+//     p = d>>3 — conflict-free for the b128 stores, the tr16 B-frag reads
+//     AND plain b128 A-frag reads, so the per-iteration Q/dO A-fragments
+//     come from LDS instead of re-reading global (halves Q/dO traffic).
+template <bool DOC, int PREF, int DBN = 0, bool DBLATE = false, int TRMODE = 0>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1462,11 +1467,13 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
       for (int u = 0; u < 4; ++u) {
         int unit = tid + u * 256;
         int q, d0;
-        if constexpr (TR) {
+        if constexpr (TRMODE == 1) {
           // phase-conflict-free assignment for the permuted natural image
           q = ((unit >> 2) & 15) | (((unit >> 6) & 3) << 4);
           d0 = (unit & 3) * 8 + ((unit >> 8) & 3) * 32;
         } else {
+          // TRMODE 2: the L16 image is conflict-free under the simple
+          // 64-consecutive-q assignment
           q = unit & 63;
           d0 = (unit >> 6) * 8;
         }
@@ -1478,13 +1485,20 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           vq = *reinterpret_cast<const bf16x8*>(Qb + (q0t + q) * DH + d0);
           vd = *reinterpret_cast<const bf16x8*>(dOb + (q0t + q) * DH + d0);
         }
-        if constexpr (TR) {
+        if constexpr (TRMODE == 1) {
           int c = d0 >> 2;
           int g = (c & 7) | ((((q & 3) ^ (c >> 3)) & 3) << 3);
           *reinterpret_cast<bf16x8*>(
               reinterpret_cast<char*>(qtr) + q * 256 + g * 8) = vq;
           *reinterpret_cast<bf16x8*>(
               reinterpret_cast<char*>(dotr) + q * 256 + g * 8) = vd;
+        } else if constexpr (TRMODE == 2) {
+          int pp = d0 >> 3;
+          int l16 = (q & 15) ^ (((pp & 3) << 2) | (pp >> 2));
+          *reinterpret_cast<bf16x8*>(
+              reinterpret_cast<char*>(qtr) + q * 256 + l16 * 16) = vq;
+          *reinterpret_cast<bf16x8*>(
+              reinterpret_cast<char*>(dotr) + q * 256 + l16 * 16) = vd;
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
@@ -1521,6 +1535,16 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           if (c < PREF) {
             qc = qrow[c < PREF ? c : 0];
             dc = dorow[c < PREF ? c : 0];
+          } else if constexpr (TRMODE == 2) {
+            // A-frags from the staged L16 image (conflict-free b128):
+            // chunk pair p = c*2 + half, row q0+col
+            const int pp = c * 2 + half;
+            const int rl = qsub * 32 + col;  // local image row
+            const int l16 = (rl & 15) ^ (((pp & 3) << 2) | (pp >> 2));
+            qc = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(qtr) + rl * 256 + l16 * 16);
+            dc = *reinterpret_cast<const bf16frag*>(
+                reinterpret_cast<const char*>(dotr) + rl * 256 + l16 * 16);
           } else {
             qc = *reinterpret_cast<const bf16frag*>(
                 Qb + (q0 + col) * DH + c * 16 + half * 8);
@@ -1581,7 +1605,31 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
             bf16frag dof, qf;
-            if constexpr (TR) {
+            if constexpr (TRMODE == 2) {
+              const int m_ = lane & 15;
+              const int colhi_ = (lane >> 4) & 1;
+              const int qr_ = qsub * 32 + mch * 16 + half * 8 + (m_ >> 2);
+              const int c_r = dblk * 8 + colhi_ * 4 + (m_ & 3);
+              const int perm_c = (((c_r >> 1) & 3) << 2) | (c_r >> 3);
+              const int l16a = (qr_ & 15) ^ perm_c;
+              const int l16b = ((qr_ + 4) & 15) ^ perm_c;  // q&15 changes at +4
+              const int a0 = qr_ * 256 + l16a * 16 + (c_r & 1) * 8;
+              const int a1 = (qr_ + 4) * 256 + l16b * 16 + (c_r & 1) * 8;
+              auto* dob3 = (__attribute__((address_space(3))) char*)dotr;
+              auto* qb3 = (__attribute__((address_space(3))) char*)qtr;
+              typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+              typedef __attribute__((address_space(3))) bf16x4t as3b4;
+              bf16x4t d0v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(dob3 + a0));
+              bf16x4t d1v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(dob3 + a1));
+              bf16x4t q0v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(qb3 + a0));
+              bf16x4t q1v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(qb3 + a1));
+              dof = __builtin_shufflevector(d0v, d1v, 0, 1, 2, 3, 4, 5, 6, 7);
+              qf = __builtin_shufflevector(q0v, q1v, 0, 1, 2, 3, 4, 5, 6, 7);
+            } else if constexpr (TRMODE == 1) {
               // two 4-q hardware-transpose reads per frag; lane m of each
               // 16-lane group supplies (q = qb+(m>>2), chunk (m&3)) and
               // receives (q = qb+0..3, d = db + lane%16)
@@ -1898,7 +1946,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0, 0, false, true>), grid_kv,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0, 0, false, 1>), grid_kv,
                        dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
@@ -1916,7 +1964,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, true>), grid_kv,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 1>), grid_kv,
                        dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
@@ -1967,8 +2015,17 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  if (pref == 20)
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, true>), grid,
+  if (pref == 21)
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 2>), grid,
+                       dim3(256), 65536, s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dK),
+                       reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
+                       Hkv, S, scale);
+  else if (pref == 20)
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 1>), grid,
                        dim3(256), 65536, s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
