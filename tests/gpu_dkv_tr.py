@@ -80,7 +80,7 @@ def main():
     for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
         q, k, v, do, delta, lse2, scale = prep(*shp)
         dk0, dv0 = run(fn, q, k, v, do, delta, lse2, scale, 0)
-        for mode, nm in ((20, "TR"), (21, "TR2")):
+        for mode, nm in ((20, "TR"), (21, "TR2"), (22, "TR2G")):
             dk1, dv1 = run(fn, q, k, v, do, delta, lse2, scale, mode)
             ek = (dk1.float() - dk0.float()).abs().max().item()
             ev = (dv1.float() - dv0.float()).abs().max().item()
@@ -97,7 +97,7 @@ def main():
 
     # timing at the microbench shape
     q, k, v, do, delta, lse2, scale = prep(1, 32, 8, 8192)
-    for pref, name in ((0, "v6"), (20, "TR"), (21, "TR2"), (20, "TRb"), (21, "TR2b")):
+    for pref, name in ((0, "v6"), (21, "TR2"), (22, "TR2G"), (21, "TR2b"), (22, "TR2Gb")):
         for _ in range(3):
             run(fn, q, k, v, do, delta, lse2, scale, pref)
         t0 = torch.cuda.Event(enable_timing=True)

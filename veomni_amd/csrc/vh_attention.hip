@@ -1463,6 +1463,20 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
             dOb + (q0 + col) * DH + c * 16 + half * 8);
       }
       // stage Q^T / dO^T [128][64] (1024 units of [1 q][8 d] / 256 thr = 4)
+      if constexpr (TRMODE == 3) {
+        // glds fill of the L16 image: window win = wave*4+u covers 4 q rows
+        // (1024 B, lane-linear dest); lane sources pair p = P(slot ^ (q&15))
+        // — zero VGPR staging, zero ds_writes
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int win = wave * 4 + u;
+          const int qrow = win * 4 + (lane >> 4);
+          const int pp0 = (lane & 15) ^ (qrow & 15);
+          const int pg = ((pp0 & 3) << 2) | (pp0 >> 2);
+          glds16a(Qb + (q0t + qrow) * DH + pg * 8, qtr + win * 512);
+          glds16a(dOb + (q0t + qrow) * DH + pg * 8, dotr + win * 512);
+        }
+      } else
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         int unit = tid + u * 256;
@@ -1535,7 +1549,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
           if (c < PREF) {
             qc = qrow[c < PREF ? c : 0];
             dc = dorow[c < PREF ? c : 0];
-          } else if constexpr (TRMODE == 2) {
+          } else if constexpr (TRMODE >= 2) {
             // A-frags from the staged L16 image (conflict-free b128):
             // chunk pair p = c*2 + half, row q0+col
             const int pp = c * 2 + half;
@@ -1605,7 +1619,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 #pragma unroll
           for (int dblk = 0; dblk < 4; ++dblk) {
             bf16frag dof, qf;
-            if constexpr (TRMODE == 2) {
+            if constexpr (TRMODE >= 2) {
               const int m_ = lane & 15;
               const int colhi_ = (lane >> 4) & 1;
               const int qr_ = qsub * 32 + mch * 16 + half * 8 + (m_ >> 2);
@@ -2016,7 +2030,16 @@ extern "C" int vh_attn_bwd2_dkv6probe_bf16(const uint16_t* Q, const uint16_t* K,
                      reinterpret_cast<bf16_t*>(dK),                           \
                      reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,  \
                      Hkv, S, scale)
-  if (pref == 21)
+  if (pref == 22)
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 3>), grid,
+                       dim3(256), 65536, s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dK),
+                       reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
+                       Hkv, S, scale);
+  else if (pref == 21)
     hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 2>), grid,
                        dim3(256), 65536, s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
