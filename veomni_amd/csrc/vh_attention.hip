@@ -1939,10 +1939,11 @@ extern "C" int vh_attn_bwd2_dkv4probe_bf16(const uint16_t* Q, const uint16_t* K,
 }
 
 /* Dispatched split backward: GQA-folded dkv (dK/dV [B,Hkv,S,D] written
- * once — no per-Q-head intermediates, no host group sum) with the TR2
- * (tr16 hardware-transpose, L16 latin-square image, A-frags from LDS)
- * staging — bit-identical to v6, measured -43% (tests/gpu_dkv_tr.py) —
- * + per-Q-head dq.
+ * once — no per-Q-head intermediates, no host group sum) with the TR2G
+ * staging (tr16 hardware-transpose reads over the L16 latin-square image,
+ * A-frags from LDS, image filled by glds direct-to-LDS with per-lane
+ * permuted sources) — bit-identical to v6, measured -47% cumulative
+ * (tests/gpu_dkv_tr.py) — + per-Q-head dq.
  * doc_start/doc_end (nullable, B == 1) select the packed-varlen
  * block-diagonal causal mask. */
 extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
@@ -1961,7 +1962,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
   dim3 grid_kv((uint32_t)(S / 64), (uint32_t)(B * Hkv));
   dim3 grid_q((uint32_t)(S / 128), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0, 0, false, 2>), grid_kv,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<true, 0, 0, false, 3>), grid_kv,
                        dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
@@ -1979,7 +1980,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), doc_start, B, Hq, Hkv,
                        S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 2>), grid_kv,
+    hipLaunchKernelGGL((k_attn_bwd_dkv_g<false, 0, 0, false, 3>), grid_kv,
                        dim3(256), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
