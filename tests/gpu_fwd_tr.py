@@ -38,17 +38,18 @@ def main():
         k = (torch.randn(B, Hkv, S, 128, device="cuda") * 0.5).to(torch.bfloat16)
         v = (torch.randn(B, Hkv, S, 128, device="cuda") * 0.5).to(torch.bfloat16)
         o0, l0 = run(q, k, v, 0)
-        o1, l1 = run(q, k, v, 20)
-        eo = (o1.float() - o0.float()).abs().max().item()
-        el = (l1 - l0).abs().max().item()
-        print(f"shape {shp}: TRF-vs-v0 |O|={eo:.4g} |LSE|={el:.4g}", flush=True)
-        assert eo == 0.0 and el == 0.0
+        for mode, nm in ((20, "TRF"), (21, "TRF2")):
+            o1, l1 = run(q, k, v, mode)
+            eo = (o1.float() - o0.float()).abs().max().item()
+            el = (l1 - l0).abs().max().item()
+            print(f"shape {shp}: {nm}-vs-v0 |O|={eo:.4g} |LSE|={el:.4g}", flush=True)
+            assert eo == 0.0 and el == 0.0
 
     torch.manual_seed(0)
     q = (torch.randn(1, 32, 8192, 128, device="cuda") * 0.5).to(torch.bfloat16)
     k = (torch.randn(1, 8, 8192, 128, device="cuda") * 0.5).to(torch.bfloat16)
     v = (torch.randn(1, 8, 8192, 128, device="cuda") * 0.5).to(torch.bfloat16)
-    for mode, name in ((0, "v0"), (20, "TRF"), (0, "v0b"), (20, "TRFb")):
+    for mode, name in ((0, "v0"), (21, "TRF2"), (0, "v0b"), (21, "TRF2b")):
         for _ in range(3):
             run(q, k, v, mode)
         t0 = torch.cuda.Event(enable_timing=True)

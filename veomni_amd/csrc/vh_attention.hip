@@ -66,7 +66,10 @@ __device__ __forceinline__ float xor32h(float v, int half) {
 // TRF: natural-layout V image + tr16 hardware-transpose PV reads (same
 // scheme as the dkv TR variant; image off(kv,d) = kv*256 + g*8 + (d&3)*2,
 // g = (c&7)|(((kv&3)^(c>>3))&3)<<3, c = d>>2).
-template <int MODE, bool DOC, bool TRF = false>
+// TRF 2: V image = L16 latin square filled by glds direct-to-LDS (per-lane
+// permuted sources), PV B-frags via tr16 — the dkv TR2G scheme on the
+// forward's V tile (drops the vn register carry + all V ds_writes).
+template <int MODE, bool DOC, int TRF = 0>
 __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
@@ -133,10 +136,10 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
   }
   // TRF store assignment keeps every 16-lane phase on 16 distinct 16-B
   // slots of the permuted image (conflict-free)
-  const int v_kv = TRF ? (((tid >> 2) & 15) | (((tid >> 6) & 3) << 4))
-                       : (tid & 63);
-  const int v_d0 = TRF ? ((tid & 3) * 8 + ((tid >> 8) & 1) * 32)
-                       : ((tid >> 6) * 8);  // 2 units: rows v_d0 and v_d0+64
+  const int v_kv = (TRF == 1) ? (((tid >> 2) & 15) | (((tid >> 6) & 3) << 4))
+                              : (tid & 63);
+  const int v_d0 = (TRF == 1) ? ((tid & 3) * 8 + ((tid >> 8) & 1) * 32)
+                              : ((tid >> 6) * 8);  // rows v_d0, v_d0+64
   const bf16_t* vsrc = Vb + (int64_t)v_kv * DH + v_d0;
 
   f32x16 oacc[4];
@@ -152,9 +155,19 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 #pragma unroll
     for (int i = 0; i < 2; ++i)
       glds16a(ksrc[i] + (int64_t)t0 * KB * DH, kt(0) + klds[i]);
+    if constexpr (TRF == 2) {
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int win = wave * 2 + u;
+        const int kvr = win * 4 + (lane >> 4);
+        const int pp0 = (lane & 15) ^ (kvr & 15);
+        const int pg = ((pp0 & 3) << 2) | (pp0 >> 2);
+        glds16a(Vb + ((int64_t)t0 * KB + kvr) * DH + pg * 8, vt(0) + win * 512);
+      }
+    } else {
     bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH);
     bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)t0 * KB * DH + 64);
-    if constexpr (TRF) {
+    if constexpr (TRF == 1) {
       int c0 = v_d0 >> 2, c1 = (v_d0 + 64) >> 2;
       int g0 = (c0 & 7) | ((((v_kv & 3) ^ (c0 >> 3)) & 3) << 3);
       int g1 = (c1 & 7) | ((((v_kv & 3) ^ (c1 >> 3)) & 3) << 3);
@@ -170,6 +183,7 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
         vt(0)[(r1 * 128 + vswz(r1, v_kv * 2)) >> 1] = v1.v[j];
       }
     }
+    }
   }
   __syncthreads();
 
@@ -182,8 +196,20 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 #pragma unroll
       for (int i = 0; i < 2; ++i)
         glds16a(ksrc[i] + (int64_t)(t + 1) * KB * DH, kt(cur ^ 1) + klds[i]);
-      vn0 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)(t + 1) * KB * DH);
-      vn1 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)(t + 1) * KB * DH + 64);
+      if constexpr (TRF == 2) {
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const int win = wave * 2 + u;
+          const int kvr = win * 4 + (lane >> 4);
+          const int pp0 = (lane & 15) ^ (kvr & 15);
+          const int pg = ((pp0 & 3) << 2) | (pp0 >> 2);
+          glds16a(Vb + ((int64_t)(t + 1) * KB + kvr) * DH + pg * 8,
+                  vt(cur ^ 1) + win * 512);
+        }
+      } else {
+        vn0 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)(t + 1) * KB * DH);
+        vn1 = *reinterpret_cast<const bf16x8*>(vsrc + (int64_t)(t + 1) * KB * DH + 64);
+      }
     }
 
     const bool diag = ((int64_t)(t + 1) * KB) > ((int64_t)qb * QB + wave * WQ);
@@ -318,7 +344,25 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
             bf16frag vf;
-            if constexpr (TRF) {
+            if constexpr (TRF == 2) {
+              const int m_ = lane & 15;
+              const int colhi_ = (lane >> 4) & 1;
+              const int kvb0 = sub * 32 + mch * 16 + half * 8 + (m_ >> 2);
+              const int c_r = d * 8 + colhi_ * 4 + (m_ & 3);
+              const int perm_ = (((c_r >> 1) & 3) << 2) | (c_r >> 3);
+              const int a0 =
+                  kvb0 * 256 + ((kvb0 & 15) ^ perm_) * 16 + (c_r & 1) * 8;
+              const int a1 = (kvb0 + 4) * 256 +
+                             (((kvb0 + 4) & 15) ^ perm_) * 16 + (c_r & 1) * 8;
+              auto* vb3 = (__attribute__((address_space(3))) char*)vtc;
+              typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+              typedef __attribute__((address_space(3))) bf16x4t as3b4;
+              bf16x4t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(vb3 + a0));
+              bf16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                  (as3b4*)(vb3 + a1));
+              vf = __builtin_shufflevector(r0, r1, 0, 1, 2, 3, 4, 5, 6, 7);
+            } else if constexpr (TRF == 1) {
               const int m_ = lane & 15;
               const int colhi_ = (lane >> 4) & 1;
               const int kvb0 = sub * 32 + mch * 16 + half * 8 + (m_ >> 2);
@@ -348,8 +392,8 @@ __global__ __launch_bounds__(512, 2) void k_attn_fwd(
     }
 
     // ---- T14 write-late: flush the next V tile, then the tile barrier
-    if (more) {
-      if constexpr (TRF) {
+    if (more && TRF != 2) {
+      if constexpr (TRF == 1) {
         int c0 = v_d0 >> 2, c1 = (v_d0 + 64) >> 2;
         int g0 = (c0 & 7) | ((((v_kv & 3) ^ (c0 >> 3)) & 3) << 3);
         int g1 = (c1 & 7) | ((((v_kv & 3) ^ (c1 >> 3)) & 3) << 3);
@@ -430,8 +474,15 @@ extern "C" int vh_attn_fwd_probe_bf16(const uint16_t* Q, const uint16_t* K,
   if (mode == 1) VH_AM(1);
   else if (mode == 2) VH_AM(2);
   else if (mode == 3) VH_AM(3);
+  else if (mode == 21)  // TRF2: glds-filled L16 V image A/B
+    hipLaunchKernelGGL((k_attn_fwd<0, false, 2>), grid, dim3(512), 65536,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<bf16_t*>(O), LSE, nullptr, B, Hq, Hkv,
+                       S, scale);
   else if (mode == 20)  // TRF: tr16 V image A/B
-    hipLaunchKernelGGL((k_attn_fwd<0, false, true>), grid, dim3(512), 65536,
+    hipLaunchKernelGGL((k_attn_fwd<0, false, 1>), grid, dim3(512), 65536,
                        s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1727,7 +1778,8 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 
 // TRQ: tr16 hardware-transpose K^T reads over the bank-permuted natural
 // [32 kv][128 d] image (same scheme as dkv TR; see that comment).
-template <bool DOC, bool TRQ = false>
+// TRQ 2: the L16 latin-square image filled by glds (dkv TR2G scheme).
+template <bool DOC, int TRQ = 0>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -1800,11 +1852,22 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
           reinterpret_cast<char*>(vrow) + row * 256 + kswz(row, colb)) =
           *reinterpret_cast<const bf16x8*>(Vb + (kvt0 + row) * DH + (colb >> 1));
     }
+    if constexpr (TRQ == 2) {
+      // glds fill of the L16 K^T image: 8 windows of 4 kv rows, 4 waves x 2
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int win = wave * 2 + u;
+        const int kvr = win * 4 + (lane >> 4);
+        const int pp0 = (lane & 15) ^ (kvr & 15);
+        const int pg = ((pp0 & 3) << 2) | (pp0 >> 2);
+        glds16a(Kb + (kvt0 + kvr) * DH + pg * 8, ktr + win * 512);
+      }
+    } else
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
       int unit = tid + u * 256;
       int kv, d0;
-      if constexpr (TRQ) {
+      if constexpr (TRQ == 1) {
         kv = ((unit >> 2) & 7) | (((unit >> 5) & 3) << 3);
         d0 = (unit & 3) * 8 + ((unit >> 7) & 3) * 32;
       } else {
@@ -1812,7 +1875,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
         d0 = (unit >> 5) * 8;
       }
       bf16x8 v = *reinterpret_cast<const bf16x8*>(Kb + (kvt0 + kv) * DH + d0);
-      if constexpr (TRQ) {
+      if constexpr (TRQ == 1) {
         int c = d0 >> 2;
         int g = (c & 7) | ((((kv & 3) ^ (c >> 3)) & 3) << 3);
         *reinterpret_cast<bf16x8*>(
@@ -1877,7 +1940,25 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
           bf16frag ktf;
-          if constexpr (TRQ) {
+          if constexpr (TRQ == 2) {
+            const int m_ = lane & 15;
+            const int colhi_ = (lane >> 4) & 1;
+            const int kvb0 = mch * 16 + half * 8 + (m_ >> 2);
+            const int c_r = dblk * 8 + colhi_ * 4 + (m_ & 3);
+            const int perm_ = (((c_r >> 1) & 3) << 2) | (c_r >> 3);
+            const int a0 =
+                kvb0 * 256 + ((kvb0 & 15) ^ perm_) * 16 + (c_r & 1) * 8;
+            const int a1 = (kvb0 + 4) * 256 +
+                           (((kvb0 + 4) & 15) ^ perm_) * 16 + (c_r & 1) * 8;
+            auto* kb3 = (__attribute__((address_space(3))) char*)ktr;
+            typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4t;
+            typedef __attribute__((address_space(3))) bf16x4t as3b4;
+            bf16x4t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (as3b4*)(kb3 + a0));
+            bf16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (as3b4*)(kb3 + a1));
+            ktf = __builtin_shufflevector(r0, r1, 0, 1, 2, 3, 4, 5, 6, 7);
+          } else if constexpr (TRQ == 1) {
             const int m_ = lane & 15;
             const int colhi_ = (lane >> 4) & 1;
             const int kvb0 = mch * 16 + half * 8 + (m_ >> 2);
@@ -1972,7 +2053,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dV), doc_start, doc_end, B,
                        Hq, Hkv, S, scale);
     VH_HIP(hipGetLastError());
-    hipLaunchKernelGGL((k_attn_bwd_dq<true, true>), grid_q, dim3(256), 24576, s,
+    hipLaunchKernelGGL((k_attn_bwd_dq<true, 1>), grid_q, dim3(256), 24576, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -1990,7 +2071,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
                        Hkv, S, scale);
     VH_HIP(hipGetLastError());
-    hipLaunchKernelGGL((k_attn_bwd_dq<false, true>), grid_q, dim3(256), 24576, s,
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 1>), grid_q, dim3(256), 24576, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -2081,8 +2162,16 @@ extern "C" int vh_attn_bwd2_dqprobe_bf16(const uint16_t* Q, const uint16_t* K,
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
-  if (mode == 20)
-    hipLaunchKernelGGL((k_attn_bwd_dq<false, true>), grid, dim3(256), 24576,
+  if (mode == 21)
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 2>), grid, dim3(256), 24576,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
+                       scale);
+  else if (mode == 20)
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 1>), grid, dim3(256), 24576,
                        s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -2090,7 +2179,7 @@ extern "C" int vh_attn_bwd2_dqprobe_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
                        scale);
   else
-    hipLaunchKernelGGL((k_attn_bwd_dq<false, false>), grid, dim3(256), 24576,
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 0>), grid, dim3(256), 24576,
                        s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
