@@ -444,14 +444,14 @@ extern "C" int vh_attn_fwd_bf16(const uint16_t* Q, const uint16_t* K,
   VH_CHECK(doc_start == nullptr || B == 1, "varlen requires packed B == 1");
   dim3 grid((uint32_t)(S / QB), (uint32_t)(B * Hq));
   if (doc_start) {
-    hipLaunchKernelGGL((k_attn_fwd<0, true>), grid, dim3(512), 65536, s,
+    hipLaunchKernelGGL((k_attn_fwd<0, true, 2>), grid, dim3(512), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
                        reinterpret_cast<bf16_t*>(O), LSE, doc_start, B, Hq,
                        Hkv, S, scale);
   } else {
-    hipLaunchKernelGGL((k_attn_fwd<0, false>), grid, dim3(512), 65536, s,
+    hipLaunchKernelGGL((k_attn_fwd<0, false, 2>), grid, dim3(512), 65536, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
