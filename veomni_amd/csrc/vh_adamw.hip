@@ -23,21 +23,33 @@ struct AdamArgs {
 };
 
 __global__ void k_adamw(AdamArgs a, int64_t total_vec) {
-  int64_t vi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  // per-BLOCK contiguous chunk, threads stride within it: wave reads stay
+  // coalesced while the tensor id is MONOTONE per thread — one binary
+  // search at entry, then an O(1) forward walk (the grid-stride form paid
+  // an 11-step prefix search per 8 elements: measured 86 ms -> the walk
+  // version targets the ~53 ms HBM bound for the 427 GB sweep)
+  const int64_t chunk =
+      (total_vec + gridDim.x - 1) / gridDim.x;
+  const int64_t v_begin = (int64_t)blockIdx.x * chunk + threadIdx.x;
+  const int64_t v_end = min((int64_t)(blockIdx.x + 1) * chunk, total_vec);
+  const int64_t stride = blockDim.x;
   float inv_scale = 1.f;
   if (a.grad_scale != nullptr) inv_scale = 1.f / *a.grad_scale;
   const float wd_mul = 1.f - a.lr * a.weight_decay;
   const float step_lr = a.lr / a.bc1;
   const float inv_bc2 = 1.f / a.bc2;
-  for (; vi < total_vec; vi += stride) {
-    int64_t e = vi * 8;
-    // binary search tensor id
-    int lo = 0, hi = a.T;
+  int lo = 0;
+  {
+    int64_t e0 = v_begin * 8;
+    int hi = a.T;
     while (lo + 1 < hi) {
       int mid = (lo + hi) >> 1;
-      if (e >= a.prefix[mid]) lo = mid; else hi = mid;
+      if (e0 >= a.prefix[mid]) lo = mid; else hi = mid;
     }
+  }
+  for (int64_t vi = v_begin; vi < v_end; vi += stride) {
+    int64_t e = vi * 8;
+    while (lo + 1 < a.T && e >= a.prefix[lo + 1]) ++lo;
     int64_t off = e - a.prefix[lo];
     bf16x8* pp = reinterpret_cast<bf16x8*>(a.p_ptrs[lo]) + (off >> 3);
     const bf16x8* gp = reinterpret_cast<const bf16x8*>(a.g_ptrs[lo]) + (off >> 3);
@@ -86,7 +98,7 @@ extern "C" int vh_adamw_bf16(const uint64_t* p_ptrs, const uint64_t* g_ptrs,
   a.grad_scale = grad_scale;
   int64_t total_vec = total / 8;
   int blocks = (int)((total_vec + 255) / 256);
-  if (blocks > 8192) blocks = 8192;
+  if (blocks > 4096) blocks = 4096;
   hipLaunchKernelGGL(k_adamw, dim3(blocks), dim3(256), 0, s, a, total_vec);
   VH_HIP(hipGetLastError());
   return 0;
