@@ -90,7 +90,7 @@ def main():
     for shp in ((1, 4, 2, 256), (2, 8, 2, 512)):
         q, k, v, do, delta, lse2, scale = prep(*shp)
         dq0 = run_dq(fnq, q, k, v, do, delta, lse2, scale, 0)
-        for mode, nm in ((20, "TRQ"), (21, "TRQ2")):
+        for mode, nm in ((20, "TRQ"), (21, "TRQ2"), (22, "TRQ3")):
             dq1 = run_dq(fnq, q, k, v, do, delta, lse2, scale, mode)
             eq = (dq1.float() - dq0.float()).abs().max().item()
             print(f"shape {shp}: dq {nm}-vs-v0 |dQ|={eq:.4g}", flush=True)
@@ -109,7 +109,7 @@ def main():
         t1.record()
         torch.cuda.synchronize()
         print(f"{name}: {t0.elapsed_time(t1) / 10 * 1000:.0f} us", flush=True)
-    for mode, name in ((0, "dq-v0"), (20, "dq-TRQ"), (21, "dq-TRQ2"), (20, "dq-TRQb"), (21, "dq-TRQ2b")):
+    for mode, name in ((0, "dq-v0"), (20, "dq-TRQ"), (22, "dq-TRQ3"), (20, "dq-TRQb"), (22, "dq-TRQ3b")):
         for _ in range(3):
             run_dq(fnq, q, k, v, do, delta, lse2, scale, mode)
         t0 = torch.cuda.Event(enable_timing=True)

@@ -1779,6 +1779,10 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dkv_g(
 // TRQ: tr16 hardware-transpose K^T reads over the bank-permuted natural
 // [32 kv][128 d] image (same scheme as dkv TR; see that comment).
 // TRQ 2: the L16 latin-square image filled by glds (dkv TR2G scheme).
+// TRQ 3: TRQ 2 + the WHOLE stage (krow/vrow kswz images + ktr L16) goes
+// glds into DOUBLE buffers (24 KB spare LDS), pipelined one kv tile
+// ahead with a single barrier per tile — the staging drain leaves the
+// barrier path entirely.
 template <bool DOC, int TRQ = 0>
 __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
@@ -1790,6 +1794,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
   bf16_t* krow = reinterpret_cast<bf16_t*>(smem);            // [32][128] 8 K
   bf16_t* vrow = reinterpret_cast<bf16_t*>(smem + 8192);     // [32][128] 8 K
   bf16_t* ktr = reinterpret_cast<bf16_t*>(smem + 16384);     // [128][32] 8 K
+  // TRQ 3 double buffers: second set at +24 KiB (launch carves 48 KiB)
 
   const int qb = blockIdx.x;           // q block of 128 rows
   const int bh = blockIdx.y;
@@ -1837,8 +1842,38 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
   for (int d = 0; d < 4; ++d) dq4[d] = f32x16{};
 
   const int kvtn = (int)((q0b + 128) / 32);
+  // TRQ 3: glds one whole tile set (krow/vrow kswz + ktr L16) into buffer Bf
+#define VH_DQ_STAGE(KVT2, BF)                                                  \
+  {                                                                            \
+    const int64_t kv0_ = (int64_t)(KVT2) * 32;                                 \
+    char* kb_ = reinterpret_cast<char*>(smem) + (BF) * 24576;                  \
+    _Pragma("unroll") for (int u = 0; u < 2; ++u) {                            \
+      const int win_ = wave * 2 + u;                                           \
+      const int row_ = win_ * 4 + (lane >> 4);                                 \
+      const int c16_ = (lane & 15) ^ (row_ & 15);                              \
+      const int pg_ = (((c16_ & 3) << 2) | (c16_ >> 2));                       \
+      glds16a(Kb + (kv0_ + row_) * DH + c16_ * 8,                              \
+              reinterpret_cast<bf16_t*>(kb_) + win_ * 512);                    \
+      glds16a(Vb + (kv0_ + row_) * DH + c16_ * 8,                              \
+              reinterpret_cast<bf16_t*>(kb_ + 8192) + win_ * 512);             \
+      glds16a(Kb + (kv0_ + row_) * DH + pg_ * 8,                               \
+              reinterpret_cast<bf16_t*>(kb_ + 16384) + win_ * 512);            \
+    }                                                                          \
+  }
+  int cur3 = 0;
+  if constexpr (TRQ == 3) {
+    if (kvt_begin < kvtn) VH_DQ_STAGE(kvt_begin, 0);
+    __syncthreads();
+  }
   for (int kvt = kvt_begin; kvt < kvtn; ++kvt) {
     const int64_t kvt0 = (int64_t)kvt * 32;
+    if constexpr (TRQ == 3) {
+      char* base_ = reinterpret_cast<char*>(smem) + cur3 * 24576;
+      krow = reinterpret_cast<bf16_t*>(base_);
+      vrow = reinterpret_cast<bf16_t*>(base_ + 8192);
+      ktr = reinterpret_cast<bf16_t*>(base_ + 16384);
+      if (kvt + 1 < kvtn) VH_DQ_STAGE(kvt + 1, cur3 ^ 1);
+    } else {
     // stage K/V rows [32][128] (2 passes) + K^T [128][32] (2 units each)
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
@@ -1889,6 +1924,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
       }
     }
     __syncthreads();
+    }
 
     const bool live = (kvt0 <= q0b + wave * 32 + 31) &&
                       (!DOC || kvt0 + 31 >= ds_wave);
@@ -1940,7 +1976,7 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
 #pragma unroll
         for (int mch = 0; mch < 2; ++mch) {
           bf16frag ktf;
-          if constexpr (TRQ == 2) {
+          if constexpr (TRQ >= 2) {
             const int m_ = lane & 15;
             const int colhi_ = (lane >> 4) & 1;
             const int kvb0 = mch * 16 + half * 8 + (m_ >> 2);
@@ -1983,7 +2019,9 @@ __global__ __launch_bounds__(256, 2) void k_attn_bwd_dq(
       }
     }
     __syncthreads();
+    cur3 ^= 1;
   }
+#undef VH_DQ_STAGE
 
   // single-contributor store: this block covers every kv for its q rows
 #pragma unroll
@@ -2162,7 +2200,15 @@ extern "C" int vh_attn_bwd2_dqprobe_bf16(const uint16_t* Q, const uint16_t* K,
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   VH_CHECK(S % 128 == 0, "S %% 128 != 0");
   dim3 grid((uint32_t)(S / 128), (uint32_t)(B * Hq));
-  if (mode == 21)
+  if (mode == 22)
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 3>), grid, dim3(256), 49152,
+                       s, reinterpret_cast<const bf16_t*>(Q),
+                       reinterpret_cast<const bf16_t*>(K),
+                       reinterpret_cast<const bf16_t*>(V),
+                       reinterpret_cast<const bf16_t*>(dO), delta, lse2,
+                       reinterpret_cast<bf16_t*>(dQ), nullptr, B, Hq, Hkv, S,
+                       scale);
+  else if (mode == 21)
     hipLaunchKernelGGL((k_attn_bwd_dq<false, 2>), grid, dim3(256), 24576,
                        s, reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
