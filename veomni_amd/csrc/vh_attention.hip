@@ -2091,7 +2091,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dV), doc_start, doc_end, B,
                        Hq, Hkv, S, scale);
     VH_HIP(hipGetLastError());
-    hipLaunchKernelGGL((k_attn_bwd_dq<true, 1>), grid_q, dim3(256), 24576, s,
+    hipLaunchKernelGGL((k_attn_bwd_dq<true, 3>), grid_q, dim3(256), 49152, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
@@ -2109,7 +2109,7 @@ extern "C" int vh_attn_bwd2_bf16(const uint16_t* Q, const uint16_t* K,
                        reinterpret_cast<bf16_t*>(dV), nullptr, nullptr, B, Hq,
                        Hkv, S, scale);
     VH_HIP(hipGetLastError());
-    hipLaunchKernelGGL((k_attn_bwd_dq<false, 1>), grid_q, dim3(256), 24576, s,
+    hipLaunchKernelGGL((k_attn_bwd_dq<false, 3>), grid_q, dim3(256), 49152, s,
                        reinterpret_cast<const bf16_t*>(Q),
                        reinterpret_cast<const bf16_t*>(K),
                        reinterpret_cast<const bf16_t*>(V),
