@@ -810,3 +810,53 @@ def _vl_moe_ep_async_sp(rank, ws):
 
 def test_vl_moe_ep_async_sp():
     spawn(_vl_moe_ep_async_sp)
+
+
+def _packed_varlen_sp(rank, ws):
+    """Packed multi-document batch under Ulysses SP (sync AND async): the
+    collator's cu_seq_lens describe the FULL sequence (computed before the
+    sp slice, ref data_collator.py:415), the attention core applies the
+    block-diagonal mask after the gather. Loss must match the
+    single-process packed run — and must DIFFER from a run that drops the
+    doc bounds (proves the mask is live)."""
+    from veomni_amd.data import sp_collate, synthetic_batch
+    from veomni_amd.distributed.parallel_state import (init_parallel_state,
+                                                       set_parallel_state)
+    from veomni_amd.models import build_model
+    from veomni_amd.models.modeling import bind_ops
+
+    bind_ops("eager")
+    torch.manual_seed(0)
+
+    full = synthetic_batch(512, 256, seed=3)
+    # pack two documents: positions restart at 128
+    pos = torch.cat([torch.arange(128), torch.arange(128)])[None]
+    full["position_ids"] = pos
+    from veomni_amd.data import fa_kwargs_from_position_ids
+
+    set_parallel_state(None)
+    ps_plain = init_parallel_state()
+    ref = build_model("tiny-dense")
+    full_ref = dict(full)
+    cu, ml = fa_kwargs_from_position_ids(pos)
+    full_ref["cu_seq_lens_q"] = cu
+    full_ref["cu_seq_lens_k"] = cu
+    rloss, _ = ref(**full_ref)
+    # dropping the bounds must change the loss (mask actually applied)
+    rloss_nodoc, _ = ref(**full)
+    assert not torch.allclose(rloss.detach(), rloss_nodoc.detach())
+
+    for async_u in (False, True):
+        set_parallel_state(None)
+        init_parallel_state(ulysses_size=ws, async_ulysses=async_u)
+        batch = sp_collate(full)
+        assert "cu_seq_lens_q" in batch  # full-sequence fa kwargs survive
+        loss, _ = ref(**batch)
+        torch.testing.assert_close(loss.detach().float(),
+                                   rloss.detach().float(),
+                                   rtol=5e-3, atol=5e-4)
+    set_parallel_state(ps_plain)
+
+
+def test_packed_varlen_sp():
+    spawn(_packed_varlen_sp)

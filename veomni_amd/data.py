@@ -86,7 +86,13 @@ def sp_slice(t: torch.Tensor, sp_size: int, sp_rank: int, dim: int = -1) -> torc
 
 
 def sp_collate(batch: dict, sp_size: int | None = None, sp_rank: int | None = None) -> dict:
-    """SequenceParallelCollator semantics for the text path (ref :317-427)."""
+    """SequenceParallelCollator semantics for the text path (ref :317-427).
+
+    Packed batches: the flash-attention cu_seq_lens are computed from the
+    FULL (sp-padded) position_ids BEFORE slicing (ref :415 — the kwargs
+    describe the gathered sequence the Ulysses exchange reconstructs; the
+    sp-pad tail becomes its own 1-token segment via the position_ids == 0
+    convention)."""
     ps = get_parallel_state()
     sp_size = ps.sp_size if sp_size is None else sp_size
     sp_rank = ps.sp_rank if sp_rank is None else sp_rank
@@ -96,7 +102,13 @@ def sp_collate(batch: dict, sp_size: int | None = None, sp_rank: int | None = No
     out["labels"] = sp_slice(sp_pad(sp_shift(batch["labels"]), sp_size, IGNORE_INDEX), sp_size, sp_rank)
     out["input_ids"] = sp_slice(sp_pad(batch["input_ids"], sp_size, 0), sp_size, sp_rank)
     if "position_ids" in batch:
-        out["position_ids"] = sp_slice(sp_pad(batch["position_ids"], sp_size, 0), sp_size, sp_rank)
+        padded_pos = sp_pad(batch["position_ids"], sp_size, 0)
+        cu, max_len = fa_kwargs_from_position_ids(padded_pos)
+        out["cu_seq_lens_q"] = cu
+        out["cu_seq_lens_k"] = cu
+        out["max_length_q"] = max_len
+        out["max_length_k"] = max_len
+        out["position_ids"] = sp_slice(padded_pos, sp_size, sp_rank)
     return out
 
 
