@@ -264,12 +264,12 @@ class _A2AStartSeqHeads(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output: Tensor):
-        work = _ASYNC_WORK.pop(id(grad_output), None)
+        ent = _ASYNC_WORK.pop(id(grad_output), None)
         ws = dist.get_world_size(ctx.group)
-        if work is not None:
+        if ent is not None:
             # mirrored async: _A2AWait.backward started the reverse exchange;
             # finish it and apply the post-concat (scatter seq, gather heads)
-            work.wait()
+            ent[0].wait()
             out = torch.cat(grad_output.split(grad_output.size(0) // ws),
                             dim=ctx.head_dim)
             return (None, out, None, None)
@@ -289,11 +289,16 @@ class _A2AWait(torch.autograd.Function):
     def backward(ctx, grad_output: Tensor):
         if ctx.group is not None:
             # start the reverse exchange NOW; the paired _A2AStartSeqHeads
-            # backward (several autograd nodes later) finishes it
+            # backward (several autograd nodes later) finishes it. The stash
+            # holds (work, input, output) so both buffers stay alive until
+            # the pop even if the engine were ever to hand the pair node a
+            # different (accumulated) tensor — in that unreachable-by-
+            # construction case the pair falls back to the sync exchange
+            # and the orphan entry keeps the in-flight buffers valid.
             x = grad_output.contiguous()
             out = torch.empty_like(x)
             work = dist.all_to_all_single(out, x, group=ctx.group, async_op=True)
-            _ASYNC_WORK[id(out)] = work
+            _ASYNC_WORK[id(out)] = (work, x, out)
             return out, None
         return grad_output, None
 
