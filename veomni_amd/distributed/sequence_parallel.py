@@ -259,7 +259,8 @@ class _A2AStartSeqHeads(torch.autograd.Function):
         )
         out = torch.empty_like(xs)
         work = dist.all_to_all_single(out, xs, group=group, async_op=True)
-        _ASYNC_WORK[id(out)] = work
+        # (work, input, output): both buffers stay referenced until the wait
+        _ASYNC_WORK[id(out)] = (work, xs, out)
         return out
 
     @staticmethod
@@ -280,9 +281,9 @@ class _A2AWait(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: Tensor, group) -> Tensor:
         ctx.group = group
-        work = _ASYNC_WORK.pop(id(x), None)
-        if work is not None:
-            work.wait()
+        ent = _ASYNC_WORK.pop(id(x), None)
+        if ent is not None:
+            ent[0].wait()
         return x.view_as(x)
 
     @staticmethod
